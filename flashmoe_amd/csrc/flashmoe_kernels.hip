@@ -1,0 +1,877 @@
+// MI355X-native (gfx950 / CDNA4) implementation of the FlashDMoE hot path:
+//   gate -> token routing -> grouped expert FFN (MFMA) -> weighted combine
+// Built from scratch against the C-ABI in include/flashmoe_abi.h; semantics
+// restated from the reference (citations per kernel, file:line into
+// /root/reference). No CUDA compatibility paths; wave64 / MFMA / LDS only.
+//
+// Round-1 structure: separate kernels on one stream.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
+
+#include <cmath>
+#include <cstdint>
+#include <cstdio>
+#include <cstring>
+#include <algorithm>
+
+#include "../../include/flashmoe_abi.h"
+
+// ---------------------------------------------------------------------------
+// Common types / helpers
+// ---------------------------------------------------------------------------
+
+using bf16 = __hip_bfloat16;
+
+// token slot descriptor (reference TPS, types.cuh:299-312): token index +
+// sum of the token's selected top-k probabilities (gate.cuh:669,711-715)
+struct __align__(8) TPS {
+  uint32_t tokenIdx;
+  float probSum;
+};
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(4))) uint32_t u32x4;
+
+__device__ __forceinline__ float toF(float v) { return v; }
+__device__ __forceinline__ float toF(bf16 v) { return __bfloat162float(v); }
+__device__ __forceinline__ void fromF(float v, float& o) { o = v; }
+__device__ __forceinline__ void fromF(float v, bf16& o) { o = __float2bfloat16(v); }
+
+__device__ __forceinline__ float applyAct(float v, int act) {
+  // hidden_act 0: ReLU, 1: exact-erf GELU (cutlass epilogue::thread::{ReLU,GELU},
+  // schema csrc/flashmoe_config.schema.json:33-37, types.cuh:151-159)
+  if (act == 0) return fmaxf(v, 0.0f);
+  return 0.5f * v * (1.0f + erff(v * 0.70710678118654752f));
+}
+
+#define DIVUP(a, b) (((a) + (b) - 1) / (b))
+
+// ---------------------------------------------------------------------------
+// Gate kernel (reference: moe/gate.cuh:474-720 singleBlock semantics; the
+// multiBlock ring for E>64 computes the same values, gate.cuh:93-468).
+//
+// One block per 128-token tile, 256 threads.  Per tile:
+//   1. logits[128][E] fp32 via LDS-staged chunked dot products (2 thr/token)
+//   2. per token: online softmax over E (padded cols excluded,
+//      gate.cuh:578-585); iterative strict-> first-index-wins top-k on the
+//      LOGITS (monotone-equivalent to probs, gate.cuh:662-668); mCw = sum of
+//      selected probs (gate.cuh:669); write gate_out row (probs, Element)
+//   3. per expert: intra-tile count in token order + one atomicAdd(eC) base
+//      (gate.cuh:688-716 - inter-tile order nondeterministic, as reference)
+//   4. per token: tokenIds[e][base+local] = {token, mCw} if slot < EC
+//      (capacity, gate.cuh:711-715)
+// Supports E <= 128 this round (covers BASELINE configs 1-4; E=256 next).
+// ---------------------------------------------------------------------------
+
+template <typename T, int K>
+__global__ __launch_bounds__(256) void k_gate(
+    const T* __restrict__ x, const T* __restrict__ gate_w,
+    T* __restrict__ gate_out, TPS* __restrict__ tokenIds,
+    uint32_t* __restrict__ eC, int S, int H, int E, int PX, int EC, int pEC) {
+  constexpr int BM = 128;
+  constexpr int BK = 64;
+  constexpr int RPAD = 8;  // row pad (elements) to break LDS bank alignment
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // arena layout (host mirrors gate_lds_bytes()):
+  float* logits = reinterpret_cast<float*>(smem);            // [BM][E+1]
+  T* aCh = reinterpret_cast<T*>(logits + BM * (E + 1));      // [BM][BK+RPAD]
+  T* gCh = aCh + BM * (BK + RPAD);                           // [E][BK+RPAD]
+  // post-logits arena reuses the A-chunk region:
+  uint16_t* sel = reinterpret_cast<uint16_t*>(aCh);          // [BM][K]
+  uint16_t* localIdx = sel + BM * K;                         // [BM][K]
+  uint32_t* base = reinterpret_cast<uint32_t*>(localIdx + BM * K);  // [E]
+
+  const int tid = threadIdx.x;
+  const int m0 = blockIdx.x * BM;
+  const int tok = tid & (BM - 1);
+  const int half = tid >> 7;              // 2 threads per token
+  const int E2 = (E + 1) / 2;
+  const int e0 = half * E2;
+  const int e1 = min(e0 + E2, E);
+
+  // zero the logits accumulator (accumulated in LDS across K chunks;
+  // a register array would be runtime-indexed -> scratch, guide rule 20)
+  for (int i = tid; i < BM * (E + 1); i += 256) logits[i] = 0.0f;
+
+  const int LDA = BK + RPAD;
+  for (int kc = 0; kc < H; kc += BK) {
+    // stage A chunk [BM][BK] and G chunk [E][BK], 16B units, coalesced
+    constexpr int EPU = 16 / sizeof(T);  // elements per 16B unit
+    {
+      const int unitsA = BM * BK / EPU;
+      for (int u = tid; u < unitsA; u += 256) {
+        const int row = u / (BK / EPU);
+        const int cu = u % (BK / EPU);
+        const T* src = x + (size_t)(m0 + row) * H + kc + cu * EPU;
+        T* dst = aCh + row * LDA + cu * EPU;
+        *reinterpret_cast<u32x4*>(dst) = *reinterpret_cast<const u32x4*>(src);
+      }
+      const int unitsG = E * BK / EPU;
+      for (int u = tid; u < unitsG; u += 256) {
+        const int row = u / (BK / EPU);
+        const int cu = u % (BK / EPU);
+        const T* src = gate_w + (size_t)row * H + kc + cu * EPU;
+        T* dst = gCh + row * LDA + cu * EPU;
+        *reinterpret_cast<u32x4*>(dst) = *reinterpret_cast<const u32x4*>(src);
+      }
+    }
+    __syncthreads();
+    const T* arow = aCh + tok * LDA;
+    for (int e = e0; e < e1; ++e) {
+      const T* grow = gCh + e * LDA;
+      float s = 0.0f;
+#pragma unroll 8
+      for (int j = 0; j < BK; ++j) s = fmaf(toF(arow[j]), toF(grow[j]), s);
+      logits[tok * (E + 1) + e] += s;
+    }
+    __syncthreads();
+  }
+
+  // phase 2: softmax + top-k (thread t < 128 owns token t)
+  float mCw = 0.0f;
+  uint16_t mySel[K];
+  if (tid < BM) {
+    const float* lrow = logits + tid * (E + 1);
+    float m = -INFINITY;
+    for (int e = 0; e < E; ++e) m = fmaxf(m, lrow[e]);
+    float d = 0.0f;
+    for (int e = 0; e < E; ++e) d += __expf(lrow[e] - m);
+    const float inv_d = 1.0f / d;
+    T* grow = gate_out + (size_t)(m0 + tid) * PX;
+    for (int e = 0; e < E; ++e) fromF(__expf(lrow[e] - m) * inv_d, grow[e]);
+    for (int e = E; e < PX; ++e) fromF(0.0f, grow[e]);
+    // iterative argmax on logits, strict >, first index wins
+    uint64_t taken_lo = 0, taken_hi = 0;  // E <= 128
+#pragma unroll
+    for (int i = 0; i < K; ++i) {
+      float sV = -INFINITY;
+      int sIdx = 0;
+      for (int e = 0; e < E; ++e) {
+        const bool taken = (e < 64) ? ((taken_lo >> e) & 1)
+                                    : ((taken_hi >> (e - 64)) & 1);
+        if (!taken && lrow[e] > sV) { sV = lrow[e]; sIdx = e; }
+      }
+      if (sIdx < 64) taken_lo |= 1ull << sIdx; else taken_hi |= 1ull << (sIdx - 64);
+      mySel[i] = (uint16_t)sIdx;
+      mCw += __expf(sV - m) * inv_d;
+    }
+  }
+  __syncthreads();  // logits no longer needed; sel arena reuses A-chunk space
+  if (tid < BM) {
+#pragma unroll
+    for (int i = 0; i < K; ++i) sel[tid * K + i] = mySel[i];
+  }
+  __syncthreads();
+
+  // phase 3: per-expert intra-tile ordering + global base via atomicAdd
+  if (tid < E) {
+    uint32_t cnt = 0;
+    for (int mj = 0; mj < BM * K; ++mj) {
+      if (sel[mj] == tid) localIdx[mj] = (uint16_t)cnt++;
+    }
+    base[tid] = atomicAdd(eC + tid, cnt);
+  }
+  __syncthreads();
+
+  // phase 4: write token slots (capacity-clipped)
+  if (tid < BM) {
+#pragma unroll
+    for (int i = 0; i < K; ++i) {
+      const int e = mySel[i];
+      const uint32_t slot = base[e] + localIdx[tid * K + i];
+      if (slot < (uint32_t)EC) {
+        tokenIds[(size_t)e * pEC + slot] = TPS{(uint32_t)(m0 + tid), mCw};
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Grouped expert GEMM, bf16 MFMA (reference: os/processor/processor.cuh
+// fGET preGEMM :339-468 / postGEMM :711-751, gemm.cuh FAA epilogue, combine
+// :44-205). 128x128 tile, 4 waves (2x2), 64x64 per wave as 4x4
+// v_mfma_f32_16x16x32_bf16 fragments, BK=64, LDS-staged NT operands.
+//
+// PHASE 0 (up):   A = x rows gathered via tokenIds[e]; B = Wup[e] [P,H];
+//                 epilogue act(acc + b_up) -> xM[e]
+// PHASE 1 (down): A = xM[e]; B = Wdn[e] viewed [H,P] (reference
+//                 reinterpretation, moe.cuh:114-116); epilogue
+//                 z = acc + b_dn; k>1: atomicAdd(O32, gate_out*z/probSum)
+//                 (processor.cuh:126-168); k==1: unscaled store to moe_out
+//                 (processor.cuh:173-204)
+// PHASE 2 (down-direct, EP path): like 1 but plain store to out_rows[m].
+// If tokenIds == nullptr, the row gather is identity and routed = n_rows
+// (packed-rows mode for fm_expert_ffn).
+// ---------------------------------------------------------------------------
+
+struct GemmArgs {
+  const void* A;        // phase0: x [S,H]; phase1/2: xM_e rows [pEC,K]
+  const void* B;        // weight rows [N,K] (K contiguous)
+  const void* bias;     // [N] or null
+  void* out;            // phase0: xM_e [pEC,P]; phase2: out_rows
+  float* O32;           // phase1 k>1
+  void* moe_out;        // phase1 k==1
+  const void* gate_out; // [S,PX]
+  const TPS* tokenIds;  // [E,pEC] (this expert's row = tokenIds + e*pEC)
+  const uint32_t* eC;
+  long long strideAExpert;  // elements between experts in A (phase1: pEC*K)
+  long long strideBExpert;  // elements between experts in B (2*P*H)
+  long long strideOExpert;  // elements between experts in out
+  int K;                // reduction dim
+  int N;                // output dim
+  int EC, pEC, PX;
+  int topk;
+  int act;
+  int expertOffset;     // global expert id of blockIdx.z==0
+  int nRows;            // packed-rows mode row count
+  int H;                // row stride of x / O32 / moe_out
+};
+
+template <int PHASE>
+__global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
+  constexpr int BM = 128, BN = 128, BK = 64;
+  constexpr int LDT = BK + 8;  // 16B row pad -> conflict-free b128 frag reads
+  __shared__ __attribute__((aligned(16))) bf16 Alds[BM * LDT];
+  __shared__ __attribute__((aligned(16))) bf16 Blds[BN * LDT];
+  __shared__ TPS sTps[BM];
+  __shared__ uint32_t sRouted;
+
+  const int e = blockIdx.z;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int m0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+  const int K = a.K, N = a.N;
+
+  const TPS* tpsE = a.tokenIds ? a.tokenIds + (size_t)e * a.pEC : nullptr;
+  if (tid == 0) {
+    sRouted = tpsE ? min(a.eC[e], (uint32_t)a.EC) : (uint32_t)a.nRows;
+  }
+  __syncthreads();
+  const uint32_t routed = sRouted;
+  if ((uint32_t)m0 >= routed) return;  // empty tile (0-token expert / tail)
+
+  if (tid < BM) {
+    TPS t{0u, 1.0f};
+    if ((uint32_t)(m0 + tid) < routed) {
+      t = tpsE ? tpsE[m0 + tid] : TPS{(uint32_t)(m0 + tid), 1.0f};
+    }
+    sTps[tid] = t;
+  }
+  __syncthreads();
+
+  const bf16* __restrict__ Ag = reinterpret_cast<const bf16*>(a.A);
+  const bf16* __restrict__ Bg =
+      reinterpret_cast<const bf16*>(a.B) + (size_t)e * a.strideBExpert;
+
+  // accumulators: wave (wr,wc) owns the 64x64 subtile at (wr*64, wc*64)
+  const int wr = wave >> 1, wc = wave & 1;
+  f32x4 accv[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) accv[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  // staging map: 16B units; unit u -> row u/8, col8 u%8 (8 bf16 per unit)
+  // thread t stages units {t, t+256, t+512, t+768} -> coalesced
+  const int aRowStride = (PHASE == 0) ? a.H : K;  // x rows vs xM rows
+  for (int kt = 0; kt < K; kt += BK) {
+#pragma unroll
+    for (int u4 = 0; u4 < 4; ++u4) {
+      const int u = tid + u4 * 256;
+      const int row = u >> 3, c8 = (u & 7) * 8;
+      // A row source: phase0 gathers x[token]; else xM row (m0+row)
+      size_t srow;
+      if (PHASE == 0) {
+        srow = (size_t)sTps[row].tokenIdx;
+      } else {
+        srow = (size_t)(m0 + row) + (size_t)e * 0;  // A already expert-based
+      }
+      const bf16* src = (PHASE == 0)
+          ? Ag + srow * aRowStride + kt + c8
+          : Ag + (size_t)e * a.strideAExpert + (size_t)(m0 + row) * K + kt + c8;
+      *reinterpret_cast<u32x4*>(&Alds[row * LDT + c8]) =
+          *reinterpret_cast<const u32x4*>(src);
+      const int brow = min(n0 + row, N - 1);
+      const bf16* bsrc = Bg + (size_t)brow * K + kt + c8;
+      *reinterpret_cast<u32x4*>(&Blds[row * LDT + c8]) =
+          *reinterpret_cast<const u32x4*>(bsrc);
+    }
+    __syncthreads();
+#pragma unroll
+    for (int s = 0; s < 2; ++s) {  // two K=32 MFMA steps per tile
+      bf16x8 af[4], bf[4];
+      const int kofs = 32 * s + (lane >> 4) * 8;
+      const int rl = lane & 15;
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        af[mi] = *reinterpret_cast<const bf16x8*>(
+            &Alds[(wr * 64 + mi * 16 + rl) * LDT + kofs]);
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        bf[ni] = *reinterpret_cast<const bf16x8*>(
+            &Blds[(wc * 64 + ni * 16 + rl) * LDT + kofs]);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          accv[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bf[ni], accv[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: C/D map for 16x16 MFMA: col = lane&15, row = (lane>>4)*4 + r
+  const int cl = lane & 15;
+  const int r0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = wr * 64 + mi * 16 + r0 + r;
+      const int m = m0 + row;
+      if ((uint32_t)m >= routed) continue;
+      const TPS tp = sTps[row];
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        const int col = n0 + wc * 64 + ni * 16 + cl;
+        if (col >= N) continue;
+        float v = accv[mi][ni][r];
+        if (a.bias)
+          v += toF(reinterpret_cast<const bf16*>(a.bias)[col]);
+        if constexpr (PHASE == 0) {
+          v = applyAct(v, a.act);
+          reinterpret_cast<bf16*>(a.out)[(size_t)e * a.strideOExpert +
+                                         (size_t)m * N + col] =
+              __float2bfloat16(v);
+        } else if constexpr (PHASE == 1) {
+          if (a.topk > 1) {
+            const float prob = toF(reinterpret_cast<const bf16*>(
+                a.gate_out)[(size_t)tp.tokenIdx * a.PX + a.expertOffset + e]);
+            atomicAdd(&a.O32[(size_t)tp.tokenIdx * a.H + col],
+                      v * prob / tp.probSum);
+          } else {
+            reinterpret_cast<bf16*>(
+                a.moe_out)[(size_t)tp.tokenIdx * a.H + col] =
+                __float2bfloat16(v);
+          }
+        } else {  // PHASE 2: packed-rows direct output
+          reinterpret_cast<bf16*>(a.out)[(size_t)m * N + col] =
+              __float2bfloat16(v);
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// fp32 grouped GEMM (config-1 "CPU correctness plumbing" shapes; VALU
+// tiled). 64x64 tile, 256 threads, each thread a 4x4 sub-block, BK=16.
+// Same phase semantics as the bf16 kernel.
+// ---------------------------------------------------------------------------
+
+template <int PHASE>
+__global__ __launch_bounds__(256) void k_group_gemm_f32(GemmArgs a) {
+  constexpr int BM = 64, BN = 64, BK = 16;
+  constexpr int LDT = BK + 1;
+  __shared__ float Alds[BM * LDT];
+  __shared__ float Blds[BN * LDT];
+  __shared__ TPS sTps[BM];
+  __shared__ uint32_t sRouted;
+
+  const int e = blockIdx.z;
+  const int tid = threadIdx.x;
+  const int m0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+  const int K = a.K, N = a.N;
+  const TPS* tpsE = a.tokenIds ? a.tokenIds + (size_t)e * a.pEC : nullptr;
+  if (tid == 0)
+    sRouted = tpsE ? min(a.eC[e], (uint32_t)a.EC) : (uint32_t)a.nRows;
+  __syncthreads();
+  const uint32_t routed = sRouted;
+  if ((uint32_t)m0 >= routed) return;
+  if (tid < BM) {
+    TPS t{0u, 1.0f};
+    if ((uint32_t)(m0 + tid) < routed)
+      t = tpsE ? tpsE[m0 + tid] : TPS{(uint32_t)(m0 + tid), 1.0f};
+    sTps[tid] = t;
+  }
+  __syncthreads();
+
+  const float* __restrict__ Ag = reinterpret_cast<const float*>(a.A);
+  const float* __restrict__ Bg =
+      reinterpret_cast<const float*>(a.B) + (size_t)e * a.strideBExpert;
+
+  const int tr = (tid / 16) * 4;  // thread rows [tr, tr+4)
+  const int tc = (tid % 16) * 4;
+  float acc[4][4] = {};
+  for (int kt = 0; kt < K; kt += BK) {
+    // stage: BM*BK = 1024 floats, 256 threads x 4
+#pragma unroll
+    for (int u4 = 0; u4 < 4; ++u4) {
+      const int u = tid + u4 * 256;
+      const int row = u / BK, cc = u % BK;
+      const float* src = (PHASE == 0)
+          ? Ag + (size_t)sTps[row].tokenIdx * a.H + kt + cc
+          : Ag + (size_t)e * a.strideAExpert + (size_t)(m0 + row) * K + kt + cc;
+      Alds[row * LDT + cc] = *src;
+      const int brow = min(n0 + row, N - 1);
+      Blds[row * LDT + cc] = Bg[(size_t)brow * K + kt + cc];
+    }
+    __syncthreads();
+#pragma unroll
+    for (int kk = 0; kk < BK; ++kk) {
+      float av[4], bv[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) av[i] = Alds[(tr + i) * LDT + kk];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) bv[j] = Blds[(tc + j) * LDT + kk];
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j) acc[i][j] = fmaf(av[i], bv[j], acc[i][j]);
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int m = m0 + tr + i;
+    if ((uint32_t)m >= routed) continue;
+    const TPS tp = sTps[tr + i];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int col = n0 + tc + j;
+      if (col >= N) continue;
+      float v = acc[i][j];
+      if (a.bias) v += reinterpret_cast<const float*>(a.bias)[col];
+      if constexpr (PHASE == 0) {
+        v = applyAct(v, a.act);
+        reinterpret_cast<float*>(a.out)[(size_t)e * a.strideOExpert +
+                                        (size_t)m * N + col] = v;
+      } else if constexpr (PHASE == 1) {
+        if (a.topk > 1) {
+          const float prob = reinterpret_cast<const float*>(
+              a.gate_out)[(size_t)tp.tokenIdx * a.PX + a.expertOffset + e];
+          atomicAdd(&a.O32[(size_t)tp.tokenIdx * a.H + col],
+                    v * prob / tp.probSum);
+        } else {
+          reinterpret_cast<float*>(a.moe_out)[(size_t)tp.tokenIdx * a.H +
+                                              col] = v;
+        }
+      } else {
+        reinterpret_cast<float*>(a.out)[(size_t)m * N + col] = v;
+      }
+    }
+  }
+}
+
+// cast the fp32 combine accumulator into the Element output (k>1 path)
+template <typename T>
+__global__ void k_cast_out(const float* __restrict__ O32, T* __restrict__ out,
+                           size_t n) {
+  const size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (size_t j = i; j < n; j += stride) fromF(O32[j], out[j]);
+}
+
+// combine pre-packed return rows (EP host path): one block per row
+template <typename T, bool SCALED>
+__global__ void k_combine_rows(const T* __restrict__ rows,
+                               const uint32_t* __restrict__ tokenIdx,
+                               const float* __restrict__ scale,
+                               float* __restrict__ O32, T* __restrict__ moeOut,
+                               int H) {
+  const int i = blockIdx.x;
+  const uint32_t t = tokenIdx[i];
+  const float s = SCALED ? scale[i] : 1.0f;
+  for (int h = threadIdx.x; h < H; h += blockDim.x) {
+    const float v = toF(rows[(size_t)i * H + h]);
+    if (SCALED)
+      atomicAdd(&O32[(size_t)t * H + h], s * v);
+    else
+      moeOut[(size_t)t * H + h] = rows[(size_t)i * H + h];
+  }
+}
+
+// MFMA layout probe (test-only): D = A[16x32] x B[32x16] via one
+// mfma_f32_16x16x32_bf16, written with the assumed C/D mapping.
+__global__ void k_mfma_probe(const bf16* A, const bf16* B, float* D) {
+  const int lane = threadIdx.x & 63;
+  bf16x8 af, bfr;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    af[j] = *reinterpret_cast<const __bf16*>(
+        &A[(lane & 15) * 32 + (lane >> 4) * 8 + j]);
+    bfr[j] = *reinterpret_cast<const __bf16*>(
+        &B[(lane & 15) * 32 + (lane >> 4) * 8 + j]);  // B given as [16 col][32 k]
+  }
+  f32x4 c{0.f, 0.f, 0.f, 0.f};
+  c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bfr, c, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) D[((lane >> 4) * 4 + r) * 16 + (lane & 15)] = c[r];
+}
+
+// ===========================================================================
+// Host side: state, workspace, ABI
+// ===========================================================================
+
+namespace {
+
+struct State {
+  fm_config cfg{};
+  bool initialized = false;
+  int rank = 0, world = 1;
+  int S = 0, H = 0, P = 0, E = 0, PX = 0, nLx = 0, EC = 0, pEC = 0;
+  size_t esz = 0;
+  // workspace
+  TPS* tokenIds = nullptr;   // [E, pEC]
+  uint32_t* eC = nullptr;    // [E]
+  void* xM = nullptr;        // [nLx_alloc, pEC, P] Element
+  float* O32 = nullptr;      // [S, H]
+  int nLxAlloc = 0;
+};
+State g;
+
+thread_local char g_err[512] = "";
+
+void setErr(const char* msg) {
+  snprintf(g_err, sizeof(g_err), "%s", msg);
+}
+
+#define FM_HIP_CHECK(expr)                                                   \
+  do {                                                                       \
+    hipError_t _e = (expr);                                                  \
+    if (_e != hipSuccess) {                                                  \
+      snprintf(g_err, sizeof(g_err), "HIP error %s at %s:%d: %s",            \
+               hipGetErrorString(_e), __FILE__, __LINE__, #expr);            \
+      return FM_ERR_HIP;                                                     \
+    }                                                                        \
+  } while (0)
+
+size_t gate_lds_bytes(int E, size_t esz) {
+  // mirrors the k_gate arena: logits + A chunk + G chunk
+  return 128 * (E + 1) * sizeof(float) + (128 + E) * (64 + 8) * esz;
+}
+
+int launch_gate(hipStream_t st, const void* x, const void* gate_w,
+                void* gate_out, int64_t S) {
+  dim3 grid(S / 128), block(256);
+  const size_t lds = gate_lds_bytes(g.E, g.esz);
+#define GATE_CASE(T, KK)                                                      \
+  do {                                                                        \
+    if (lds > 64 * 1024) {                                                    \
+      hipFuncSetAttribute(                                                    \
+          reinterpret_cast<const void*>(&k_gate<T, KK>),                      \
+          hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);              \
+    }                                                                         \
+    hipLaunchKernelGGL((k_gate<T, KK>), grid, block, lds, st,                 \
+                       reinterpret_cast<const T*>(x),                         \
+                       reinterpret_cast<const T*>(gate_w),                    \
+                       reinterpret_cast<T*>(gate_out), g.tokenIds, g.eC,      \
+                       (int)S, g.H, g.E, g.PX, g.EC, g.pEC);                  \
+  } while (0)
+#define GATE_K(T)                                                             \
+  switch (g.cfg.expert_top_k) {                                               \
+    case 1: GATE_CASE(T, 1); break;                                           \
+    case 2: GATE_CASE(T, 2); break;                                           \
+    case 4: GATE_CASE(T, 4); break;                                           \
+    case 8: GATE_CASE(T, 8); break;                                           \
+    default: setErr("unsupported expert_top_k (1,2,4,8)"); return FM_ERR_UNSUPPORTED; \
+  }
+  if (g.cfg.dtype == 2) {
+    GATE_K(bf16)
+  } else {
+    GATE_K(float)
+  }
+#undef GATE_K
+#undef GATE_CASE
+  FM_HIP_CHECK(hipGetLastError());
+  return FM_OK;
+}
+
+}  // namespace
+
+extern "C" {
+
+const char* fm_last_error(void) { return g_err; }
+int fm_built_for_gfx950(void) { return 1; }
+
+int fm_initialize(const fm_config* cfg, int rank, int world_size) {
+  if (g.initialized) { setErr("already initialized"); return FM_ERR_STATE; }
+  if (!cfg) { setErr("null config"); return FM_ERR_STATE; }
+  g.cfg = *cfg;
+  g.rank = rank;
+  g.world = world_size;
+  g.S = cfg->sequence_len * cfg->mini_batch;
+  g.H = cfg->hidden_size;
+  g.P = cfg->intermediate_size;
+  g.E = cfg->num_experts;
+  g.PX = DIVUP(g.E, 64) * 64;
+  if (cfg->num_experts % world_size != 0) {
+    setErr("num_experts must divide by world_size (uniform EP split)");
+    return FM_ERR_SHAPE;
+  }
+  g.nLx = g.E / world_size;
+  const int base = cfg->drop_tokens ? DIVUP(g.S, g.E) : g.S;
+  g.EC = base * cfg->capacity_factor * cfg->expert_top_k;
+  g.pEC = DIVUP(g.EC, 128) * 128;
+  switch (cfg->dtype) {
+    case 0: case 1: g.esz = 4; break;
+    case 2: g.esz = 2; break;
+    default: setErr("dtype not supported this round (fp32/bf16)");
+             return FM_ERR_UNSUPPORTED;
+  }
+  if (g.E > 128) { setErr("E > 128 not supported this round"); return FM_ERR_UNSUPPORTED; }
+  if (g.H % 64 || g.P % 64) { setErr("H and P must be multiples of 64"); return FM_ERR_SHAPE; }
+  if (g.esz == 2 && (g.H % 128 || g.P % 128)) {
+    setErr("bf16 path requires H, P multiples of 128 this round");
+    return FM_ERR_SHAPE;
+  }
+  // workspace: tokenIds/eC global-E; xM sized for the worst consumer
+  // (single-rank path: E experts; EP path: world*EC rows per local expert)
+  g.nLxAlloc = (world_size == 1) ? g.E
+                                 : g.nLx * DIVUP(world_size * g.EC, g.pEC);
+  FM_HIP_CHECK(hipMalloc(&g.tokenIds, (size_t)g.E * g.pEC * sizeof(TPS)));
+  FM_HIP_CHECK(hipMalloc(&g.eC, (size_t)g.E * sizeof(uint32_t)));
+  FM_HIP_CHECK(hipMalloc(&g.xM, (size_t)g.nLxAlloc * g.pEC * g.P * g.esz));
+  FM_HIP_CHECK(hipMalloc(&g.O32, (size_t)g.S * g.H * sizeof(float)));
+  g.initialized = true;
+  return FM_OK;
+}
+
+int fm_finalize(void) {
+  if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
+  hipFree(g.tokenIds); hipFree(g.eC); hipFree(g.xM); hipFree(g.O32);
+  g = State{};
+  return FM_OK;
+}
+
+int fm_get_compiled_config(int64_t* S, int64_t* H, int64_t* E, int64_t* P,
+                           int64_t* PX, int64_t* element_size) {
+  if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
+  if (S) *S = g.S;
+  if (H) *H = g.H;
+  if (E) *E = g.E;
+  if (P) *P = g.P;
+  if (PX) *PX = g.PX;
+  if (element_size) *element_size = (int64_t)g.esz;
+  return FM_OK;
+}
+
+int fm_get_num_local_experts(void) { return g.initialized ? g.nLx : -1; }
+
+static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
+                             int mTiles, int nTiles, int nE) {
+  dim3 block(256);
+  if (g.esz == 2) {
+    dim3 grid(mTiles, nTiles, nE);
+    switch (phase) {
+      case 0: hipLaunchKernelGGL(k_group_gemm_bf16<0>, grid, block, 0, st, a); break;
+      case 1: hipLaunchKernelGGL(k_group_gemm_bf16<1>, grid, block, 0, st, a); break;
+      default: hipLaunchKernelGGL(k_group_gemm_bf16<2>, grid, block, 0, st, a); break;
+    }
+  } else {
+    dim3 grid(mTiles * 2, nTiles * 2, nE);  // 64x64 tiles
+    switch (phase) {
+      case 0: hipLaunchKernelGGL(k_group_gemm_f32<0>, grid, block, 0, st, a); break;
+      case 1: hipLaunchKernelGGL(k_group_gemm_f32<1>, grid, block, 0, st, a); break;
+      default: hipLaunchKernelGGL(k_group_gemm_f32<2>, grid, block, 0, st, a); break;
+    }
+  }
+  FM_HIP_CHECK(hipGetLastError());
+  return FM_OK;
+}
+
+int fm_moe_forward(void* stream, const void* x, const void* gate_w,
+                   const void* expert_w, const void* b_up, const void* b_dn,
+                   void* gate_out, void* moe_out, int64_t S) {
+  if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
+  if (g.world != 1) {
+    setErr("fm_moe_forward is the single-rank path; use the staged EP entry points");
+    return FM_ERR_STATE;
+  }
+  if (S != g.S) { setErr("S mismatch vs frozen config"); return FM_ERR_SHAPE; }
+  hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+
+  FM_HIP_CHECK(hipMemsetAsync(g.eC, 0, (size_t)g.E * sizeof(uint32_t), st));
+  if (g.cfg.expert_top_k > 1) {
+    FM_HIP_CHECK(hipMemsetAsync(g.O32, 0, (size_t)g.S * g.H * sizeof(float), st));
+  }
+  // moe_out zero either way: dropped tokens keep zeros (clearState,
+  // moe.cuh:30-70 / python_bindings.cu:82)
+  FM_HIP_CHECK(hipMemsetAsync(moe_out, 0, (size_t)g.S * g.H * g.esz, st));
+
+  int rc = launch_gate(st, x, gate_w, gate_out, S);
+  if (rc != FM_OK) return rc;
+
+  const int mTiles = g.pEC / 128;
+  GemmArgs up{};
+  up.A = x;
+  up.B = expert_w;
+  up.bias = b_up;
+  up.out = g.xM;
+  up.gate_out = gate_out;
+  up.tokenIds = g.tokenIds;
+  up.eC = g.eC;
+  up.strideAExpert = 0;
+  up.strideBExpert = 2LL * g.P * g.H;
+  up.strideOExpert = (long long)g.pEC * g.P;
+  up.K = g.H; up.N = g.P;
+  up.EC = g.EC; up.pEC = g.pEC; up.PX = g.PX;
+  up.topk = g.cfg.expert_top_k; up.act = g.cfg.hidden_act;
+  up.expertOffset = 0; up.nRows = 0; up.H = g.H;
+  rc = launch_group_gemm(st, 0, up, mTiles, DIVUP(g.P, 128), g.E);
+  if (rc != FM_OK) return rc;
+
+  GemmArgs dn = up;
+  dn.A = g.xM;
+  dn.B = reinterpret_cast<const char*>(expert_w) + (size_t)g.P * g.H * g.esz;
+  dn.bias = b_dn;
+  dn.out = nullptr;
+  dn.O32 = g.O32;
+  dn.moe_out = moe_out;
+  dn.strideAExpert = (long long)g.pEC * g.P;
+  dn.K = g.P; dn.N = g.H;
+  rc = launch_group_gemm(st, 1, dn, mTiles, DIVUP(g.H, 128), g.E);
+  if (rc != FM_OK) return rc;
+
+  if (g.cfg.expert_top_k > 1) {
+    const size_t n = (size_t)g.S * g.H;
+    const int blocks = (int)min((size_t)2048, DIVUP(n, (size_t)256 * 8));
+    if (g.esz == 2)
+      hipLaunchKernelGGL(k_cast_out<bf16>, dim3(blocks), dim3(256), 0, st,
+                         g.O32, reinterpret_cast<bf16*>(moe_out), n);
+    else
+      hipLaunchKernelGGL(k_cast_out<float>, dim3(blocks), dim3(256), 0, st,
+                         g.O32, reinterpret_cast<float*>(moe_out), n);
+    FM_HIP_CHECK(hipGetLastError());
+  }
+  return FM_OK;
+}
+
+int fm_gate_forward(void* stream, const void* x, const void* gate_w,
+                    void* gate_out, int64_t S) {
+  if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
+  if (S != g.S) { setErr("S mismatch vs frozen config"); return FM_ERR_SHAPE; }
+  hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+  FM_HIP_CHECK(hipMemsetAsync(g.eC, 0, (size_t)g.E * sizeof(uint32_t), st));
+  return launch_gate(st, x, gate_w, gate_out, S);
+}
+
+int fm_read_routing(void* stream, uint32_t* routed_counts, uint32_t* token_idx,
+                    float* prob_sum) {
+  if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
+  hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+  uint32_t* hEC = (uint32_t*)malloc((size_t)g.E * sizeof(uint32_t));
+  TPS* hT = (TPS*)malloc((size_t)g.E * g.pEC * sizeof(TPS));
+  if (!hEC || !hT) { free(hEC); free(hT); setErr("oom"); return FM_ERR_HIP; }
+  hipError_t e1 = hipMemcpyAsync(hEC, g.eC, (size_t)g.E * sizeof(uint32_t),
+                                 hipMemcpyDeviceToHost, st);
+  hipError_t e2 = hipMemcpyAsync(hT, g.tokenIds,
+                                 (size_t)g.E * g.pEC * sizeof(TPS),
+                                 hipMemcpyDeviceToHost, st);
+  hipError_t e3 = hipStreamSynchronize(st);
+  if (e1 != hipSuccess || e2 != hipSuccess || e3 != hipSuccess) {
+    free(hEC); free(hT); setErr("routing D2H failed"); return FM_ERR_HIP;
+  }
+  for (int e = 0; e < g.E; ++e) {
+    const uint32_t r = hEC[e] < (uint32_t)g.EC ? hEC[e] : (uint32_t)g.EC;
+    routed_counts[e] = r;
+    for (uint32_t i = 0; i < r; ++i) {
+      token_idx[(size_t)e * g.EC + i] = hT[(size_t)e * g.pEC + i].tokenIdx;
+      prob_sum[(size_t)e * g.EC + i] = hT[(size_t)e * g.pEC + i].probSum;
+    }
+  }
+  free(hEC); free(hT);
+  return FM_OK;
+}
+
+int fm_expert_ffn(void* stream, const void* rows, const void* expert_w,
+                  const void* b_up, const void* b_dn, void* out_rows,
+                  int64_t n_rows, int32_t local_e) {
+  if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
+  if (n_rows <= 0) return FM_OK;
+  if (n_rows > (int64_t)g.nLxAlloc * g.pEC) {
+    setErr("n_rows exceeds workspace"); return FM_ERR_SHAPE;
+  }
+  hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+  const int mTiles128 = (int)DIVUP(n_rows, 128);
+  GemmArgs up{};
+  up.A = rows;
+  up.B = reinterpret_cast<const char*>(expert_w) +
+         (size_t)local_e * 2 * g.P * g.H * g.esz;
+  up.bias = b_up;
+  up.out = g.xM;  // scratch [n_rows, P]
+  up.tokenIds = nullptr; up.eC = nullptr;
+  up.strideAExpert = 0; up.strideBExpert = 0; up.strideOExpert = 0;
+  up.K = g.H; up.N = g.P; up.EC = 0; up.pEC = 0; up.PX = g.PX;
+  up.topk = 1; up.act = g.cfg.hidden_act; up.expertOffset = 0;
+  up.nRows = (int)n_rows; up.H = g.H;
+  int rc = launch_group_gemm(st, 0, up, mTiles128, DIVUP(g.P, 128), 1);
+  if (rc != FM_OK) return rc;
+  // note: PHASE 0 with tokenIds==nullptr writes out[e=0 stride 0] = xM rows
+  GemmArgs dn = up;
+  dn.A = g.xM;
+  dn.B = reinterpret_cast<const char*>(expert_w) +
+         ((size_t)local_e * 2 + 1) * g.P * g.H * g.esz;
+  dn.bias = b_dn;
+  dn.out = out_rows;
+  dn.K = g.P; dn.N = g.H;
+  return launch_group_gemm(st, 2, dn, mTiles128, DIVUP(g.H, 128), 1);
+}
+
+int fm_combine(void* stream, const void* rows, const uint32_t* token_idx,
+               const float* scale, int64_t n_rows, int32_t zero_first) {
+  if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
+  hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+  if (zero_first)
+    FM_HIP_CHECK(hipMemsetAsync(g.O32, 0, (size_t)g.S * g.H * sizeof(float), st));
+  if (n_rows <= 0) return FM_OK;
+  // always accumulate into the fp32 buffer; caller passes scale=1.0 rows
+  // for the k==1 unscaled semantics (each token appears at most once)
+  if (g.esz == 2) {
+    hipLaunchKernelGGL((k_combine_rows<bf16, true>), dim3((int)n_rows),
+                       dim3(256), 0, st, reinterpret_cast<const bf16*>(rows),
+                       token_idx, scale, g.O32, (bf16*)nullptr, g.H);
+  } else {
+    hipLaunchKernelGGL((k_combine_rows<float, true>), dim3((int)n_rows),
+                       dim3(256), 0, st, reinterpret_cast<const float*>(rows),
+                       token_idx, scale, g.O32, (float*)nullptr, g.H);
+  }
+  FM_HIP_CHECK(hipGetLastError());
+  return FM_OK;
+}
+
+int fm_combine_finalize(void* stream, void* moe_out, int64_t S) {
+  if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
+  if (S != g.S) { setErr("S mismatch"); return FM_ERR_SHAPE; }
+  hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+  const size_t n = (size_t)g.S * g.H;
+  const int blocks = (int)min((size_t)2048, DIVUP(n, (size_t)256 * 8));
+  if (g.esz == 2)
+    hipLaunchKernelGGL(k_cast_out<bf16>, dim3(blocks), dim3(256), 0, st, g.O32,
+                       reinterpret_cast<bf16*>(moe_out), n);
+  else
+    hipLaunchKernelGGL(k_cast_out<float>, dim3(blocks), dim3(256), 0, st,
+                       g.O32, reinterpret_cast<float*>(moe_out), n);
+  FM_HIP_CHECK(hipGetLastError());
+  return FM_OK;
+}
+
+// test-only probe (see k_mfma_probe)
+int fm_debug_mfma(void* stream, const void* A, const void* B, void* D) {
+  hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+  hipLaunchKernelGGL(k_mfma_probe, dim3(1), dim3(64), 0, st,
+                     reinterpret_cast<const bf16*>(A),
+                     reinterpret_cast<const bf16*>(B),
+                     reinterpret_cast<float*>(D));
+  FM_HIP_CHECK(hipGetLastError());
+  return FM_OK;
+}
+
+}  // extern "C"

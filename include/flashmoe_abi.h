@@ -1,0 +1,134 @@
+/* C-ABI boundary of the MI355X-native FlashDMoE hot path.
+ *
+ * Each entry point cites the reference interface it replaces
+ * (file:line into /root/reference). No torch types: plain pointers,
+ * sizes and an opaque HIP stream. All device pointers are raw HIP
+ * device memory on the current device; the caller owns input/output
+ * buffers, the library owns its internal workspace.
+ *
+ * Threading model: one process per GPU, all calls from
+ * one host thread per process, kernels enqueued on the caller's stream.
+ *
+ * The Python mirror (flashmoe_amd/moe.py, ctypes) reproduces the
+ * reference's pybind11 surface `flashmoe._C`
+ * (csrc/python_bindings.cu:194-217) on top of these functions; the
+ * binding stub a maintainer would add upstream is in INTEGRATION.md.
+ */
+#ifndef FLASHMOE_ABI_H
+#define FLASHMOE_ABI_H
+
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* Error codes (negative) / 0 on success. fm_last_error() returns a
+ * human-readable message for the most recent failure on this thread. */
+#define FM_OK 0
+#define FM_ERR_HIP (-1)          /* HIP runtime failure */
+#define FM_ERR_STATE (-2)        /* initialize/finalize ordering */
+#define FM_ERR_SHAPE (-3)        /* shape mismatch vs frozen config */
+#define FM_ERR_UNSUPPORTED (-4)  /* dtype/feature not built */
+
+/* Mirror of csrc/flashmoe_config.json (schema:
+ * csrc/flashmoe_config.schema.json). dtype: 0 fp32, 1 tf32(=fp32 on
+ * CDNA4 - no xf32), 2 bf16, 3 fp16. */
+typedef struct fm_config {
+  int32_t num_experts;
+  int32_t expert_top_k;
+  int32_t capacity_factor;
+  int32_t drop_tokens;
+  int32_t hidden_act; /* 0 relu, 1 gelu */
+  int32_t hidden_size;        /* H */
+  int32_t intermediate_size;  /* P */
+  int32_t sequence_len;
+  int32_t mini_batch;
+  int32_t dtype;
+  int32_t is_training;
+} fm_config;
+
+/* Replaces flashmoe::initialize (python_bindings.cu:157-159,
+ * bootstrap.cuh:533-547): freezes the config, selects the device,
+ * allocates the library workspace (tokenIds, eC, xM, O32) sized from the
+ * config. rank/world describe the EP layout (nLx = num_experts/world,
+ * uniform split, bootstrap.cuh:36-52); comms bootstrap itself lives on
+ * the Python side (torch.distributed over RCCL). */
+int fm_initialize(const fm_config* cfg, int rank, int world_size);
+
+/* Replaces flashmoe::finalize (python_bindings.cu:163-168,
+ * bootstrap.cuh:561-588): frees the workspace. */
+int fm_finalize(void);
+
+/* Replaces _C.get_compiled_config (python_bindings.cu:170-183 /
+ * flashmoe/ops.py:63-71): S, H, E, P, PX, element byte size. */
+int fm_get_compiled_config(int64_t* S, int64_t* H, int64_t* E, int64_t* P,
+                           int64_t* PX, int64_t* element_size);
+
+/* Replaces _C.get_num_local_experts (python_bindings.cu:185-189). */
+int fm_get_num_local_experts(void);
+
+/* THE hot path. Replaces _C.moe_forward (python_bindings.cu:17-151) for
+ * the single-rank case: gate -> route -> expert FFN -> combine on this
+ * rank's tokens against this rank's nLx experts (world 1: all E).
+ *
+ * stream:    hipStream_t as void* (0 = default stream)
+ * x:         [S, H] Element, device, contiguous
+ * gate_w:    flat E*H Element buffer (the torch [H,E] tensor's storage,
+ *            used as a row-major [E,H] matrix - reference quirk,
+ *            moe.cuh:107-109; see oracle/moe_oracle.py docstring)
+ * expert_w:  [nLx, 2, P, H] Element ([e][0]=Wup [P,H]; [e][1] flat,
+ *            used as [H,P] - moe.cuh:114-116)
+ * b_up/b_dn: [nLx, P] / [nLx, H] Element or NULL (reference zero-fills,
+ *            python_bindings.cu:80-82)
+ * gate_out:  [S, PX] Element out (softmax probs, moe.cuh:128-131)
+ * moe_out:   [S, H] Element out
+ * S:         token count this call (validated against frozen config)
+ *
+ * Asynchronous: enqueues on `stream`; no sync, no allocation. */
+int fm_moe_forward(void* stream, const void* x, const void* gate_w,
+                   const void* expert_w, const void* b_up, const void* b_dn,
+                   void* gate_out, void* moe_out, int64_t S);
+
+/* --- Staged entry points for the expert-parallel (multi-GPU) path.
+ * The reference runs these stages inside one kernel with one-sided
+ * NVSHMEM/P2P exchange between dispatch and FFN (moe.cuh:134-143); here
+ * the host pipeline calls them around the RCCL all-to-all. --- */
+
+/* Gate only: writes gate_out, and the library's tokenIds/eC workspace
+ * (gate.cuh:474-720 semantics). */
+int fm_gate_forward(void* stream, const void* x, const void* gate_w,
+                    void* gate_out, int64_t S);
+
+/* Copy routing results out: routed[e] = min(eC[e], EC) and the ordered
+ * token slots (token index + probSum) per expert, for the host to build
+ * the all-to-all. Device-to-host; synchronizes `stream`. */
+int fm_read_routing(void* stream, uint32_t* routed_counts /* [E] */,
+                    uint32_t* token_idx /* [E*EC] */,
+                    float* prob_sum /* [E*EC] */);
+
+/* Expert FFN on pre-packed rows (identity gather): rows [n_rows, H]
+ * belong to local expert `local_e`; writes z = (act(rows@WupT+b))@WdnT+b
+ * to out_rows [n_rows, H] (processor.cuh:685-751). */
+int fm_expert_ffn(void* stream, const void* rows, const void* expert_w,
+                  const void* b_up, const void* b_dn, void* out_rows,
+                  int64_t n_rows, int32_t local_e);
+
+/* Combine pre-scaled return rows into moe_out: for i < n_rows,
+ * moe_out[token_idx[i]] += scale[i] * rows[i] (k>1 path,
+ * processor.cuh:126-168); k==1: unscaled overwrite. Caller passes
+ * scale[i] = gate_out[t,e]/probSum[t]. zero_first resets the fp32
+ * accumulator before adding. finalize converts the accumulator into
+ * moe_out. */
+int fm_combine(void* stream, const void* rows, const uint32_t* token_idx,
+               const float* scale, int64_t n_rows, int32_t zero_first);
+int fm_combine_finalize(void* stream, void* moe_out, int64_t S);
+
+/* Version/introspection */
+const char* fm_last_error(void);
+int fm_built_for_gfx950(void);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* FLASHMOE_ABI_H */

@@ -1,0 +1,234 @@
+"""GPU parity tests: the HIP path vs the CPU oracle through the C-ABI.
+
+Bit-exact bar for routing (top-k indices, counts; near-tie escape below),
+rtol 2e-2 / atol 2e-3 for bf16 values, 1e-5 fp32.
+
+Determinism note (mirrors the reference, gate.cuh:688-716): inter-tile
+order in an expert's token list comes from atomicAdd, so when an expert
+OVERFLOWS capacity the kept-token SET is schedule-dependent — in the
+reference too. Exact-output tests therefore use either one 128-token tile
+(deterministic) or capacity that cannot overflow; full-size runs check
+invariants and a capacity-safe configuration.
+"""
+import ctypes
+import json
+import os
+import tempfile
+
+import numpy as np
+import pytest
+
+torch = pytest.importorskip("torch")
+
+from oracle.moe_oracle import OracleConfig, moe_forward as oracle_forward
+
+pytestmark = pytest.mark.gpu
+
+
+def make_cfg(**kw):
+    cfg = {
+        "capacity_factor": 1, "drop_tokens": 1, "expert_top_k": 2,
+        "global_batch": 256, "is_training": 0, "hidden_act": 0,
+        "hidden_size": 128, "intermediate_size": 256, "mini_batch": 1,
+        "moe_frequency": 1, "num_experts": 8, "num_layers": 1,
+        "sequence_len": 128, "torch_dtype": 2, "vocab_size": 32000,
+    }
+    cfg.update(kw)
+    f = tempfile.NamedTemporaryFile("w", suffix=".json", delete=False)
+    json.dump(cfg, f)
+    f.close()
+    return cfg, f.name
+
+
+@pytest.fixture
+def fresh_moe():
+    from flashmoe_amd import moe
+
+    yield moe
+    try:
+        moe.finalize()
+    except Exception:
+        pass
+
+
+def run_pair(moe, cfg, cfg_path, seed=47):
+    """Run HIP moe_forward and the oracle on identical inputs."""
+    from flashmoe_amd.config import torch_dtype_of
+
+    moe.initialize(cfg_path, rank=0, world_size=1)
+    S = cfg["sequence_len"] * cfg["mini_batch"]
+    H, P, E = cfg["hidden_size"], cfg["intermediate_size"], cfg["num_experts"]
+    dt = torch_dtype_of(cfg["torch_dtype"])
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    x = torch.randn(cfg["mini_batch"], cfg["sequence_len"], H, generator=g).to(dt).cuda()
+    gw = torch.randn(H, E, generator=g).to(dt).cuda()
+    ew = torch.randn(E, 2, P, H, generator=g).to(dt).cuda()
+    out = moe.moe_forward(x, gw, ew)
+    gate_out = moe.gate_output().clone()
+    torch.cuda.synchronize()
+
+    element = {0: "fp32", 1: "fp32", 2: "bf16", 3: "fp16"}[cfg["torch_dtype"]]
+    ocfg = OracleConfig(
+        num_experts=E, expert_top_k=cfg["expert_top_k"],
+        capacity_factor=cfg["capacity_factor"], drop_tokens=cfg["drop_tokens"],
+        hidden_act=cfg["hidden_act"], element=element,
+    )
+    ref = oracle_forward(
+        x.view(S, H).float().cpu().numpy(),
+        gw.float().cpu().numpy().reshape(-1),
+        ew.float().cpu().numpy(),
+        ocfg,
+    )
+    return out.view(S, H), gate_out, ref, ocfg
+
+
+def assert_values(got_t, want_np, element, what, rtol=None, atol_scale=None):
+    got = got_t.float().cpu().numpy()
+    scale = max(1.0, float(np.abs(want_np).max()))
+    if element == "bf16":
+        rtol = rtol or 2e-2
+        atol = (atol_scale or 2e-3) * scale
+    else:
+        rtol = rtol or 1e-5
+        atol = 1e-5 * scale
+    ok = np.isclose(got, want_np, rtol=rtol, atol=atol)
+    frac = 1.0 - ok.mean()
+    assert ok.all(), (
+        f"{what}: {100*frac:.3f}% mismatched, max abs err "
+        f"{np.abs(got-want_np).max():.5f} (scale {scale:.2f})"
+    )
+
+
+def routing_from_lib(E, EC):
+    """Read the kernel's routing via fm_read_routing."""
+    import flashmoe_amd._ext as _ext
+
+    lib = _ext.load()
+    counts = np.zeros(E, dtype=np.uint32)
+    tok = np.zeros(E * EC, dtype=np.uint32)
+    ps = np.zeros(E * EC, dtype=np.float32)
+    _ext.check(lib.fm_read_routing(
+        None,
+        ctypes.c_void_p(counts.ctypes.data),
+        ctypes.c_void_p(tok.ctypes.data),
+        ctypes.c_void_p(ps.ctypes.data)), "fm_read_routing")
+    return counts, tok.reshape(E, EC), ps.reshape(E, EC)
+
+
+# ---------------------------------------------------------------------------
+
+
+def test_mfma_layout_probe():
+    """Verify the assumed mfma_f32_16x16x32_bf16 operand/result layout
+    with an asymmetric product (guide par.3: always asymmetric B)."""
+    import __graft_entry__  # noqa: F401  (path setup)
+    import flashmoe_amd._ext as _ext
+
+    lib = _ext.load()
+    g = torch.Generator().manual_seed(0)
+    A = torch.randn(16, 32, generator=g).to(torch.bfloat16).cuda()
+    Bt = torch.randn(16, 32, generator=g).to(torch.bfloat16).cuda()  # [col][k]
+    D = torch.zeros(16, 16, dtype=torch.float32, device="cuda")
+    _ext.check(lib.fm_debug_mfma(
+        None, ctypes.c_void_p(A.data_ptr()), ctypes.c_void_p(Bt.data_ptr()),
+        ctypes.c_void_p(D.data_ptr())), "fm_debug_mfma")
+    torch.cuda.synchronize()
+    want = A.float() @ Bt.float().T
+    assert torch.allclose(D, want, rtol=1e-2, atol=1e-2), (
+        f"MFMA layout mismatch: max err {(D-want).abs().max().item()}"
+    )
+
+
+def test_single_tile_bf16_top2(fresh_moe):
+    cfg, path = make_cfg()
+    out, gate_out, ref, ocfg = run_pair(fresh_moe, cfg, path)
+    # routing bit-exact (single tile -> deterministic even with drop)
+    E, EC = 8, 32
+    counts, tok, ps = routing_from_lib(E, EC)
+    clipped = np.minimum(ref["eC"], EC)
+    assert np.array_equal(counts.astype(np.int64), clipped)
+    for e in range(E):
+        assert np.array_equal(tok[e, : counts[e]].astype(np.int64),
+                              ref["token_lists"][e]), f"expert {e} token list"
+    assert_values(gate_out, ref["gate_out"], "bf16", "gate_out")
+    assert_values(out, ref["moe_out"], "bf16", "moe_out")
+
+
+def test_single_tile_fp32_top1_config1(fresh_moe):
+    # BASELINE config 1: 4 experts top-1, seq=128, H=512, P=2048, fp32
+    cfg, path = make_cfg(num_experts=4, expert_top_k=1, hidden_size=512,
+                         intermediate_size=2048, torch_dtype=0)
+    out, gate_out, ref, ocfg = run_pair(fresh_moe, cfg, path)
+    assert_values(gate_out, ref["gate_out"], "fp32", "gate_out")
+    assert_values(out, ref["moe_out"], "fp32", "moe_out")
+
+
+def test_single_tile_gelu(fresh_moe):
+    cfg, path = make_cfg(hidden_act=1)
+    out, gate_out, ref, _ = run_pair(fresh_moe, cfg, path)
+    assert_values(out, ref["moe_out"], "bf16", "moe_out")
+
+
+def test_single_tile_top4(fresh_moe):
+    cfg, path = make_cfg(expert_top_k=4, num_experts=16)
+    out, gate_out, ref, _ = run_pair(fresh_moe, cfg, path)
+    assert_values(out, ref["moe_out"], "bf16", "moe_out")
+
+
+def test_multi_tile_no_overflow_bf16(fresh_moe):
+    """Multi-tile S with capacity_factor 2 (overflow-free): exact output
+    parity across the nondeterministic tile order."""
+    cfg, path = make_cfg(sequence_len=1024, capacity_factor=2,
+                         hidden_size=256, intermediate_size=512)
+    out, gate_out, ref, _ = run_pair(fresh_moe, cfg, path)
+    S, E = 1024, 8
+    EC = (S // E) * 2 * 2  # ceil(S/E)*CF*k = 512
+    counts, _, _ = routing_from_lib(E, EC)
+    assert counts.astype(np.int64).sum() == ref["eC"].sum() == S * 2
+    assert np.array_equal(np.sort(counts.astype(np.int64)), np.sort(ref["eC"]))
+    assert_values(gate_out, ref["gate_out"], "bf16", "gate_out")
+    assert_values(out, ref["moe_out"], "bf16", "moe_out")
+
+
+def test_bench_config_invariants(fresh_moe):
+    """BASELINE config 2 at full size (S=4096, H=1024, P=4096, E=8, k=2,
+    CF=1): size-independent properties + routing counts vs oracle
+    (kept SETS under overflow are schedule-dependent, as in the
+    reference — checked via counts, probs, and finiteness)."""
+    cfg, path = make_cfg(sequence_len=4096, hidden_size=1024,
+                         intermediate_size=4096)
+    out, gate_out, ref, _ = run_pair(fresh_moe, cfg, path)
+    S, E, EC = 4096, 8, 1024
+    counts, tok, ps = routing_from_lib(E, EC)
+    # raw eC is deterministic -> clipped counts match the oracle's
+    assert np.array_equal(counts.astype(np.int64), np.minimum(ref["eC"], EC))
+    # gate probabilities are schedule-independent
+    assert_values(gate_out, ref["gate_out"], "bf16", "gate_out")
+    g = gate_out.float().cpu().numpy()
+    np.testing.assert_allclose(g[:, :E].sum(1), 1.0, atol=2e-2)
+    assert np.isfinite(out.float().cpu().numpy()).all()
+    # each kernel-kept (token, expert) slot carries the oracle's mCw
+    mcw = ref["mCw"]
+    for e in range(E):
+        sl = slice(0, counts[e])
+        np.testing.assert_allclose(ps[e, sl], mcw[tok[e, sl]], rtol=2e-2)
+
+
+def test_error_behaviour(fresh_moe):
+    """Shape/device validation mirrors the reference's TORCH_CHECKs
+    (python_bindings.cu:22-70)."""
+    cfg, path = make_cfg()
+    fresh_moe.initialize(path, rank=0, world_size=1)
+    H, E, P = 128, 8, 256
+    x = torch.randn(1, 128, H, dtype=torch.bfloat16, device="cuda")
+    gw = torch.randn(H, E, dtype=torch.bfloat16, device="cuda")
+    ew = torch.randn(E, 2, P, H, dtype=torch.bfloat16, device="cuda")
+    with pytest.raises(ValueError, match="batch\\*seq"):
+        fresh_moe.moe_forward(x[:, :64], gw, ew)
+    with pytest.raises(ValueError, match="Gate weights"):
+        fresh_moe.moe_forward(x, gw.T.contiguous(), ew)
+    xx = torch.randn(1, 128, 2 * H, dtype=torch.bfloat16, device="cuda")[..., ::2]
+    with pytest.raises(ValueError, match="contiguous"):
+        fresh_moe.moe_forward(xx, gw, ew)
+    with pytest.raises(ValueError, match="Expert count"):
+        fresh_moe.moe_forward(x, gw, ew[:4])
