@@ -1,0 +1,241 @@
+#!/usr/bin/env python3
+"""FlashMoE-AMD benchmark — BASELINE.json metric on the BASELINE config.
+
+Workload (N=1): BASELINE.json configs[1] — 1xMI355X, 8 experts top-2,
+S=4096, H=1024, P=4096, bf16 (the largest single-GPU configuration; the
+quoted-metric case). A "step" is one DMoE forward over one batch of
+synthetic tokens already resident in HBM. At N>1 ranks: weak scaling the
+reference's way (experts grow with GPUs: E = 8N, nLx = 8 per rank,
+per-rank tokens fixed — README.md:46 / plots/scaling_gpus_8.png
+protocol), expert parallelism over RCCL.
+
+Timing: W untimed warmups, then exactly K steps bracketed by
+barrier + torch.cuda.synchronize on both sides; value = whole-job
+tokens/s = N*S*K / max-over-ranks elapsed. Inputs: torch.randn seed
+47+rank (reference protocol, flashmoe/worker.py:56-58,
+csrc/benchmarks/flash_bench.cu:40-41).
+
+Rank 0 prints ONE JSON line (driver contract).
+"""
+import argparse
+import ctypes
+import json
+import os
+import sys
+import tempfile
+import time
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+BF16_MFMA_PEAK_TFLOPS = 2500.0  # gfx950 dense bf16 MFMA peak (MI355X_MICROARCH.md)
+
+
+def build_config(n_gpus: int) -> dict:
+    return {
+        "capacity_factor": 1, "drop_tokens": 1, "expert_top_k": 2,
+        "global_batch": 256, "is_training": 0, "hidden_act": 0,
+        "hidden_size": 1024, "intermediate_size": 4096, "mini_batch": 1,
+        "moe_frequency": 1, "num_experts": 8 * n_gpus, "num_layers": 1,
+        "sequence_len": 4096, "torch_dtype": 2, "vocab_size": 32000,
+    }
+
+
+def measure_roofline(moe, lib, x, gw, ew, S, H, P, E, EC, iters=10):
+    """Per-phase HIP-event timing of the dominant kernels (expert GEMM
+    pair), outside the timed region. Algorithmic flops use the ACTUAL
+    routed token count (capacity drops included)."""
+    import numpy as np
+    import torch
+
+    stream = torch.cuda.current_stream().cuda_stream
+    gate_out = moe.gate_output()
+    out = torch.empty_like(x)
+    ms = (ctypes.c_float * 4)()
+    acc = [0.0, 0.0, 0.0, 0.0]
+    for _ in range(iters):
+        import flashmoe_amd._ext as _ext
+
+        _ext.check(lib.fm_moe_forward_phased(
+            ctypes.c_void_p(stream),
+            ctypes.c_void_p(x.data_ptr()), ctypes.c_void_p(gw.data_ptr()),
+            ctypes.c_void_p(ew.data_ptr()), None, None,
+            ctypes.c_void_p(gate_out.data_ptr()),
+            ctypes.c_void_p(out.data_ptr()), S, ms), "fm_moe_forward_phased")
+        for i in range(4):
+            acc[i] += ms[i] / iters
+    # routed rows for the algorithmic flop count
+    import flashmoe_amd._ext as _ext
+
+    counts = np.zeros(E, dtype=np.uint32)
+    tok = np.zeros(E * EC, dtype=np.uint32)
+    ps = np.zeros(E * EC, dtype=np.float32)
+    _ext.check(lib.fm_read_routing(
+        ctypes.c_void_p(stream),
+        ctypes.c_void_p(counts.ctypes.data), ctypes.c_void_p(tok.ctypes.data),
+        ctypes.c_void_p(ps.ctypes.data)), "fm_read_routing")
+    routed = int(counts.sum())
+    flops = 4.0 * routed * H * P  # up (2*r*P*H) + down (2*r*H*P)
+    gemm_ms = acc[1] + acc[2]
+    achieved_tf = flops / (gemm_ms * 1e-3) / 1e12 if gemm_ms > 0 else 0.0
+    return {
+        "bound": "mfma",
+        "achieved": round(achieved_tf, 1),
+        "peak": BF16_MFMA_PEAK_TFLOPS,
+        "unit": "TFLOP/s",
+        "frac": round(achieved_tf / BF16_MFMA_PEAK_TFLOPS, 4),
+        "traffic": None,
+        "detail": {
+            "phase_ms": {"gate": round(acc[0], 4), "gemm_up": round(acc[1], 4),
+                         "gemm_down_combine": round(acc[2], 4),
+                         "memset_cast": round(acc[3], 4)},
+            "routed_rows": routed,
+            "algorithmic_flops_per_fwd": flops,
+        },
+    }
+
+
+def measure_cpu_baseline(cfg, target_seconds=15.0):
+    """Oracle (kind 'port') on this box's host cores: a bounded sample of
+    the same workload, scaled to tokens/s."""
+    import numpy as np
+
+    from oracle.moe_oracle import OracleConfig, moe_forward
+
+    S_s = 512  # sample tokens (bounded: ~10-30 s of CPU work at cfg2 shapes)
+    H, P, E = cfg["hidden_size"], cfg["intermediate_size"], cfg["num_experts"]
+    ocfg = OracleConfig(num_experts=E, expert_top_k=cfg["expert_top_k"],
+                        capacity_factor=cfg["capacity_factor"],
+                        drop_tokens=cfg["drop_tokens"],
+                        hidden_act=cfg["hidden_act"], element="bf16")
+    g = np.random.default_rng(47)
+    x = g.standard_normal((S_s, H), dtype=np.float32)
+    gw = g.standard_normal((H * E,), dtype=np.float32)
+    ew = g.standard_normal((E, 2, P, H), dtype=np.float32)
+    moe_forward(x, gw, ew, ocfg)  # warm
+    t0 = time.perf_counter()
+    reps = 0
+    while time.perf_counter() - t0 < target_seconds and reps < 50:
+        moe_forward(x, gw, ew, ocfg)
+        reps += 1
+    dt = time.perf_counter() - t0
+    toks_per_s = S_s * reps / dt
+    return {
+        "value": round(toks_per_s, 1),
+        "unit": "tokens/s",
+        "cores": os.cpu_count(),
+        "kind": "port",
+        "sample": f"oracle moe_forward on {S_s} tokens x{reps} reps "
+                  f"({dt:.1f}s), same E={E}/H={H}/P={P}/top-2 shapes",
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=50)
+    ap.add_argument("--warmup", type=int, default=20)
+    ap.add_argument("--skip-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    import torch
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
+    n_gpus = max(args.gpus, world)
+    distributed = world > 1
+
+    cfg = build_config(n_gpus)
+    with tempfile.NamedTemporaryFile("w", suffix=".json", delete=False) as f:
+        json.dump(cfg, f)
+        cfg_path = f.name
+
+    from flashmoe_amd import moe
+    import flashmoe_amd._ext as _ext
+
+    moe.initialize(cfg_path, rank=rank, world_size=world)
+    lib = _ext.load()
+    if distributed:
+        import torch.distributed as dist
+
+        dist.init_process_group("nccl")
+
+    S = cfg["sequence_len"] * cfg["mini_batch"]
+    H, P, E = cfg["hidden_size"], cfg["intermediate_size"], cfg["num_experts"]
+    nLx = moe.get_num_local_experts()
+    EC = (-(-S // E)) * cfg["capacity_factor"] * cfg["expert_top_k"]
+
+    torch.manual_seed(47 + rank)
+    x = torch.randn(1, S, H, dtype=torch.bfloat16, device="cuda")
+    gw = torch.randn(H, E, dtype=torch.bfloat16, device="cuda")
+    ew = torch.randn(nLx, 2, P, H, dtype=torch.bfloat16, device="cuda")
+
+    def step():
+        return moe.moe_forward(x, gw, ew)
+
+    def barrier():
+        if distributed:
+            import torch.distributed as dist
+
+            dist.barrier()
+        torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        step()
+    barrier()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    barrier()
+    elapsed = time.perf_counter() - t0
+    if distributed:
+        import torch.distributed as dist
+
+        t = torch.tensor([elapsed], device="cuda")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1e3
+    tokens_per_s = n_gpus * S * args.steps / elapsed
+
+    if rank == 0:
+        roofline = None
+        cpu_baseline = None
+        if world == 1:
+            roofline = measure_roofline(moe, lib, x.view(S, H), gw, ew, S, H,
+                                        P, E, EC)
+            if not args.skip_cpu_baseline:
+                cpu_baseline = measure_cpu_baseline(cfg)
+        line = {
+            "metric": "DMoE fwd tokens/s",
+            "value": round(tokens_per_s, 1),
+            "unit": "tokens/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 4),
+            "us_per_fwd": round(ms_per_step * 1e3, 1),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # no published absolute numbers
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "workload": "BASELINE configs[1]: 1xMI355X E=8 top-2 S=4096 "
+                            "H=1024 P=4096 bf16 (E=8N weak scaling at N ranks)",
+                "seq_len": S, "d_model": H, "d_ff": P,
+                "experts": E, "top_k": 2, "capacity_factor": 1,
+                "parallelism": f"ep{n_gpus}",
+            },
+            "roofline": roofline,
+            "cpu_baseline": cpu_baseline,
+        }
+        print(json.dumps(line), flush=True)
+    if distributed:
+        import torch.distributed as dist
+
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

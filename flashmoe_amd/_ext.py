@@ -45,6 +45,9 @@ def _bind(lib: ctypes.CDLL) -> ctypes.CDLL:
     lib.fm_get_num_local_experts.restype = ctypes.c_int
     lib.fm_moe_forward.argtypes = [p, p, p, p, p, p, p, p, i64]
     lib.fm_moe_forward.restype = ctypes.c_int
+    lib.fm_moe_forward_phased.argtypes = [p, p, p, p, p, p, p, p, i64,
+                                          ctypes.POINTER(ctypes.c_float)]
+    lib.fm_moe_forward_phased.restype = ctypes.c_int
     lib.fm_gate_forward.argtypes = [p, p, p, p, i64]
     lib.fm_gate_forward.restype = ctypes.c_int
     lib.fm_read_routing.argtypes = [p, p, p, p]

@@ -687,17 +687,10 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
   return FM_OK;
 }
 
-int fm_moe_forward(void* stream, const void* x, const void* gate_w,
-                   const void* expert_w, const void* b_up, const void* b_dn,
-                   void* gate_out, void* moe_out, int64_t S) {
-  if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
-  if (g.world != 1) {
-    setErr("fm_moe_forward is the single-rank path; use the staged EP entry points");
-    return FM_ERR_STATE;
-  }
-  if (S != g.S) { setErr("S mismatch vs frozen config"); return FM_ERR_SHAPE; }
-  hipStream_t st = reinterpret_cast<hipStream_t>(stream);
-
+static int moe_forward_impl(hipStream_t st, const void* x, const void* gate_w,
+                            const void* expert_w, const void* b_up,
+                            const void* b_dn, void* gate_out, void* moe_out,
+                            int64_t S, hipEvent_t* evs) {
   FM_HIP_CHECK(hipMemsetAsync(g.eC, 0, (size_t)g.E * sizeof(uint32_t), st));
   if (g.cfg.expert_top_k > 1) {
     FM_HIP_CHECK(hipMemsetAsync(g.O32, 0, (size_t)g.S * g.H * sizeof(float), st));
@@ -705,9 +698,11 @@ int fm_moe_forward(void* stream, const void* x, const void* gate_w,
   // moe_out zero either way: dropped tokens keep zeros (clearState,
   // moe.cuh:30-70 / python_bindings.cu:82)
   FM_HIP_CHECK(hipMemsetAsync(moe_out, 0, (size_t)g.S * g.H * g.esz, st));
+  if (evs) FM_HIP_CHECK(hipEventRecord(evs[1], st));
 
   int rc = launch_gate(st, x, gate_w, gate_out, S);
   if (rc != FM_OK) return rc;
+  if (evs) FM_HIP_CHECK(hipEventRecord(evs[2], st));
 
   const int mTiles = g.pEC / 128;
   GemmArgs up{};
@@ -727,6 +722,7 @@ int fm_moe_forward(void* stream, const void* x, const void* gate_w,
   up.expertOffset = 0; up.nRows = 0; up.H = g.H;
   rc = launch_group_gemm(st, 0, up, mTiles, DIVUP(g.P, 128), g.E);
   if (rc != FM_OK) return rc;
+  if (evs) FM_HIP_CHECK(hipEventRecord(evs[3], st));
 
   GemmArgs dn = up;
   dn.A = g.xM;
@@ -739,6 +735,7 @@ int fm_moe_forward(void* stream, const void* x, const void* gate_w,
   dn.K = g.P; dn.N = g.H;
   rc = launch_group_gemm(st, 1, dn, mTiles, DIVUP(g.H, 128), g.E);
   if (rc != FM_OK) return rc;
+  if (evs) FM_HIP_CHECK(hipEventRecord(evs[4], st));
 
   if (g.cfg.expert_top_k > 1) {
     const size_t n = (size_t)g.S * g.H;
@@ -752,6 +749,47 @@ int fm_moe_forward(void* stream, const void* x, const void* gate_w,
     FM_HIP_CHECK(hipGetLastError());
   }
   return FM_OK;
+}
+
+int fm_moe_forward(void* stream, const void* x, const void* gate_w,
+                   const void* expert_w, const void* b_up, const void* b_dn,
+                   void* gate_out, void* moe_out, int64_t S) {
+  if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
+  if (g.world != 1) {
+    setErr("fm_moe_forward is the single-rank path; use the staged EP entry points");
+    return FM_ERR_STATE;
+  }
+  if (S != g.S) { setErr("S mismatch vs frozen config"); return FM_ERR_SHAPE; }
+  return moe_forward_impl(reinterpret_cast<hipStream_t>(stream), x, gate_w,
+                          expert_w, b_up, b_dn, gate_out, moe_out, S, nullptr);
+}
+
+int fm_moe_forward_phased(void* stream, const void* x, const void* gate_w,
+                          const void* expert_w, const void* b_up,
+                          const void* b_dn, void* gate_out, void* moe_out,
+                          int64_t S, float ms[4]) {
+  if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
+  if (g.world != 1) { setErr("single-rank path only"); return FM_ERR_STATE; }
+  if (S != g.S) { setErr("S mismatch vs frozen config"); return FM_ERR_SHAPE; }
+  hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+  hipEvent_t evs[6];
+  for (auto& e : evs) FM_HIP_CHECK(hipEventCreate(&e));
+  FM_HIP_CHECK(hipEventRecord(evs[0], st));
+  int rc = moe_forward_impl(st, x, gate_w, expert_w, b_up, b_dn, gate_out,
+                            moe_out, S, evs);
+  if (rc == FM_OK) {
+    FM_HIP_CHECK(hipEventRecord(evs[5], st));
+    FM_HIP_CHECK(hipStreamSynchronize(st));
+    float pre = 0, gate = 0, upT = 0, dnT = 0, post = 0;
+    hipEventElapsedTime(&pre, evs[0], evs[1]);
+    hipEventElapsedTime(&gate, evs[1], evs[2]);
+    hipEventElapsedTime(&upT, evs[2], evs[3]);
+    hipEventElapsedTime(&dnT, evs[3], evs[4]);
+    hipEventElapsedTime(&post, evs[4], evs[5]);
+    ms[0] = gate; ms[1] = upT; ms[2] = dnT; ms[3] = pre + post;
+  }
+  for (auto& e : evs) hipEventDestroy(e);
+  return rc;
 }
 
 int fm_gate_forward(void* stream, const void* x, const void* gate_w,

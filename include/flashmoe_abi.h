@@ -124,6 +124,17 @@ int fm_combine(void* stream, const void* rows, const uint32_t* token_idx,
                const float* scale, int64_t n_rows, int32_t zero_first);
 int fm_combine_finalize(void* stream, void* moe_out, int64_t S);
 
+/* Phase-timed variant of fm_moe_forward for roofline measurement
+ * (replaces the reference's in-API benchmark loop semantics,
+ * moe.cuh:146-184 forwardHostBench): HIP events are recorded between the
+ * phases on `stream`, the stream is synchronized, and ms[0..3] receives
+ * {gate, expert-up GEMM, expert-down GEMM+combine, other (memset/cast)}
+ * durations in milliseconds. Not for use inside a timed region. */
+int fm_moe_forward_phased(void* stream, const void* x, const void* gate_w,
+                          const void* expert_w, const void* b_up,
+                          const void* b_dn, void* gate_out, void* moe_out,
+                          int64_t S, float ms[4]);
+
 /* Version/introspection */
 const char* fm_last_error(void);
 int fm_built_for_gfx950(void);
