@@ -119,12 +119,35 @@ __global__ __launch_bounds__(256) void k_gate(
       }
     }
     __syncthreads();
+    // hoist this token's A chunk to registers as 16B vectors, then run
+    // every owned expert against it with vector LDS reads (scalar bf16
+    // LDS reads were the gate's dominant cost, profiles/r01)
     const T* arow = aCh + tok * LDA;
+    u32x4 a8[BK * sizeof(T) / 16];
+#pragma unroll
+    for (int jb = 0; jb < BK * (int)sizeof(T) / 16; ++jb)
+      a8[jb] = *reinterpret_cast<const u32x4*>(&arow[jb * (16 / sizeof(T))]);
     for (int e = e0; e < e1; ++e) {
       const T* grow = gCh + e * LDA;
       float s = 0.0f;
-#pragma unroll 8
-      for (int j = 0; j < BK; ++j) s = fmaf(toF(arow[j]), toF(grow[j]), s);
+#pragma unroll
+      for (int jb = 0; jb < BK * (int)sizeof(T) / 16; ++jb) {
+        const u32x4 g8 = *reinterpret_cast<const u32x4*>(&grow[jb * (16 / sizeof(T))]);
+#pragma unroll
+        for (int w = 0; w < 4; ++w) {
+          const uint32_t aw = a8[jb][w], gw = g8[w];
+          if constexpr (sizeof(T) == 2) {
+            const float2 av = __bfloat1622float2(
+                *reinterpret_cast<const __hip_bfloat162*>(&aw));
+            const float2 gv = __bfloat1622float2(
+                *reinterpret_cast<const __hip_bfloat162*>(&gw));
+            s = fmaf(av.x, gv.x, s);
+            s = fmaf(av.y, gv.y, s);
+          } else {
+            s = fmaf(__uint_as_float(aw), __uint_as_float(gw), s);
+          }
+        }
+      }
       logits[tok * (E + 1) + e] += s;
     }
     __syncthreads();
@@ -230,14 +253,23 @@ struct GemmArgs {
   int H;                // row stride of x / O32 / moe_out
 };
 
-template <int PHASE>
+// address-space helpers for global_load_lds (direct HBM->LDS DMA)
+typedef __attribute__((address_space(1))) const uint32_t gas_u32;
+typedef __attribute__((address_space(3))) uint32_t las_u32;
+
+template <int PHASE, int ACT, bool HAS_BIAS>
 __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
   constexpr int BM = 128, BN = 128, BK = 64;
-  constexpr int LDT = BK + 8;  // 16B row pad -> conflict-free b128 frag reads
-  __shared__ __attribute__((aligned(16))) bf16 Alds[BM * LDT];
-  __shared__ __attribute__((aligned(16))) bf16 Blds[BN * LDT];
-  __shared__ TPS sTps[BM];
-  __shared__ uint32_t sRouted;
+  // ONE shared arena (a second __shared__ object would force a vmcnt(0)
+  // drain before every ds_read beside glds - guide par.5 trap 4a).
+  // A/B tiles are LINEAR [128][64] bf16 (glds writes lane-linearly); the
+  // bank swizzle lives on the SOURCE chunk index and the fragment-read
+  // address (rule 21): chunk' = chunk ^ (row & 7).
+  __shared__ __attribute__((aligned(16))) char smem[2 * BM * BK * 2 + BM * 8 + 16];
+  bf16* Alds = reinterpret_cast<bf16*>(smem);
+  bf16* Blds = Alds + BM * BK;
+  TPS* sTps = reinterpret_cast<TPS*>(Blds + BN * BK);
+  uint32_t* sRouted = reinterpret_cast<uint32_t*>(sTps + BM);
 
   const int e = blockIdx.z;
   const int tid = threadIdx.x;
@@ -249,10 +281,10 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
 
   const TPS* tpsE = a.tokenIds ? a.tokenIds + (size_t)e * a.pEC : nullptr;
   if (tid == 0) {
-    sRouted = tpsE ? min(a.eC[e], (uint32_t)a.EC) : (uint32_t)a.nRows;
+    *sRouted = tpsE ? min(a.eC[e], (uint32_t)a.EC) : (uint32_t)a.nRows;
   }
   __syncthreads();
-  const uint32_t routed = sRouted;
+  const uint32_t routed = *sRouted;
   if ((uint32_t)m0 >= routed) return;  // empty tile (0-token expert / tail)
 
   if (tid < BM) {
@@ -276,45 +308,46 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
 #pragma unroll
     for (int j = 0; j < 4; ++j) accv[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  // staging map: 16B units; unit u -> row u/8, col8 u%8 (8 bf16 per unit)
-  // thread t stages units {t, t+256, t+512, t+768} -> coalesced
-  const int aRowStride = (PHASE == 0) ? a.H : K;  // x rows vs xM rows
+  // glds staging: 16 KiB per tile = 16 wave-instructions of 1 KiB; wave w
+  // issues groups w*4..w*4+3; lane l covers row grp*8 + l/8, chunk l%8.
+  // Source chunk is XOR-swizzled: sc = (l%8) ^ (l/8) (row&7 == l/8).
+  const int grow8 = lane >> 3;           // row within the 8-row group
+  const int schunk = (lane & 7) ^ grow8;  // swizzled 16B chunk index
+  const int aRowStride = (PHASE == 0) ? a.H : K;
+  const size_t aBase = (PHASE == 0) ? 0 : (size_t)e * a.strideAExpert;
   for (int kt = 0; kt < K; kt += BK) {
 #pragma unroll
-    for (int u4 = 0; u4 < 4; ++u4) {
-      const int u = tid + u4 * 256;
-      const int row = u >> 3, c8 = (u & 7) * 8;
-      // A row source: phase0 gathers x[token]; else xM row (m0+row)
-      size_t srow;
-      if (PHASE == 0) {
-        srow = (size_t)sTps[row].tokenIdx;
-      } else {
-        srow = (size_t)(m0 + row) + (size_t)e * 0;  // A already expert-based
-      }
-      const bf16* src = (PHASE == 0)
-          ? Ag + srow * aRowStride + kt + c8
-          : Ag + (size_t)e * a.strideAExpert + (size_t)(m0 + row) * K + kt + c8;
-      *reinterpret_cast<u32x4*>(&Alds[row * LDT + c8]) =
-          *reinterpret_cast<const u32x4*>(src);
+    for (int i = 0; i < 4; ++i) {
+      const int grp = wave * 4 + i;
+      const int row = grp * 8 + grow8;
+      const size_t arow = (PHASE == 0) ? (size_t)sTps[row].tokenIdx
+                                       : (size_t)(m0 + row);
+      const bf16* asrc = Ag + aBase + arow * aRowStride + kt + schunk * 8;
+      __builtin_amdgcn_global_load_lds(
+          (gas_u32*)asrc, (las_u32*)(Alds + grp * 512), 16, 0, 0);
       const int brow = min(n0 + row, N - 1);
-      const bf16* bsrc = Bg + (size_t)brow * K + kt + c8;
-      *reinterpret_cast<u32x4*>(&Blds[row * LDT + c8]) =
-          *reinterpret_cast<const u32x4*>(bsrc);
+      const bf16* bsrc = Bg + (size_t)brow * K + kt + schunk * 8;
+      __builtin_amdgcn_global_load_lds(
+          (gas_u32*)bsrc, (las_u32*)(Blds + grp * 512), 16, 0, 0);
     }
-    __syncthreads();
+    __syncthreads();  // carries the vmcnt(0) glds drain (guide par.5)
 #pragma unroll
     for (int s = 0; s < 2; ++s) {  // two K=32 MFMA steps per tile
       bf16x8 af[4], bf[4];
-      const int kofs = 32 * s + (lane >> 4) * 8;
       const int rl = lane & 15;
+      const int cbase = 4 * s + (lane >> 4);  // 16B chunk before swizzle
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
+      for (int mi = 0; mi < 4; ++mi) {
+        const int R = wr * 64 + mi * 16 + rl;
         af[mi] = *reinterpret_cast<const bf16x8*>(
-            &Alds[(wr * 64 + mi * 16 + rl) * LDT + kofs]);
+            &Alds[R * BK + ((cbase ^ (R & 7)) * 8)]);
+      }
 #pragma unroll
-      for (int ni = 0; ni < 4; ++ni)
+      for (int ni = 0; ni < 4; ++ni) {
+        const int R = wc * 64 + ni * 16 + rl;
         bf[ni] = *reinterpret_cast<const bf16x8*>(
-            &Blds[(wc * 64 + ni * 16 + rl) * LDT + kofs]);
+            &Blds[R * BK + ((cbase ^ (R & 7)) * 8)]);
+      }
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
@@ -325,9 +358,37 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
     __syncthreads();
   }
 
-  // epilogue: C/D map for 16x16 MFMA: col = lane&15, row = (lane>>4)*4 + r
+  // epilogue: C/D map for 16x16 MFMA: col = lane&15, row = (lane>>4)*4 + r.
+  // Bias and combine scales are HOISTED (a per-element bias load made
+  // hipcc emit one dependent global_load+vmcnt(0) per output - guide
+  // par.5 trap 4c, profiles/r01); activation is a compile-time template.
   const int cl = lane & 15;
   const int r0 = (lane >> 4) * 4;
+  float* sScale = reinterpret_cast<float*>(Alds);  // reuse; loop is done
+  const bool multi = (PHASE == 1) && a.topk > 1;
+  if constexpr (PHASE == 1) {
+    if (multi) {
+      __syncthreads();
+      if (tid < BM) {
+        const TPS tp = sTps[tid];
+        float sc = 0.0f;
+        if ((uint32_t)(m0 + tid) < routed)
+          sc = toF(reinterpret_cast<const bf16*>(
+                   a.gate_out)[(size_t)tp.tokenIdx * a.PX +
+                               a.expertOffset + e]) / tp.probSum;
+        sScale[tid] = sc;
+      }
+      __syncthreads();
+    }
+  }
+  float bv[4] = {0.f, 0.f, 0.f, 0.f};
+  if constexpr (HAS_BIAS) {
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      const int col = n0 + wc * 64 + ni * 16 + cl;
+      if (col < N) bv[ni] = toF(reinterpret_cast<const bf16*>(a.bias)[col]);
+    }
+  }
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi) {
 #pragma unroll
@@ -336,24 +397,21 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
       const int m = m0 + row;
       if ((uint32_t)m >= routed) continue;
       const TPS tp = sTps[row];
+      const float rowScale = multi ? sScale[row] : 1.0f;
 #pragma unroll
       for (int ni = 0; ni < 4; ++ni) {
         const int col = n0 + wc * 64 + ni * 16 + cl;
         if (col >= N) continue;
-        float v = accv[mi][ni][r];
-        if (a.bias)
-          v += toF(reinterpret_cast<const bf16*>(a.bias)[col]);
+        float v = accv[mi][ni][r] + bv[ni];
         if constexpr (PHASE == 0) {
-          v = applyAct(v, a.act);
+          v = (ACT == 0) ? fmaxf(v, 0.0f)
+                         : 0.5f * v * (1.0f + erff(v * 0.70710678118654752f));
           reinterpret_cast<bf16*>(a.out)[(size_t)e * a.strideOExpert +
                                          (size_t)m * N + col] =
               __float2bfloat16(v);
         } else if constexpr (PHASE == 1) {
-          if (a.topk > 1) {
-            const float prob = toF(reinterpret_cast<const bf16*>(
-                a.gate_out)[(size_t)tp.tokenIdx * a.PX + a.expertOffset + e]);
-            atomicAdd(&a.O32[(size_t)tp.tokenIdx * a.H + col],
-                      v * prob / tp.probSum);
+          if (multi) {
+            atomicAdd(&a.O32[(size_t)tp.tokenIdx * a.H + col], v * rowScale);
           } else {
             reinterpret_cast<bf16*>(
                 a.moe_out)[(size_t)tp.tokenIdx * a.H + col] =
@@ -374,7 +432,7 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
 // Same phase semantics as the bf16 kernel.
 // ---------------------------------------------------------------------------
 
-template <int PHASE>
+template <int PHASE, int ACT, bool HAS_BIAS>
 __global__ __launch_bounds__(256) void k_group_gemm_f32(GemmArgs a) {
   constexpr int BM = 64, BN = 64, BK = 16;
   constexpr int LDT = BK + 1;
@@ -438,27 +496,38 @@ __global__ __launch_bounds__(256) void k_group_gemm_f32(GemmArgs a) {
     __syncthreads();
   }
 
+  const bool multi = (PHASE == 1) && a.topk > 1;
+  float bv[4] = {0.f, 0.f, 0.f, 0.f};
+  if constexpr (HAS_BIAS) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int col = n0 + tc + j;
+      if (col < N) bv[j] = reinterpret_cast<const float*>(a.bias)[col];
+    }
+  }
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
     const int m = m0 + tr + i;
     if ((uint32_t)m >= routed) continue;
     const TPS tp = sTps[tr + i];
+    float rowScale = 1.0f;
+    if (multi)
+      rowScale = reinterpret_cast<const float*>(
+                     a.gate_out)[(size_t)tp.tokenIdx * a.PX +
+                                 a.expertOffset + e] / tp.probSum;
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
       const int col = n0 + tc + j;
       if (col >= N) continue;
-      float v = acc[i][j];
-      if (a.bias) v += reinterpret_cast<const float*>(a.bias)[col];
+      float v = acc[i][j] + bv[j];
       if constexpr (PHASE == 0) {
-        v = applyAct(v, a.act);
+        v = (ACT == 0) ? fmaxf(v, 0.0f)
+                       : 0.5f * v * (1.0f + erff(v * 0.70710678118654752f));
         reinterpret_cast<float*>(a.out)[(size_t)e * a.strideOExpert +
                                         (size_t)m * N + col] = v;
       } else if constexpr (PHASE == 1) {
-        if (a.topk > 1) {
-          const float prob = reinterpret_cast<const float*>(
-              a.gate_out)[(size_t)tp.tokenIdx * a.PX + a.expertOffset + e];
-          atomicAdd(&a.O32[(size_t)tp.tokenIdx * a.H + col],
-                    v * prob / tp.probSum);
+        if (multi) {
+          atomicAdd(&a.O32[(size_t)tp.tokenIdx * a.H + col], v * rowScale);
         } else {
           reinterpret_cast<float*>(a.moe_out)[(size_t)tp.tokenIdx * a.H +
                                               col] = v;
@@ -668,21 +737,35 @@ int fm_get_num_local_experts(void) { return g.initialized ? g.nLx : -1; }
 static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
                              int mTiles, int nTiles, int nE) {
   dim3 block(256);
+  const bool hasBias = a.bias != nullptr;
+  const int act = (phase == 0) ? a.act : 0;
+#define GG_LAUNCH(KER, GRID, PH, AC, HB)                                      \
+  hipLaunchKernelGGL((KER<PH, AC, HB>), GRID, block, 0, st, a)
+#define GG_DISPATCH(KER, GRID)                                                \
+  do {                                                                        \
+    const int sel = phase * 4 + act * 2 + (hasBias ? 1 : 0);                  \
+    switch (sel) {                                                            \
+      case 0: GG_LAUNCH(KER, GRID, 0, 0, false); break;                       \
+      case 1: GG_LAUNCH(KER, GRID, 0, 0, true); break;                        \
+      case 2: GG_LAUNCH(KER, GRID, 0, 1, false); break;                       \
+      case 3: GG_LAUNCH(KER, GRID, 0, 1, true); break;                        \
+      case 4: GG_LAUNCH(KER, GRID, 1, 0, false); break;                       \
+      case 5: GG_LAUNCH(KER, GRID, 1, 0, true); break;                        \
+      default:                                                                \
+        if (hasBias) GG_LAUNCH(KER, GRID, 2, 0, true);                        \
+        else GG_LAUNCH(KER, GRID, 2, 0, false);                               \
+        break;                                                                \
+    }                                                                         \
+  } while (0)
   if (g.esz == 2) {
     dim3 grid(mTiles, nTiles, nE);
-    switch (phase) {
-      case 0: hipLaunchKernelGGL(k_group_gemm_bf16<0>, grid, block, 0, st, a); break;
-      case 1: hipLaunchKernelGGL(k_group_gemm_bf16<1>, grid, block, 0, st, a); break;
-      default: hipLaunchKernelGGL(k_group_gemm_bf16<2>, grid, block, 0, st, a); break;
-    }
+    GG_DISPATCH(k_group_gemm_bf16, grid);
   } else {
     dim3 grid(mTiles * 2, nTiles * 2, nE);  // 64x64 tiles
-    switch (phase) {
-      case 0: hipLaunchKernelGGL(k_group_gemm_f32<0>, grid, block, 0, st, a); break;
-      case 1: hipLaunchKernelGGL(k_group_gemm_f32<1>, grid, block, 0, st, a); break;
-      default: hipLaunchKernelGGL(k_group_gemm_f32<2>, grid, block, 0, st, a); break;
-    }
+    GG_DISPATCH(k_group_gemm_f32, grid);
   }
+#undef GG_DISPATCH
+#undef GG_LAUNCH
   FM_HIP_CHECK(hipGetLastError());
   return FM_OK;
 }
@@ -695,9 +778,11 @@ static int moe_forward_impl(hipStream_t st, const void* x, const void* gate_w,
   if (g.cfg.expert_top_k > 1) {
     FM_HIP_CHECK(hipMemsetAsync(g.O32, 0, (size_t)g.S * g.H * sizeof(float), st));
   }
-  // moe_out zero either way: dropped tokens keep zeros (clearState,
-  // moe.cuh:30-70 / python_bindings.cu:82)
-  FM_HIP_CHECK(hipMemsetAsync(moe_out, 0, (size_t)g.S * g.H * g.esz, st));
+  // k==1: zero moe_out so dropped tokens keep zeros (clearState,
+  // moe.cuh:30-70); k>1 writes every element via k_cast_out instead
+  if (g.cfg.expert_top_k == 1) {
+    FM_HIP_CHECK(hipMemsetAsync(moe_out, 0, (size_t)g.S * g.H * g.esz, st));
+  }
   if (evs) FM_HIP_CHECK(hipEventRecord(evs[1], st));
 
   int rc = launch_gate(st, x, gate_w, gate_out, S);
