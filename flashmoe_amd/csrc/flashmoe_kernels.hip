@@ -427,6 +427,209 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
 }
 
 // ---------------------------------------------------------------------------
+// Big-tile grouped GEMM, bf16 MFMA: 256xBN (BN 256 or 128), BK=64,
+// 512 threads = 8 waves (2M x 4N), per-wave 128 x BN/4 as 8 x BN/64
+// mfma_f32_16x16x32_bf16 fragments. Double-buffered glds staging with
+// COUNTED vmcnt + raw s_barrier (the guide's deep-pipeline recipe: a
+// __syncthreads here would drain vmcnt(0) and expose full HBM latency
+// every K-tile - measured 47% wave-park, profiles/r01 PMC). Used for the
+// large-M expert shapes; the 128^2 kernel remains for small tiles.
+// Same phase semantics/epilogues as k_group_gemm_bf16.
+// ---------------------------------------------------------------------------
+
+template <int PHASE, int ACT, bool HAS_BIAS, int BN>
+__global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
+  constexpr int BM = 256, BK = 64;
+  constexpr int NF = BN / 64;           // B fragments per wave (4 or 2)
+  constexpr int AGRP = BM * BK * 2 / 1024;  // glds 1KiB groups per A tile (32)
+  constexpr int BGRP = BN * BK * 2 / 1024;  // (16 or 8)
+  constexpr int GPW_A = AGRP / 8, GPW_B = BGRP / 8;  // per wave
+  constexpr int GPT = GPW_A + GPW_B;    // glds per wave per K-tile
+  __shared__ __attribute__((aligned(16))) char smem[
+      2 * BM * BK * 2 + 2 * BN * BK * 2 + BM * 8 + 16];
+  bf16* Abase = reinterpret_cast<bf16*>(smem);          // 2 x [BM][BK]
+  bf16* Bbase = Abase + 2 * BM * BK;                    // 2 x [BN][BK]
+  TPS* sTps = reinterpret_cast<TPS*>(Bbase + 2 * BN * BK);
+  uint32_t* sRouted = reinterpret_cast<uint32_t*>(sTps + BM);
+
+  const int e = blockIdx.z;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int m0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+  const int K = a.K, N = a.N;
+
+  const TPS* tpsE = a.tokenIds ? a.tokenIds + (size_t)e * a.pEC : nullptr;
+  if (tid == 0)
+    *sRouted = tpsE ? min(a.eC[e], (uint32_t)a.EC) : (uint32_t)a.nRows;
+  __syncthreads();
+  const uint32_t routed = *sRouted;
+  if ((uint32_t)m0 >= routed) return;
+  const int mCap = a.tokenIds ? a.pEC : a.nRows;  // A-row clamp bound
+  if (tid < BM) {
+    TPS t{0u, 1.0f};
+    if ((uint32_t)(m0 + tid) < routed)
+      t = tpsE ? tpsE[m0 + tid] : TPS{(uint32_t)(m0 + tid), 1.0f};
+    sTps[tid] = t;
+  }
+  __syncthreads();
+
+  const bf16* __restrict__ Ag = reinterpret_cast<const bf16*>(a.A);
+  const bf16* __restrict__ Bg =
+      reinterpret_cast<const bf16*>(a.B) + (size_t)e * a.strideBExpert;
+
+  // per-lane glds source bases, hoisted out of the K loop
+  const int grow8 = lane >> 3;
+  const int schunk = (lane & 7) ^ grow8;
+  const int aRowStride = (PHASE == 0) ? a.H : K;
+  const size_t aBase = (PHASE == 0) ? 0 : (size_t)e * a.strideAExpert;
+  const bf16* aSrc[GPW_A];
+  const bf16* bSrc[GPW_B];
+#pragma unroll
+  for (int i = 0; i < GPW_A; ++i) {
+    const int row = (wave * GPW_A + i) * 8 + grow8;
+    const size_t arow = (PHASE == 0) ? (size_t)sTps[row].tokenIdx
+                                     : (size_t)min(m0 + row, mCap - 1);
+    aSrc[i] = Ag + aBase + arow * aRowStride + schunk * 8;
+  }
+#pragma unroll
+  for (int i = 0; i < GPW_B; ++i) {
+    const int row = (wave * GPW_B + i) * 8 + grow8;
+    bSrc[i] = Bg + (size_t)min(n0 + row, N - 1) * K + schunk * 8;
+  }
+
+  auto stage = [&](int kt, int buf) {
+#pragma unroll
+    for (int i = 0; i < GPW_A; ++i)
+      __builtin_amdgcn_global_load_lds(
+          (gas_u32*)(aSrc[i] + kt),
+          (las_u32*)(Abase + buf * BM * BK + (wave * GPW_A + i) * 512), 16, 0, 0);
+#pragma unroll
+    for (int i = 0; i < GPW_B; ++i)
+      __builtin_amdgcn_global_load_lds(
+          (gas_u32*)(bSrc[i] + kt),
+          (las_u32*)(Bbase + buf * BN * BK + (wave * GPW_B + i) * 512), 16, 0, 0);
+  };
+
+  const int wr = wave >> 2, wc = wave & 3;  // 2M x 4N wave grid
+  f32x4 accv[8][NF];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < NF; ++j) accv[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int nK = K / BK;
+  stage(0, 0);
+  if (nK > 1) stage(BK, 1);
+  for (int t = 0; t < nK; ++t) {
+    // wait THIS tile's glds (t+1's stay in flight), then rendezvous
+    if (t + 1 < nK) {
+      asm volatile("s_waitcnt vmcnt(%0)" ::"i"(GPT) : "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+    const bf16* Al = Abase + (t & 1) * BM * BK;
+    const bf16* Bl = Bbase + (t & 1) * BN * BK;
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int s = 0; s < 2; ++s) {
+      bf16x8 af[8], bfr[NF];
+      const int rl = lane & 15;
+      const int cbase = 4 * s + (lane >> 4);
+#pragma unroll
+      for (int mi = 0; mi < 8; ++mi) {
+        const int R = wr * 128 + mi * 16 + rl;
+        af[mi] = *reinterpret_cast<const bf16x8*>(
+            &Al[R * BK + ((cbase ^ (R & 7)) * 8)]);
+      }
+#pragma unroll
+      for (int ni = 0; ni < NF; ++ni) {
+        const int R = wc * (BN / 4) + ni * 16 + rl;
+        bfr[ni] = *reinterpret_cast<const bf16x8*>(
+            &Bl[R * BK + ((cbase ^ (R & 7)) * 8)]);
+      }
+#pragma unroll
+      for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < NF; ++ni)
+          accv[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bfr[ni], accv[mi][ni], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_s_barrier();  // everyone done reading buf (t&1)
+    if (t + 2 < nK) stage((t + 2) * BK, t & 1);
+  }
+
+  // epilogue (same semantics as k_group_gemm_bf16)
+  const int cl = lane & 15;
+  const int r0 = (lane >> 4) * 4;
+  float* sScale = reinterpret_cast<float*>(Abase);
+  const bool multi = (PHASE == 1) && a.topk > 1;
+  if constexpr (PHASE == 1) {
+    if (multi) {
+      __syncthreads();
+      if (tid < BM) {
+        const TPS tp = sTps[tid];
+        float sc = 0.0f;
+        if ((uint32_t)(m0 + tid) < routed)
+          sc = toF(reinterpret_cast<const bf16*>(
+                   a.gate_out)[(size_t)tp.tokenIdx * a.PX +
+                               a.expertOffset + e]) / tp.probSum;
+        sScale[tid] = sc;
+      }
+      __syncthreads();
+    }
+  }
+  float bv[NF];
+#pragma unroll
+  for (int ni = 0; ni < NF; ++ni) bv[ni] = 0.f;
+  if constexpr (HAS_BIAS) {
+#pragma unroll
+    for (int ni = 0; ni < NF; ++ni) {
+      const int col = n0 + wc * (BN / 4) + ni * 16 + cl;
+      if (col < N) bv[ni] = toF(reinterpret_cast<const bf16*>(a.bias)[col]);
+    }
+  }
+#pragma unroll
+  for (int mi = 0; mi < 8; ++mi) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = wr * 128 + mi * 16 + r0 + r;
+      const int m = m0 + row;
+      if ((uint32_t)m >= routed) continue;
+      const TPS tp = sTps[row];
+      const float rowScale = multi ? sScale[row] : 1.0f;
+#pragma unroll
+      for (int ni = 0; ni < NF; ++ni) {
+        const int col = n0 + wc * (BN / 4) + ni * 16 + cl;
+        if (col >= N) continue;
+        float v = accv[mi][ni][r] + bv[ni];
+        if constexpr (PHASE == 0) {
+          v = (ACT == 0) ? fmaxf(v, 0.0f)
+                         : 0.5f * v * (1.0f + erff(v * 0.70710678118654752f));
+          reinterpret_cast<bf16*>(a.out)[(size_t)e * a.strideOExpert +
+                                         (size_t)m * N + col] =
+              __float2bfloat16(v);
+        } else if constexpr (PHASE == 1) {
+          if (multi) {
+            atomicAdd(&a.O32[(size_t)tp.tokenIdx * a.H + col], v * rowScale);
+          } else {
+            reinterpret_cast<bf16*>(
+                a.moe_out)[(size_t)tp.tokenIdx * a.H + col] =
+                __float2bfloat16(v);
+          }
+        } else {
+          reinterpret_cast<bf16*>(a.out)[(size_t)m * N + col] =
+              __float2bfloat16(v);
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // fp32 grouped GEMM (config-1 "CPU correctness plumbing" shapes; VALU
 // tiled). 64x64 tile, 256 threads, each thread a 4x4 sub-block, BK=16.
 // Same phase semantics as the bf16 kernel.
@@ -735,37 +938,65 @@ int fm_get_compiled_config(int64_t* S, int64_t* H, int64_t* E, int64_t* P,
 int fm_get_num_local_experts(void) { return g.initialized ? g.nLx : -1; }
 
 static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
-                             int mTiles, int nTiles, int nE) {
-  dim3 block(256);
+                             int M, int nE) {
   const bool hasBias = a.bias != nullptr;
   const int act = (phase == 0) ? a.act : 0;
-#define GG_LAUNCH(KER, GRID, PH, AC, HB)                                      \
-  hipLaunchKernelGGL((KER<PH, AC, HB>), GRID, block, 0, st, a)
-#define GG_DISPATCH(KER, GRID)                                                \
-  do {                                                                        \
-    const int sel = phase * 4 + act * 2 + (hasBias ? 1 : 0);                  \
-    switch (sel) {                                                            \
-      case 0: GG_LAUNCH(KER, GRID, 0, 0, false); break;                       \
-      case 1: GG_LAUNCH(KER, GRID, 0, 0, true); break;                        \
-      case 2: GG_LAUNCH(KER, GRID, 0, 1, false); break;                       \
-      case 3: GG_LAUNCH(KER, GRID, 0, 1, true); break;                        \
-      case 4: GG_LAUNCH(KER, GRID, 1, 0, false); break;                       \
-      case 5: GG_LAUNCH(KER, GRID, 1, 0, true); break;                        \
-      default:                                                                \
-        if (hasBias) GG_LAUNCH(KER, GRID, 2, 0, true);                        \
-        else GG_LAUNCH(KER, GRID, 2, 0, false);                               \
-        break;                                                                \
-    }                                                                         \
-  } while (0)
+  const int N = a.N;
   if (g.esz == 2) {
-    dim3 grid(mTiles, nTiles, nE);
-    GG_DISPATCH(k_group_gemm_bf16, grid);
+    // tile selection: prefer the deep-pipelined 256-row kernel when the
+    // grid still fills the 256 CUs at 1 block/CU
+    const int b256 = DIVUP(M, 256) * DIVUP(N, 256) * nE;
+    const int b128n = DIVUP(M, 256) * DIVUP(N, 128) * nE;
+    int mode;  // 0: big BN=256, 1: big BN=128, 2: small 128x128
+    if (M >= 256 && b256 >= 256) mode = 0;
+    else if (M >= 256 && b128n >= 256) mode = 1;
+    else mode = 2;
+    dim3 block(mode == 2 ? 256 : 512);
+    dim3 grid(DIVUP(M, mode == 2 ? 128 : 256),
+              DIVUP(N, mode == 1 ? 128 : (mode == 0 ? 256 : 128)), nE);
+    const int sel = phase * 4 + act * 2 + (hasBias ? 1 : 0);
+#define GG_CASE(PH, AC, HB)                                                   \
+    do {                                                                      \
+      if (mode == 0)                                                          \
+        hipLaunchKernelGGL((k_group_gemm_bf16_big<PH, AC, HB, 256>), grid,    \
+                           block, 0, st, a);                                  \
+      else if (mode == 1)                                                     \
+        hipLaunchKernelGGL((k_group_gemm_bf16_big<PH, AC, HB, 128>), grid,    \
+                           block, 0, st, a);                                  \
+      else                                                                    \
+        hipLaunchKernelGGL((k_group_gemm_bf16<PH, AC, HB>), grid, block, 0,   \
+                           st, a);                                            \
+    } while (0)
+    switch (sel) {
+      case 0: GG_CASE(0, 0, false); break;
+      case 1: GG_CASE(0, 0, true); break;
+      case 2: GG_CASE(0, 1, false); break;
+      case 3: GG_CASE(0, 1, true); break;
+      case 4: GG_CASE(1, 0, false); break;
+      case 5: GG_CASE(1, 0, true); break;
+      default:
+        if (hasBias) GG_CASE(2, 0, true);
+        else GG_CASE(2, 0, false);
+        break;
+    }
+#undef GG_CASE
   } else {
-    dim3 grid(mTiles * 2, nTiles * 2, nE);  // 64x64 tiles
-    GG_DISPATCH(k_group_gemm_f32, grid);
+    dim3 block(256);
+    dim3 grid(DIVUP(M, 64), DIVUP(N, 64), nE);
+    const int sel = phase * 4 + act * 2 + (hasBias ? 1 : 0);
+    switch (sel) {
+      case 0: hipLaunchKernelGGL((k_group_gemm_f32<0, 0, false>), grid, block, 0, st, a); break;
+      case 1: hipLaunchKernelGGL((k_group_gemm_f32<0, 0, true>), grid, block, 0, st, a); break;
+      case 2: hipLaunchKernelGGL((k_group_gemm_f32<0, 1, false>), grid, block, 0, st, a); break;
+      case 3: hipLaunchKernelGGL((k_group_gemm_f32<0, 1, true>), grid, block, 0, st, a); break;
+      case 4: hipLaunchKernelGGL((k_group_gemm_f32<1, 0, false>), grid, block, 0, st, a); break;
+      case 5: hipLaunchKernelGGL((k_group_gemm_f32<1, 0, true>), grid, block, 0, st, a); break;
+      default:
+        if (hasBias) hipLaunchKernelGGL((k_group_gemm_f32<2, 0, true>), grid, block, 0, st, a);
+        else hipLaunchKernelGGL((k_group_gemm_f32<2, 0, false>), grid, block, 0, st, a);
+        break;
+    }
   }
-#undef GG_DISPATCH
-#undef GG_LAUNCH
   FM_HIP_CHECK(hipGetLastError());
   return FM_OK;
 }
@@ -789,7 +1020,6 @@ static int moe_forward_impl(hipStream_t st, const void* x, const void* gate_w,
   if (rc != FM_OK) return rc;
   if (evs) FM_HIP_CHECK(hipEventRecord(evs[2], st));
 
-  const int mTiles = g.pEC / 128;
   GemmArgs up{};
   up.A = x;
   up.B = expert_w;
@@ -805,7 +1035,7 @@ static int moe_forward_impl(hipStream_t st, const void* x, const void* gate_w,
   up.EC = g.EC; up.pEC = g.pEC; up.PX = g.PX;
   up.topk = g.cfg.expert_top_k; up.act = g.cfg.hidden_act;
   up.expertOffset = 0; up.nRows = 0; up.H = g.H;
-  rc = launch_group_gemm(st, 0, up, mTiles, DIVUP(g.P, 128), g.E);
+  rc = launch_group_gemm(st, 0, up, g.pEC, g.E);
   if (rc != FM_OK) return rc;
   if (evs) FM_HIP_CHECK(hipEventRecord(evs[3], st));
 
@@ -818,13 +1048,13 @@ static int moe_forward_impl(hipStream_t st, const void* x, const void* gate_w,
   dn.moe_out = moe_out;
   dn.strideAExpert = (long long)g.pEC * g.P;
   dn.K = g.P; dn.N = g.H;
-  rc = launch_group_gemm(st, 1, dn, mTiles, DIVUP(g.H, 128), g.E);
+  rc = launch_group_gemm(st, 1, dn, g.pEC, g.E);
   if (rc != FM_OK) return rc;
   if (evs) FM_HIP_CHECK(hipEventRecord(evs[4], st));
 
   if (g.cfg.expert_top_k > 1) {
     const size_t n = (size_t)g.S * g.H;
-    const int blocks = (int)min((size_t)2048, DIVUP(n, (size_t)256 * 8));
+    const int blocks = (int)std::min((size_t)2048, (size_t)DIVUP(n, (size_t)256 * 8));
     if (g.esz == 2)
       hipLaunchKernelGGL(k_cast_out<bf16>, dim3(blocks), dim3(256), 0, st,
                          g.O32, reinterpret_cast<bf16*>(moe_out), n);
@@ -923,7 +1153,6 @@ int fm_expert_ffn(void* stream, const void* rows, const void* expert_w,
     setErr("n_rows exceeds workspace"); return FM_ERR_SHAPE;
   }
   hipStream_t st = reinterpret_cast<hipStream_t>(stream);
-  const int mTiles128 = (int)DIVUP(n_rows, 128);
   GemmArgs up{};
   up.A = rows;
   up.B = reinterpret_cast<const char*>(expert_w) +
@@ -935,7 +1164,7 @@ int fm_expert_ffn(void* stream, const void* rows, const void* expert_w,
   up.K = g.H; up.N = g.P; up.EC = 0; up.pEC = 0; up.PX = g.PX;
   up.topk = 1; up.act = g.cfg.hidden_act; up.expertOffset = 0;
   up.nRows = (int)n_rows; up.H = g.H;
-  int rc = launch_group_gemm(st, 0, up, mTiles128, DIVUP(g.P, 128), 1);
+  int rc = launch_group_gemm(st, 0, up, (int)n_rows, 1);
   if (rc != FM_OK) return rc;
   // note: PHASE 0 with tokenIds==nullptr writes out[e=0 stride 0] = xM rows
   GemmArgs dn = up;
@@ -945,7 +1174,7 @@ int fm_expert_ffn(void* stream, const void* rows, const void* expert_w,
   dn.bias = b_dn;
   dn.out = out_rows;
   dn.K = g.P; dn.N = g.H;
-  return launch_group_gemm(st, 2, dn, mTiles128, DIVUP(g.H, 128), 1);
+  return launch_group_gemm(st, 2, dn, (int)n_rows, 1);
 }
 
 int fm_combine(void* stream, const void* rows, const uint32_t* token_idx,
