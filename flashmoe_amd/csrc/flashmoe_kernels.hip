@@ -253,6 +253,163 @@ struct GemmArgs {
   int H;                // row stride of x / O32 / moe_out
 };
 
+// ---------------------------------------------------------------------------
+// Split gate (v2): the single-kernel gate ran on only S/128 blocks (32 for
+// config 2 - 12% of the CUs, 61% wave-park, profiles/r01). k_gate_logits
+// parallelises the logit GEMM over (token-tile, H-chunk) blocks with fp32
+// atomicAdd partials into a global logits buffer; k_gate_route then does
+// softmax/top-k/ordering per token tile (same semantics as k_gate,
+// gate.cuh:474-720).
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ __launch_bounds__(256) void k_gate_logits(
+    const T* __restrict__ x, const T* __restrict__ gate_w,
+    float* __restrict__ logits32, int S, int H, int E, int Hc) {
+  constexpr int BM = 128;
+  constexpr int BK = 64;
+  constexpr int RPAD = 8;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* part = reinterpret_cast<float*>(smem);          // [BM][E+1]
+  T* aCh = reinterpret_cast<T*>(part + BM * (E + 1));    // [BM][BK+RPAD]
+  T* gCh = aCh + BM * (BK + RPAD);                       // [E][BK+RPAD]
+  const int tid = threadIdx.x;
+  const int m0 = blockIdx.x * BM;
+  const int tok = tid & (BM - 1);
+  const int half = tid >> 7;
+  const int E2 = (E + 1) / 2;
+  const int e0 = half * E2;
+  const int e1 = min(e0 + E2, E);
+  const int k0 = blockIdx.y * Hc;
+  const int k1 = min(H, k0 + Hc);
+
+  for (int i = tid; i < BM * (E + 1); i += 256) part[i] = 0.0f;
+
+  const int LDA = BK + RPAD;
+  for (int kc = k0; kc < k1; kc += BK) {
+    constexpr int EPU = 16 / sizeof(T);
+    {
+      const int unitsA = BM * BK / EPU;
+      for (int u = tid; u < unitsA; u += 256) {
+        const int row = u / (BK / EPU);
+        const int cu = u % (BK / EPU);
+        *reinterpret_cast<u32x4*>(aCh + row * LDA + cu * EPU) =
+            *reinterpret_cast<const u32x4*>(x + (size_t)(m0 + row) * H + kc + cu * EPU);
+      }
+      const int unitsG = E * BK / EPU;
+      for (int u = tid; u < unitsG; u += 256) {
+        const int row = u / (BK / EPU);
+        const int cu = u % (BK / EPU);
+        *reinterpret_cast<u32x4*>(gCh + row * LDA + cu * EPU) =
+            *reinterpret_cast<const u32x4*>(gate_w + (size_t)row * H + kc + cu * EPU);
+      }
+    }
+    __syncthreads();
+    const T* arow = aCh + tok * LDA;
+    u32x4 a8[BK * sizeof(T) / 16];
+#pragma unroll
+    for (int jb = 0; jb < BK * (int)sizeof(T) / 16; ++jb)
+      a8[jb] = *reinterpret_cast<const u32x4*>(&arow[jb * (16 / sizeof(T))]);
+    for (int e = e0; e < e1; ++e) {
+      const T* grow = gCh + e * LDA;
+      float sAcc = 0.0f;
+#pragma unroll
+      for (int jb = 0; jb < BK * (int)sizeof(T) / 16; ++jb) {
+        const u32x4 g8 = *reinterpret_cast<const u32x4*>(&grow[jb * (16 / sizeof(T))]);
+#pragma unroll
+        for (int w = 0; w < 4; ++w) {
+          const uint32_t aw = a8[jb][w], gw = g8[w];
+          if constexpr (sizeof(T) == 2) {
+            const float2 av = __bfloat1622float2(
+                *reinterpret_cast<const __hip_bfloat162*>(&aw));
+            const float2 gv = __bfloat1622float2(
+                *reinterpret_cast<const __hip_bfloat162*>(&gw));
+            sAcc = fmaf(av.x, gv.x, sAcc);
+            sAcc = fmaf(av.y, gv.y, sAcc);
+          } else {
+            sAcc = fmaf(__uint_as_float(aw), __uint_as_float(gw), sAcc);
+          }
+        }
+      }
+      part[tok * (E + 1) + e] += sAcc;
+    }
+    __syncthreads();
+  }
+  // one atomic per (token, expert) per H-chunk
+  for (int e = e0; e < e1; ++e)
+    atomicAdd(logits32 + (size_t)(m0 + tok) * E + e, part[tok * (E + 1) + e]);
+}
+
+template <typename T, int K>
+__global__ __launch_bounds__(256) void k_gate_route(
+    const float* __restrict__ logits32, T* __restrict__ gate_out,
+    TPS* __restrict__ tokenIds, uint32_t* __restrict__ eC, int S, int E,
+    int PX, int EC, int pEC) {
+  constexpr int BM = 128;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* logits = reinterpret_cast<float*>(smem);                  // [BM][E+1]
+  uint16_t* sel = reinterpret_cast<uint16_t*>(logits + BM * (E + 1));
+  uint16_t* localIdx = sel + BM * K;
+  uint32_t* base = reinterpret_cast<uint32_t*>(localIdx + BM * K);  // [E]
+  const int tid = threadIdx.x;
+  const int m0 = blockIdx.x * BM;
+  // cooperative coalesced copy of the tile's logits into LDS
+  for (int i = tid; i < BM * E; i += 256) {
+    const int row = i / E, col = i % E;
+    logits[row * (E + 1) + col] = logits32[(size_t)(m0 + row) * E + col];
+  }
+  __syncthreads();
+
+  float mCw = 0.0f;
+  uint16_t mySel[K];
+  if (tid < BM) {
+    const float* lrow = logits + tid * (E + 1);
+    float m = -INFINITY;
+    for (int e = 0; e < E; ++e) m = fmaxf(m, lrow[e]);
+    float d = 0.0f;
+    for (int e = 0; e < E; ++e) d += __expf(lrow[e] - m);
+    const float inv_d = 1.0f / d;
+    T* grow = gate_out + (size_t)(m0 + tid) * PX;
+    for (int e = 0; e < E; ++e) fromF(__expf(lrow[e] - m) * inv_d, grow[e]);
+    for (int e = E; e < PX; ++e) fromF(0.0f, grow[e]);
+    uint64_t taken_lo = 0, taken_hi = 0;
+#pragma unroll
+    for (int i = 0; i < K; ++i) {
+      float sV = -INFINITY;
+      int sIdx = 0;
+      for (int e = 0; e < E; ++e) {
+        const bool taken = (e < 64) ? ((taken_lo >> e) & 1)
+                                    : ((taken_hi >> (e - 64)) & 1);
+        if (!taken && lrow[e] > sV) { sV = lrow[e]; sIdx = e; }
+      }
+      if (sIdx < 64) taken_lo |= 1ull << sIdx;
+      else taken_hi |= 1ull << (sIdx - 64);
+      mySel[i] = (uint16_t)sIdx;
+      mCw += __expf(sV - m) * inv_d;
+    }
+#pragma unroll
+    for (int i = 0; i < K; ++i) sel[tid * K + i] = mySel[i];
+  }
+  __syncthreads();
+  if (tid < E) {
+    uint32_t cnt = 0;
+    for (int mj = 0; mj < BM * K; ++mj) {
+      if (sel[mj] == tid) localIdx[mj] = (uint16_t)cnt++;
+    }
+    base[tid] = atomicAdd(eC + tid, cnt);
+  }
+  __syncthreads();
+  if (tid < BM) {
+#pragma unroll
+    for (int i = 0; i < K; ++i) {
+      const int e = mySel[i];
+      const uint32_t slot = base[e] + localIdx[tid * K + i];
+      if (slot < (uint32_t)EC)
+        tokenIds[(size_t)e * pEC + slot] = TPS{(uint32_t)(m0 + tid), mCw};
+    }
+  }
+}
+
 // address-space helpers for global_load_lds (direct HBM->LDS DMA)
 typedef __attribute__((address_space(1))) const uint32_t gas_u32;
 typedef __attribute__((address_space(3))) uint32_t las_u32;
@@ -452,12 +609,27 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
   TPS* sTps = reinterpret_cast<TPS*>(Bbase + 2 * BN * BK);
   uint32_t* sRouted = reinterpret_cast<uint32_t*>(sTps + BM);
 
-  const int e = blockIdx.z;
+  // XCD-aware block remap (perf only, placement-independent for
+  // correctness): the dispatcher places linear block b on XCD b%8; the
+  // bijective remap gives each XCD a CONTIGUOUS chunk of (e, nTile)
+  // space with mTile fastest, so the blocks sharing a B weight panel
+  // (and, for small E, a whole expert) run on one XCD and hit its L2
+  // instead of re-reading HBM/L3 (guide T1; staging-traffic-bound at
+  // 6.2 TB/s before this, profiles/r01 pmc2).
+  const int mT = gridDim.x, nT = gridDim.y;
+  const int nBlocks = mT * nT * gridDim.z;
+  const int lin = blockIdx.x + mT * (blockIdx.y + nT * blockIdx.z);
+  const int qx = nBlocks / 8, rx = nBlocks % 8;
+  const int xcd = lin % 8, pos = lin / 8;
+  const int swz =
+      (xcd < rx ? xcd * (qx + 1) : rx * (qx + 1) + (xcd - rx) * qx) + pos;
+  const int e = swz / (mT * nT);
+  const int rem = swz % (mT * nT);
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  const int m0 = blockIdx.x * BM;
-  const int n0 = blockIdx.y * BN;
+  const int m0 = (rem % mT) * BM;
+  const int n0 = (rem / mT) * BN;
   const int K = a.K, N = a.N;
 
   const TPS* tpsE = a.tokenIds ? a.tokenIds + (size_t)e * a.pEC : nullptr;
@@ -803,6 +975,7 @@ struct State {
   // workspace
   TPS* tokenIds = nullptr;   // [E, pEC]
   uint32_t* eC = nullptr;    // [E]
+  float* logits32 = nullptr; // [S, E] gate logit accumulator
   void* xM = nullptr;        // [nLx_alloc, pEC, P] Element
   float* O32 = nullptr;      // [S, H]
   int nLxAlloc = 0;
@@ -832,36 +1005,55 @@ size_t gate_lds_bytes(int E, size_t esz) {
 
 int launch_gate(hipStream_t st, const void* x, const void* gate_w,
                 void* gate_out, int64_t S) {
-  dim3 grid(S / 128), block(256);
-  const size_t lds = gate_lds_bytes(g.E, g.esz);
-#define GATE_CASE(T, KK)                                                      \
+  // split gate: logits GEMM parallelised over (token tile, H chunk),
+  // then per-tile softmax/top-k/route (see k_gate_logits/k_gate_route)
+  const int tiles = (int)(S / 128);
+  int chunksWanted = DIVUP(512, tiles);
+  int Hc = DIVUP(DIVUP(g.H, chunksWanted), 64) * 64;
+  const int chunks = DIVUP(g.H, Hc);
+  FM_HIP_CHECK(hipMemsetAsync(g.logits32, 0, (size_t)g.S * g.E * sizeof(float), st));
+  const size_t ldsL = gate_lds_bytes(g.E, g.esz);
+  const size_t ldsR = 128 * (g.E + 1) * sizeof(float) +
+                      128 * 8 * 2 * sizeof(uint16_t) + g.E * sizeof(uint32_t) + 64;
+#define GATE_LOGITS(T)                                                        \
   do {                                                                        \
-    if (lds > 64 * 1024) {                                                    \
+    if (ldsL > 64 * 1024)                                                     \
+      hipFuncSetAttribute(reinterpret_cast<const void*>(&k_gate_logits<T>),   \
+                          hipFuncAttributeMaxDynamicSharedMemorySize,         \
+                          (int)ldsL);                                         \
+    hipLaunchKernelGGL((k_gate_logits<T>), dim3(tiles, chunks), dim3(256),    \
+                       ldsL, st, reinterpret_cast<const T*>(x),               \
+                       reinterpret_cast<const T*>(gate_w), g.logits32,        \
+                       (int)S, g.H, g.E, Hc);                                 \
+  } while (0)
+#define GATE_ROUTE(T, KK)                                                     \
+  do {                                                                        \
+    if (ldsR > 64 * 1024)                                                     \
       hipFuncSetAttribute(                                                    \
-          reinterpret_cast<const void*>(&k_gate<T, KK>),                      \
-          hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);              \
-    }                                                                         \
-    hipLaunchKernelGGL((k_gate<T, KK>), grid, block, lds, st,                 \
-                       reinterpret_cast<const T*>(x),                         \
-                       reinterpret_cast<const T*>(gate_w),                    \
-                       reinterpret_cast<T*>(gate_out), g.tokenIds, g.eC,      \
-                       (int)S, g.H, g.E, g.PX, g.EC, g.pEC);                  \
+          reinterpret_cast<const void*>(&k_gate_route<T, KK>),                \
+          hipFuncAttributeMaxDynamicSharedMemorySize, (int)ldsR);             \
+    hipLaunchKernelGGL((k_gate_route<T, KK>), dim3(tiles), dim3(256), ldsR,   \
+                       st, g.logits32, reinterpret_cast<T*>(gate_out),        \
+                       g.tokenIds, g.eC, (int)S, g.E, g.PX, g.EC, g.pEC);     \
   } while (0)
 #define GATE_K(T)                                                             \
   switch (g.cfg.expert_top_k) {                                               \
-    case 1: GATE_CASE(T, 1); break;                                           \
-    case 2: GATE_CASE(T, 2); break;                                           \
-    case 4: GATE_CASE(T, 4); break;                                           \
-    case 8: GATE_CASE(T, 8); break;                                           \
+    case 1: GATE_ROUTE(T, 1); break;                                          \
+    case 2: GATE_ROUTE(T, 2); break;                                          \
+    case 4: GATE_ROUTE(T, 4); break;                                          \
+    case 8: GATE_ROUTE(T, 8); break;                                          \
     default: setErr("unsupported expert_top_k (1,2,4,8)"); return FM_ERR_UNSUPPORTED; \
   }
   if (g.cfg.dtype == 2) {
+    GATE_LOGITS(bf16);
     GATE_K(bf16)
   } else {
+    GATE_LOGITS(float);
     GATE_K(float)
   }
 #undef GATE_K
-#undef GATE_CASE
+#undef GATE_ROUTE
+#undef GATE_LOGITS
   FM_HIP_CHECK(hipGetLastError());
   return FM_OK;
 }
@@ -909,6 +1101,7 @@ int fm_initialize(const fm_config* cfg, int rank, int world_size) {
   g.nLxAlloc = (world_size == 1) ? g.E
                                  : g.nLx * DIVUP(world_size * g.EC, g.pEC);
   FM_HIP_CHECK(hipMalloc(&g.tokenIds, (size_t)g.E * g.pEC * sizeof(TPS)));
+  FM_HIP_CHECK(hipMalloc(&g.logits32, (size_t)g.S * g.E * sizeof(float)));
   FM_HIP_CHECK(hipMalloc(&g.eC, (size_t)g.E * sizeof(uint32_t)));
   FM_HIP_CHECK(hipMalloc(&g.xM, (size_t)g.nLxAlloc * g.pEC * g.P * g.esz));
   FM_HIP_CHECK(hipMalloc(&g.O32, (size_t)g.S * g.H * sizeof(float)));
@@ -919,6 +1112,7 @@ int fm_initialize(const fm_config* cfg, int rank, int world_size) {
 int fm_finalize(void) {
   if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
   hipFree(g.tokenIds); hipFree(g.eC); hipFree(g.xM); hipFree(g.O32);
+  hipFree(g.logits32);
   g = State{};
   return FM_OK;
 }
