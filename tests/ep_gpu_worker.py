@@ -22,7 +22,7 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     cfg = {
-        "capacity_factor": 1, "drop_tokens": 1, "expert_top_k": 2,
+        "capacity_factor": 2, "drop_tokens": 1, "expert_top_k": 2,
         "global_batch": 256, "is_training": 0, "hidden_act": 0,
         "hidden_size": 256, "intermediate_size": 512, "mini_batch": 1,
         "moe_frequency": 1, "num_experts": 8, "num_layers": 1,

@@ -30,7 +30,7 @@ def test_staged_entry_points_match_oracle():
     import flashmoe_amd._ext as _ext
 
     cfg = {
-        "capacity_factor": 1, "drop_tokens": 1, "expert_top_k": 2,
+        "capacity_factor": 2, "drop_tokens": 1, "expert_top_k": 2,
         "global_batch": 256, "is_training": 0, "hidden_act": 0,
         "hidden_size": 128, "intermediate_size": 256, "mini_batch": 1,
         "moe_frequency": 1, "num_experts": 4, "num_layers": 1,
@@ -43,7 +43,7 @@ def test_staged_entry_points_match_oracle():
     try:
         lib = _ext.load()
         S, H, P, E, k = 256, 128, 256, 4, 2
-        EC = (S // E) * k
+        EC = (S // E) * k * 2  # capacity_factor 2
         g = torch.Generator().manual_seed(7)
         x = torch.randn(S, H, generator=g).to(torch.bfloat16).cuda()
         gw = torch.randn(H, E, generator=g).to(torch.bfloat16).cuda()
@@ -92,7 +92,8 @@ def test_staged_entry_points_match_oracle():
             "finalize")
         torch.cuda.synchronize()
 
-        ocfg = OracleConfig(num_experts=E, expert_top_k=k, element="bf16")
+        ocfg = OracleConfig(num_experts=E, expert_top_k=k, capacity_factor=2,
+                            element="bf16")
         ref = oracle_forward(x.float().cpu().numpy(),
                              gw.float().cpu().numpy().reshape(-1),
                              ew.float().cpu().numpy(), ocfg)
