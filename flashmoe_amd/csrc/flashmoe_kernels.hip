@@ -266,26 +266,29 @@ template <typename T>
 __global__ __launch_bounds__(256) void k_gate_logits(
     const T* __restrict__ x, const T* __restrict__ gate_w,
     float* __restrict__ logits32, int S, int H, int E, int Hc) {
+  // blockIdx.z selects a chunk of <=128 experts (E up to 256, config 5)
+  const int eBase = blockIdx.z * 128;
+  const int Ec = min(E - eBase, 128);
   constexpr int BM = 128;
   constexpr int BK = 64;
   constexpr int RPAD = 8;
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  float* part = reinterpret_cast<float*>(smem);          // [BM][E+1]
-  T* aCh = reinterpret_cast<T*>(part + BM * (E + 1));    // [BM][BK+RPAD]
-  T* gCh = aCh + BM * (BK + RPAD);                       // [E][BK+RPAD]
+  float* part = reinterpret_cast<float*>(smem);          // [BM][Ec+1]
+  T* aCh = nullptr;  // assigned after Ec is known (see below)
   const int tid = threadIdx.x;
   const int m0 = blockIdx.x * BM;
   const int tok = tid & (BM - 1);
   const int half = tid >> 7;
-  const int E2 = (E + 1) / 2;
+  const int E2 = (Ec + 1) / 2;
   const int e0 = half * E2;
-  const int e1 = min(e0 + E2, E);
+  const int e1 = min(e0 + E2, Ec);
   const int k0 = blockIdx.y * Hc;
   const int k1 = min(H, k0 + Hc);
-
-  for (int i = tid; i < BM * (E + 1); i += 256) part[i] = 0.0f;
-
+  aCh = reinterpret_cast<T*>(part + BM * (Ec + 1));      // [BM][BK+RPAD]
   const int LDA = BK + RPAD;
+  T* gCh = aCh + BM * LDA;                               // [Ec][BK+RPAD]
+
+  for (int i = tid; i < BM * (Ec + 1); i += 256) part[i] = 0.0f;
   for (int kc = k0; kc < k1; kc += BK) {
     constexpr int EPU = 16 / sizeof(T);
     {
@@ -296,12 +299,13 @@ __global__ __launch_bounds__(256) void k_gate_logits(
         *reinterpret_cast<u32x4*>(aCh + row * LDA + cu * EPU) =
             *reinterpret_cast<const u32x4*>(x + (size_t)(m0 + row) * H + kc + cu * EPU);
       }
-      const int unitsG = E * BK / EPU;
+      const int unitsG = Ec * BK / EPU;
       for (int u = tid; u < unitsG; u += 256) {
         const int row = u / (BK / EPU);
         const int cu = u % (BK / EPU);
         *reinterpret_cast<u32x4*>(gCh + row * LDA + cu * EPU) =
-            *reinterpret_cast<const u32x4*>(gate_w + (size_t)row * H + kc + cu * EPU);
+            *reinterpret_cast<const u32x4*>(gate_w + (size_t)(eBase + row) * H +
+                                            kc + cu * EPU);
       }
     }
     __syncthreads();
@@ -331,13 +335,14 @@ __global__ __launch_bounds__(256) void k_gate_logits(
           }
         }
       }
-      part[tok * (E + 1) + e] += sAcc;
+      part[tok * (Ec + 1) + e] += sAcc;
     }
     __syncthreads();
   }
   // one atomic per (token, expert) per H-chunk
   for (int e = e0; e < e1; ++e)
-    atomicAdd(logits32 + (size_t)(m0 + tok) * E + e, part[tok * (E + 1) + e]);
+    atomicAdd(logits32 + (size_t)(m0 + tok) * E + eBase + e,
+              part[tok * (Ec + 1) + e]);
 }
 
 template <typename T, int K>
@@ -372,18 +377,18 @@ __global__ __launch_bounds__(256) void k_gate_route(
     T* grow = gate_out + (size_t)(m0 + tid) * PX;
     for (int e = 0; e < E; ++e) fromF(__expf(lrow[e] - m) * inv_d, grow[e]);
     for (int e = E; e < PX; ++e) fromF(0.0f, grow[e]);
-    uint64_t taken_lo = 0, taken_hi = 0;
+    // iterative argmax, strict >, first index wins; "taken" by direct
+    // comparison with earlier selections (E up to 256)
 #pragma unroll
     for (int i = 0; i < K; ++i) {
       float sV = -INFINITY;
       int sIdx = 0;
       for (int e = 0; e < E; ++e) {
-        const bool taken = (e < 64) ? ((taken_lo >> e) & 1)
-                                    : ((taken_hi >> (e - 64)) & 1);
+        bool taken = false;
+#pragma unroll
+        for (int j = 0; j < K; ++j) taken |= (j < i) && (mySel[j] == e);
         if (!taken && lrow[e] > sV) { sV = lrow[e]; sIdx = e; }
       }
-      if (sIdx < 64) taken_lo |= 1ull << sIdx;
-      else taken_hi |= 1ull << (sIdx - 64);
       mySel[i] = (uint16_t)sIdx;
       mCw += __expf(sV - m) * inv_d;
     }
@@ -1016,8 +1021,10 @@ int launch_gate(hipStream_t st, const void* x, const void* gate_w,
   int chunksWanted = DIVUP(512, tiles);
   int Hc = DIVUP(DIVUP(g.H, chunksWanted), 64) * 64;
   const int chunks = DIVUP(g.H, Hc);
+  const int eChunks = DIVUP(g.E, 128);
+  const int Ec = g.E < 128 ? g.E : 128;
   FM_HIP_CHECK(hipMemsetAsync(g.logits32, 0, (size_t)g.S * g.E * sizeof(float), st));
-  const size_t ldsL = gate_lds_bytes(g.E, g.esz);
+  const size_t ldsL = gate_lds_bytes(Ec, g.esz);
   const size_t ldsR = 128 * (g.E + 1) * sizeof(float) +
                       128 * 8 * 2 * sizeof(uint16_t) + g.E * sizeof(uint32_t) + 64;
 #define GATE_LOGITS(T)                                                        \
@@ -1026,8 +1033,8 @@ int launch_gate(hipStream_t st, const void* x, const void* gate_w,
       hipFuncSetAttribute(reinterpret_cast<const void*>(&k_gate_logits<T>),   \
                           hipFuncAttributeMaxDynamicSharedMemorySize,         \
                           (int)ldsL);                                         \
-    hipLaunchKernelGGL((k_gate_logits<T>), dim3(tiles, chunks), dim3(256),    \
-                       ldsL, st, reinterpret_cast<const T*>(x),               \
+    hipLaunchKernelGGL((k_gate_logits<T>), dim3(tiles, chunks, eChunks),     \
+                       dim3(256), ldsL, st, reinterpret_cast<const T*>(x),    \
                        reinterpret_cast<const T*>(gate_w), g.logits32,        \
                        (int)S, g.H, g.E, Hc);                                 \
   } while (0)
@@ -1095,7 +1102,7 @@ int fm_initialize(const fm_config* cfg, int rank, int world_size) {
     default: setErr("dtype not supported this round (fp32/bf16)");
              return FM_ERR_UNSUPPORTED;
   }
-  if (g.E > 128) { setErr("E > 128 not supported this round"); return FM_ERR_UNSUPPORTED; }
+  if (g.E > 256) { setErr("E > 256 not supported this round"); return FM_ERR_UNSUPPORTED; }
   if (g.H % 64 || g.P % 64) { setErr("H and P must be multiples of 64"); return FM_ERR_SHAPE; }
   if (g.esz == 2 && (g.H % 128 || g.P % 128)) {
     setErr("bf16 path requires H, P multiples of 128 this round");

@@ -163,6 +163,24 @@ def test_single_tile_fp32_top1_config1(fresh_moe):
     assert_values(out, ref["moe_out"], "fp32", "moe_out")
 
 
+def test_single_tile_bf16_top1(fresh_moe):
+    """k==1: unscaled direct store path (CombineMode::single,
+    processor.cuh:173-204)."""
+    cfg, path = make_cfg(expert_top_k=1)
+    out, gate_out, ref, _ = run_pair(fresh_moe, cfg, path)
+    assert_values(out, ref["moe_out"], "bf16", "moe_out")
+
+
+def test_single_expert_dense_fallback(fresh_moe):
+    """E == 1: the reference's dense-FFN fallback shape (moe.cuh:174-177,
+    fffn.cuh) — here the generic path routes every token to expert 0."""
+    cfg, path = make_cfg(num_experts=1, expert_top_k=1, drop_tokens=0)
+    out, gate_out, ref, _ = run_pair(fresh_moe, cfg, path)
+    g = gate_out.float().cpu().numpy()
+    assert np.allclose(g[:, 0], 1.0, atol=1e-3)  # softmax over one expert
+    assert_values(out, ref["moe_out"], "bf16", "moe_out")
+
+
 def test_single_tile_gelu(fresh_moe):
     cfg, path = make_cfg(hidden_act=1)
     out, gate_out, ref, _ = run_pair(fresh_moe, cfg, path)
@@ -172,6 +190,19 @@ def test_single_tile_gelu(fresh_moe):
 def test_single_tile_top4(fresh_moe):
     cfg, path = make_cfg(expert_top_k=4, num_experts=16)
     out, gate_out, ref, _ = run_pair(fresh_moe, cfg, path)
+    assert_values(out, ref["moe_out"], "bf16", "moe_out")
+
+
+@pytest.mark.parametrize("E,k", [(192, 2), (256, 4)])
+def test_many_experts_gate(fresh_moe, E, k):
+    """E > 128 (config-5 shape): the expert-chunked gate logits kernel
+    and >=192-way top-k routing."""
+    # single 128-token tile: routing deterministic even if an expert
+    # overflows its tiny capacity (EC = ceil(128/E)*CF*k)
+    cfg, path = make_cfg(num_experts=E, expert_top_k=k, sequence_len=128,
+                         capacity_factor=2)
+    out, gate_out, ref, _ = run_pair(fresh_moe, cfg, path)
+    assert_values(gate_out, ref["gate_out"], "bf16", "gate_out")
     assert_values(out, ref["moe_out"], "bf16", "moe_out")
 
 
