@@ -58,6 +58,8 @@ def _bind(lib: ctypes.CDLL) -> ctypes.CDLL:
     lib.fm_combine.restype = ctypes.c_int
     lib.fm_combine_finalize.argtypes = [p, p, i64]
     lib.fm_combine_finalize.restype = ctypes.c_int
+    lib.fm_read_aux_loss.argtypes = [p, p, p]
+    lib.fm_read_aux_loss.restype = ctypes.c_int
     lib.fm_last_error.argtypes = []
     lib.fm_last_error.restype = ctypes.c_char_p
     lib.fm_built_for_gfx950.argtypes = []

@@ -349,13 +349,19 @@ template <typename T, int K>
 __global__ __launch_bounds__(256) void k_gate_route(
     const float* __restrict__ logits32, T* __restrict__ gate_out,
     TPS* __restrict__ tokenIds, uint32_t* __restrict__ eC, int S, int E,
-    int PX, int EC, int pEC) {
+    int PX, int EC, int pEC, float* __restrict__ gML,
+    float* __restrict__ gMeC) {
+  // gML/gMeC: training-mode aux-loss accumulators (gate.cuh:273-299,
+  // 763-773; types.cuh:936-958): gML[e] += column-mean of softmax probs,
+  // gMeC[e] += routed fraction. Null in inference mode.
   constexpr int BM = 128;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* logits = reinterpret_cast<float*>(smem);                  // [BM][E+1]
   uint16_t* sel = reinterpret_cast<uint16_t*>(logits + BM * (E + 1));
   uint16_t* localIdx = sel + BM * K;
   uint32_t* base = reinterpret_cast<uint32_t*>(localIdx + BM * K);  // [E]
+  float* sInv = reinterpret_cast<float*>(base + E);                // [BM] 1/d
+  float* sMax = sInv + BM;                                         // [BM]
   const int tid = threadIdx.x;
   const int m0 = blockIdx.x * BM;
   // cooperative coalesced copy of the tile's logits into LDS
@@ -374,6 +380,7 @@ __global__ __launch_bounds__(256) void k_gate_route(
     float d = 0.0f;
     for (int e = 0; e < E; ++e) d += __expf(lrow[e] - m);
     const float inv_d = 1.0f / d;
+    if (gML) { sInv[tid] = inv_d; sMax[tid] = m; }
     T* grow = gate_out + (size_t)(m0 + tid) * PX;
     for (int e = 0; e < E; ++e) fromF(__expf(lrow[e] - m) * inv_d, grow[e]);
     for (int e = E; e < PX; ++e) fromF(0.0f, grow[e]);
@@ -402,6 +409,14 @@ __global__ __launch_bounds__(256) void k_gate_route(
       if (sel[mj] == tid) localIdx[mj] = (uint16_t)cnt++;
     }
     base[tid] = atomicAdd(eC + tid, cnt);
+    if (gML) {
+      // column sum of probs for this expert over the tile's tokens
+      float colSum = 0.0f;
+      for (int mj = 0; mj < BM; ++mj)
+        colSum += __expf(logits[mj * (E + 1) + tid] - sMax[mj]) * sInv[mj];
+      atomicAdd(gML + tid, colSum / (float)S);
+      atomicAdd(gMeC + tid, (float)cnt / (float)S);
+    }
   }
   __syncthreads();
   if (tid < BM) {
@@ -986,6 +1001,8 @@ struct State {
   TPS* tokenIds = nullptr;   // [E, pEC]
   uint32_t* eC = nullptr;    // [E]
   float* logits32 = nullptr; // [S, E] gate logit accumulator
+  float* gML = nullptr;      // [E] training aux: mean gate prob
+  float* gMeC = nullptr;     // [E] training aux: routed fraction
   void* xM = nullptr;        // [nLx_alloc, pEC, P] Element
   float* O32 = nullptr;      // [S, H]
   int nLxAlloc = 0;
@@ -1026,7 +1043,12 @@ int launch_gate(hipStream_t st, const void* x, const void* gate_w,
   FM_HIP_CHECK(hipMemsetAsync(g.logits32, 0, (size_t)g.S * g.E * sizeof(float), st));
   const size_t ldsL = gate_lds_bytes(Ec, g.esz);
   const size_t ldsR = 128 * (g.E + 1) * sizeof(float) +
-                      128 * 8 * 2 * sizeof(uint16_t) + g.E * sizeof(uint32_t) + 64;
+                      128 * 8 * 2 * sizeof(uint16_t) + g.E * sizeof(uint32_t) +
+                      2 * 128 * sizeof(float) + 64;
+  float* gMLp = g.cfg.is_training ? g.gML : nullptr;
+  float* gMeCp = g.cfg.is_training ? g.gMeC : nullptr;
+  if (g.cfg.is_training)
+    FM_HIP_CHECK(hipMemsetAsync(g.gML, 0, 2 * (size_t)g.E * sizeof(float), st));
 #define GATE_LOGITS(T)                                                        \
   do {                                                                        \
     if (ldsL > 64 * 1024)                                                     \
@@ -1046,7 +1068,8 @@ int launch_gate(hipStream_t st, const void* x, const void* gate_w,
           hipFuncAttributeMaxDynamicSharedMemorySize, (int)ldsR);             \
     hipLaunchKernelGGL((k_gate_route<T, KK>), dim3(tiles), dim3(256), ldsR,   \
                        st, g.logits32, reinterpret_cast<T*>(gate_out),        \
-                       g.tokenIds, g.eC, (int)S, g.E, g.PX, g.EC, g.pEC);     \
+                       g.tokenIds, g.eC, (int)S, g.E, g.PX, g.EC, g.pEC,      \
+                       gMLp, gMeCp);                                          \
   } while (0)
 #define GATE_K(T)                                                             \
   switch (g.cfg.expert_top_k) {                                               \
@@ -1114,6 +1137,8 @@ int fm_initialize(const fm_config* cfg, int rank, int world_size) {
                                  : g.nLx * DIVUP(world_size * g.EC, g.pEC);
   FM_HIP_CHECK(hipMalloc(&g.tokenIds, (size_t)g.E * g.pEC * sizeof(TPS)));
   FM_HIP_CHECK(hipMalloc(&g.logits32, (size_t)g.S * g.E * sizeof(float)));
+  FM_HIP_CHECK(hipMalloc(&g.gML, 2 * (size_t)g.E * sizeof(float)));
+  g.gMeC = g.gML + g.E;
   FM_HIP_CHECK(hipMalloc(&g.eC, (size_t)g.E * sizeof(uint32_t)));
   FM_HIP_CHECK(hipMalloc(&g.xM, (size_t)g.nLxAlloc * g.pEC * g.P * g.esz));
   FM_HIP_CHECK(hipMalloc(&g.O32, (size_t)g.S * g.H * sizeof(float)));
@@ -1125,7 +1150,7 @@ int fm_initialize(const fm_config* cfg, int rank, int world_size) {
 int fm_finalize(void) {
   if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
   hipFree(g.tokenIds); hipFree(g.eC); hipFree(g.xM); hipFree(g.O32);
-  hipFree(g.logits32);
+  hipFree(g.logits32); hipFree(g.gML);
   g = State{};
   return FM_OK;
 }
@@ -1418,6 +1443,18 @@ int fm_combine_finalize(void* stream, void* moe_out, int64_t S) {
     hipLaunchKernelGGL(k_cast_out<float>, dim3(blocks), dim3(256), 0, st,
                        g.O32, reinterpret_cast<float*>(moe_out), n);
   FM_HIP_CHECK(hipGetLastError());
+  return FM_OK;
+}
+
+int fm_read_aux_loss(void* stream, float* gML, float* gMeC) {
+  if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
+  if (!g.cfg.is_training) { setErr("aux loss requires is_training=1"); return FM_ERR_STATE; }
+  hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+  FM_HIP_CHECK(hipMemcpyAsync(gML, g.gML, (size_t)g.E * sizeof(float),
+                              hipMemcpyDeviceToHost, st));
+  FM_HIP_CHECK(hipMemcpyAsync(gMeC, g.gMeC, (size_t)g.E * sizeof(float),
+                              hipMemcpyDeviceToHost, st));
+  FM_HIP_CHECK(hipStreamSynchronize(st));
   return FM_OK;
 }
 

@@ -135,6 +135,12 @@ int fm_moe_forward_phased(void* stream, const void* x, const void* gate_w,
                           const void* b_dn, void* gate_out, void* moe_out,
                           int64_t S, float ms[4]);
 
+/* Training-mode auxiliary-loss accumulators (gate.cuh:273-299,763-773;
+ * types.cuh:936-958): gML[e] = mean softmax prob of expert e over the
+ * last forward's tokens, gMeC[e] = fraction routed to e (pre-capacity).
+ * Host arrays of E floats; synchronizes the stream. is_training=1 only. */
+int fm_read_aux_loss(void* stream, float* gML, float* gMeC);
+
 /* Version/introspection */
 const char* fm_last_error(void);
 int fm_built_for_gfx950(void);

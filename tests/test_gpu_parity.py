@@ -221,6 +221,33 @@ def test_multi_tile_no_overflow_bf16(fresh_moe):
     assert_values(out, ref["moe_out"], "bf16", "moe_out")
 
 
+def test_training_aux_loss(fresh_moe):
+    """is_training=1: gML (mean gate prob) and gMeC (routed fraction)
+    accumulators vs the oracle (gate.cuh:273-299,763-773)."""
+    from oracle.moe_oracle import gate_aux_loss
+    import flashmoe_amd._ext as _ext
+
+    cfg, path = make_cfg(is_training=1, sequence_len=512)
+    out, gate_out, ref, ocfg = run_pair(fresh_moe, cfg, path)
+    E = cfg["num_experts"]
+    lib = _ext.load()
+    gML = np.zeros(E, dtype=np.float32)
+    gMeC = np.zeros(E, dtype=np.float32)
+    _ext.check(lib.fm_read_aux_loss(
+        None, ctypes.c_void_p(gML.ctypes.data),
+        ctypes.c_void_p(gMeC.ctypes.data)), "aux_loss")
+    # oracle reference on the same inputs
+    S, H = 512, cfg["hidden_size"]
+    g = torch.Generator(device="cpu").manual_seed(47)
+    x = torch.randn(1, 512, H, generator=g).to(torch.bfloat16)
+    gw = torch.randn(H, E, generator=g).to(torch.bfloat16)
+    want_gML, want_gMeC = gate_aux_loss(
+        x.view(S, H).float().numpy(),
+        gw.float().numpy().reshape(-1).reshape(E, H), ocfg)
+    np.testing.assert_allclose(gML, want_gML, rtol=1e-2, atol=1e-4)
+    np.testing.assert_allclose(gMeC, want_gMeC, rtol=1e-5, atol=1e-6)
+
+
 def test_bench_config_invariants(fresh_moe):
     """BASELINE config 2 at full size (S=4096, H=1024, P=4096, E=8, k=2,
     CF=1): size-independent properties + routing counts vs oracle
