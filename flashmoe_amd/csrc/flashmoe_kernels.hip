@@ -730,17 +730,21 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
       bf16x8 af[8], bfr[NF];
       const int rl = lane & 15;
       const int cbase = 4 * s + (lane >> 4);
-#pragma unroll
-      for (int mi = 0; mi < 8; ++mi) {
-        const int R = wr * 128 + mi * 16 + rl;
-        af[mi] = *reinterpret_cast<const bf16x8*>(
-            &Al[R * BK + ((cbase ^ (R & 7)) * 8)]);
-      }
+      // read the B fragments FIRST: the first MFMA (af[0] x bfr[*])
+      // then depends only on the B reads + af[0], so the compiler can
+      // ladder partial lgkmcnt waits instead of draining all 12 reads
+      // before the first MFMA (seen in the .s with af-first order)
 #pragma unroll
       for (int ni = 0; ni < NF; ++ni) {
         const int R = wc * (BN / 4) + ni * 16 + rl;
         bfr[ni] = *reinterpret_cast<const bf16x8*>(
             &Bl[R * BK + ((cbase ^ (R & 7)) * 8)]);
+      }
+#pragma unroll
+      for (int mi = 0; mi < 8; ++mi) {
+        const int R = wr * 128 + mi * 16 + rl;
+        af[mi] = *reinterpret_cast<const bf16x8*>(
+            &Al[R * BK + ((cbase ^ (R & 7)) * 8)]);
       }
 #pragma unroll
       for (int mi = 0; mi < 8; ++mi)
