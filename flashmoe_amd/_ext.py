@@ -58,6 +58,8 @@ def _bind(lib: ctypes.CDLL) -> ctypes.CDLL:
     lib.fm_combine.restype = ctypes.c_int
     lib.fm_combine_finalize.argtypes = [p, p, i64]
     lib.fm_combine_finalize.restype = ctypes.c_int
+    lib.fm_export_routing.argtypes = [p, p, p]
+    lib.fm_export_routing.restype = ctypes.c_int
     lib.fm_read_aux_loss.argtypes = [p, p, p]
     lib.fm_read_aux_loss.restype = ctypes.c_int
     lib.fm_last_error.argtypes = []

@@ -971,6 +971,23 @@ __global__ void k_combine_rows(const T* __restrict__ rows,
   }
 }
 
+// export clipped routing into caller tensors: routed[e] = min(eC, EC),
+// out[e][i] = {tokenIdx, probSum-bits} for i < routed (EP host pipeline)
+__global__ void k_export_routing(const TPS* __restrict__ tokenIds,
+                                 const uint32_t* __restrict__ eC,
+                                 uint32_t* __restrict__ routed,
+                                 uint32_t* __restrict__ out, int E, int EC,
+                                 int pEC) {
+  const int e = blockIdx.x;
+  const uint32_t r = min(eC[e], (uint32_t)EC);
+  if (threadIdx.x == 0) routed[e] = r;
+  for (int i = threadIdx.x; i < EC; i += blockDim.x) {
+    TPS t = (i < (int)r) ? tokenIds[(size_t)e * pEC + i] : TPS{0u, 0.0f};
+    out[((size_t)e * EC + i) * 2] = t.tokenIdx;
+    out[((size_t)e * EC + i) * 2 + 1] = __float_as_uint(t.probSum);
+  }
+}
+
 // MFMA layout probe (test-only): D = A[16x32] x B[32x16] via one
 // mfma_f32_16x16x32_bf16, written with the assumed C/D mapping.
 __global__ void k_mfma_probe(const bf16* A, const bf16* B, float* D) {
@@ -1446,6 +1463,17 @@ int fm_combine_finalize(void* stream, void* moe_out, int64_t S) {
   else
     hipLaunchKernelGGL(k_cast_out<float>, dim3(blocks), dim3(256), 0, st,
                        g.O32, reinterpret_cast<float*>(moe_out), n);
+  FM_HIP_CHECK(hipGetLastError());
+  return FM_OK;
+}
+
+int fm_export_routing(void* stream, void* routed_dev, void* tps_dev) {
+  if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
+  hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+  hipLaunchKernelGGL(k_export_routing, dim3(g.E), dim3(256), 0, st,
+                     g.tokenIds, g.eC,
+                     reinterpret_cast<uint32_t*>(routed_dev),
+                     reinterpret_cast<uint32_t*>(tps_dev), g.E, g.EC, g.pEC);
   FM_HIP_CHECK(hipGetLastError());
   return FM_OK;
 }

@@ -69,6 +69,31 @@ def plan_dispatch(routed_counts: torch.Tensor, token_idx_2d: torch.Tensor,
     )
 
 
+def plan_dispatch_vec(counts: torch.Tensor, tok2d: torch.Tensor,
+                      ps2d: torch.Tensor, world: int) -> DispatchPlan:
+    """Vectorized plan_dispatch (no python loop over experts): same
+    canonical order — expert-major, slot-minor; experts are owned
+    contiguously so destination-rank grouping is automatic. All tensor
+    ops, so it runs on-GPU in the EP hot path (tested equivalent to
+    plan_dispatch in tests/test_ep_gloo.py)."""
+    E, ECap = tok2d.shape
+    nLx = E // world
+    slot = torch.arange(ECap, device=counts.device)
+    mask = slot[None, :] < counts[:, None]            # [E, EC]
+    token_idx = tok2d[mask]                           # row-major -> canonical
+    prob_sum = ps2d[mask]
+    e_ids = torch.arange(E, device=counts.device)[:, None].expand(E, ECap)
+    expert_of_row = e_ids[mask]
+    sec = counts.reshape(world, nLx)
+    return DispatchPlan(
+        send_counts=sec.sum(1),
+        send_expert_counts=sec,
+        token_idx=token_idx.long(),
+        prob_sum=prob_sum.float(),
+        expert_of_row=expert_of_row,
+    )
+
+
 def segment_recv(recv_expert_counts: torch.Tensor):
     """recv_expert_counts: [world, nLx] rows per (source, local expert) in
     the canonical receive order (source-major, expert-minor). Returns a
@@ -133,9 +158,9 @@ def exchange_rows(x_rows: torch.Tensor, send_counts: torch.Tensor,
 
 def moe_forward_ep(input, gate_weights, expert_weights, group=None):
     """The EP forward on GPU ranks (called from moe.moe_forward when
-    world > 1). Requires torch.distributed initialized with the nccl
-    (RCCL) backend."""
-    import numpy as np
+    world > 1). Planning runs on-GPU from fm_export_routing; the only
+    host syncs are the [world]-sized split counts the RCCL all_to_all
+    API needs as python ints."""
     import torch.distributed as dist
 
     from . import _ext, moe
@@ -151,6 +176,7 @@ def moe_forward_ep(input, gate_weights, expert_weights, group=None):
     EC = base * cfg["capacity_factor"] * k
     gate_out = moe._state["gate_out"]
     stream = torch.cuda.current_stream().cuda_stream
+    dev = input.device
     x2d = input.view(S, H)
 
     # 1. local gate (fills the library's tokenIds/eC)
@@ -160,31 +186,29 @@ def moe_forward_ep(input, gate_weights, expert_weights, group=None):
         ctypes.c_void_p(gate_weights.data_ptr()),
         ctypes.c_void_p(gate_out.data_ptr()), S), "fm_gate_forward")
 
-    # 2. routing to host
-    counts = np.zeros(E, dtype=np.uint32)
-    tok = np.zeros(E * EC, dtype=np.uint32)
-    ps = np.zeros(E * EC, dtype=np.float32)
-    _ext.check(lib.fm_read_routing(
-        ctypes.c_void_p(stream),
-        ctypes.c_void_p(counts.ctypes.data),
-        ctypes.c_void_p(tok.ctypes.data),
-        ctypes.c_void_p(ps.ctypes.data)), "fm_read_routing")
-    plan = plan_dispatch(torch.from_numpy(counts.astype(np.int64)),
-                         torch.from_numpy(tok.reshape(E, EC).astype(np.int64)),
-                         torch.from_numpy(ps.reshape(E, EC)), world)
+    # 2. GPU-resident routing export + vectorized plan
+    routed = torch.empty(E, dtype=torch.int32, device=dev)
+    tps = torch.empty(E, EC, 2, dtype=torch.int32, device=dev)
+    _ext.check(lib.fm_export_routing(
+        ctypes.c_void_p(stream), ctypes.c_void_p(routed.data_ptr()),
+        ctypes.c_void_p(tps.data_ptr())), "fm_export_routing")
+    tps_f = tps.view(torch.float32)  # reinterpret bits; [E, EC, 2]
+    plan = plan_dispatch_vec(routed.long(), tps[..., 0].long(),
+                             tps_f[..., 1], world)
 
-    # 3. exchange counts, then rows
+    # 3. exchange per-(dst, expert) counts, then rows
     all_expert_counts = exchange_rows(
-        plan.send_expert_counts.reshape(world, nLx).to(input.device),
+        plan.send_expert_counts.contiguous(),
         torch.ones(world, dtype=torch.long),
         torch.ones(world, dtype=torch.long), group).cpu()  # [world, nLx]
     recv_counts = all_expert_counts.sum(1)
-    send_rows = x2d.index_select(0, plan.token_idx.to(input.device))
-    recv_rows = exchange_rows(send_rows, plan.send_counts, recv_counts, group)
+    send_counts_cpu = plan.send_counts.cpu()  # host sync: world ints
+    send_rows = x2d.index_select(0, plan.token_idx)
+    recv_rows = exchange_rows(send_rows, send_counts_cpu, recv_counts, group)
 
     # 4. group by local expert, FFN, restore order
     perm, per_expert = segment_recv(all_expert_counts)
-    perm_d = perm.to(input.device)
+    perm_d = perm.to(dev)
     grouped = recv_rows.index_select(0, perm_d) if perm.numel() else recv_rows
     out_grouped = torch.empty_like(grouped)
     off = 0
@@ -203,19 +227,16 @@ def moe_forward_ep(input, gate_weights, expert_weights, group=None):
     result_rows = torch.empty_like(recv_rows)
     if perm.numel():
         result_rows.index_copy_(0, perm_d, out_grouped)
-    else:
-        result_rows = out_grouped
 
     # 5. return exchange (reverse splits), combine at source
-    returned = exchange_rows(result_rows, recv_counts, plan.send_counts, group)
+    returned = exchange_rows(result_rows, recv_counts, send_counts_cpu, group)
     n = returned.shape[0]
-    tok_d = plan.token_idx.to(torch.int32).to(input.device)
+    tok_d = plan.token_idx.to(torch.int32)
     if k > 1:
-        probs = gate_out[plan.token_idx.to(input.device),
-                         plan.expert_of_row.to(input.device)].float()
-        scale = (probs / plan.prob_sum.to(input.device)).contiguous()
+        probs = gate_out[plan.token_idx, plan.expert_of_row].float()
+        scale = (probs / plan.prob_sum).contiguous()
     else:
-        scale = torch.ones(n, dtype=torch.float32, device=input.device)
+        scale = torch.ones(n, dtype=torch.float32, device=dev)
     _ext.check(lib.fm_combine(
         ctypes.c_void_p(stream),
         ctypes.c_void_p(returned.data_ptr()),

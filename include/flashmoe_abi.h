@@ -135,6 +135,13 @@ int fm_moe_forward_phased(void* stream, const void* x, const void* gate_w,
                           const void* b_dn, void* gate_out, void* moe_out,
                           int64_t S, float ms[4]);
 
+/* GPU-resident routing export for the EP pipeline: routed_dev = u32[E]
+ * clipped counts (min(eC, EC)), tps_dev = u32[E][EC][2] (tokenIdx,
+ * probSum bits). Device pointers; asynchronous (no host sync) - this is
+ * what keeps the multi-GPU dispatch planning off the host critical path
+ * (replaces the reference's in-kernel decode, os/packet.cuh:288-455). */
+int fm_export_routing(void* stream, void* routed_dev, void* tps_dev);
+
 /* Training-mode auxiliary-loss accumulators (gate.cuh:273-299,763-773;
  * types.cuh:936-958): gML[e] = mean softmax prob of expert e over the
  * last forward's tokens, gMeC[e] = fraction routed to e (pre-capacity).
