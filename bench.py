@@ -78,13 +78,24 @@ def measure_roofline(moe, lib, x, gw, ew, S, H, P, E, EC, iters=10):
     flops = 4.0 * routed * H * P  # up (2*r*P*H) + down (2*r*H*P)
     gemm_ms = acc[1] + acc[2]
     achieved_tf = flops / (gemm_ms * 1e-3) / 1e12 if gemm_ms > 0 else 0.0
+    # HBM traffic of the GEMM pair per forward, from the committed PMC
+    # calibration (rocprofv3 TCC counters, profiles/r01_tcc_traffic.json;
+    # read side doubled per the microarch guide's FETCH calibration).
+    traffic = None
+    calib = os.path.join(REPO, "profiles", "r01_tcc_traffic.json")
+    if os.path.exists(calib):
+        try:
+            with open(calib) as f:
+                traffic = json.load(f).get("gemm_pair_total_bytes")
+        except Exception:
+            traffic = None
     return {
         "bound": "mfma",
         "achieved": round(achieved_tf, 1),
         "peak": BF16_MFMA_PEAK_TFLOPS,
         "unit": "TFLOP/s",
         "frac": round(achieved_tf / BF16_MFMA_PEAK_TFLOPS, 4),
-        "traffic": None,
+        "traffic": traffic,
         "detail": {
             "phase_ms": {"gate": round(acc[0], 4), "gemm_up": round(acc[1], 4),
                          "gemm_down_combine": round(acc[2], 4),
