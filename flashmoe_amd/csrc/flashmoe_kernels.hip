@@ -711,22 +711,34 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
 #pragma unroll
     for (int j = 0; j < NF; ++j) accv[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
+  // one-barrier 2-phase schedule (guide T3 minimum recipe): stage tile
+  // t+1 into the other buffer FIRST, compute tile t, then ONE
+  // vmcnt(0)+barrier per tile (the drain is cheap: the stage had the
+  // whole compute phase to land).
   const int nK = K / BK;
   stage(0, 0);
-  if (nK > 1) stage(BK, 1);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
   for (int t = 0; t < nK; ++t) {
-    // wait THIS tile's glds (t+1's stay in flight), then rendezvous
-    if (t + 1 < nK) {
-      asm volatile("s_waitcnt vmcnt(%0)" ::"i"(GPT) : "memory");
-    } else {
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    }
-    __builtin_amdgcn_s_barrier();
+    // staggered staging: waves 0-3 stage tile t+1 before their first
+    // MFMA half, waves 4-7 between the halves - so on each SIMD the
+    // paired waves run complementary {stage issue | MFMA} segments
+    // instead of lockstep (microarch guide, two-waves-per-SIMD)
+    const bool stageNow = t + 1 < nK;
+    // stagger only pays on the wide (NF==4) stream; the BN=128 kernel's
+    // shorter MFMA halves lose more to the mid-stream insertion (measured)
+    const bool late = (NF == 4) && wave >= 4;
+    if (stageNow && !late) stage((t + 1) * BK, (t + 1) & 1);
     const bf16* Al = Abase + (t & 1) * BM * BK;
     const bf16* Bl = Bbase + (t & 1) * BN * BK;
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int s = 0; s < 2; ++s) {
+      if (s == 1 && late && stageNow) {
+        __builtin_amdgcn_s_setprio(0);
+        stage((t + 1) * BK, (t + 1) & 1);
+        __builtin_amdgcn_s_setprio(1);
+      }
       bf16x8 af[8], bfr[NF];
       const int rl = lane & 15;
       const int cbase = 4 * s + (lane >> 4);
@@ -754,8 +766,8 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
               af[mi], bfr[ni], accv[mi][ni], 0, 0, 0);
     }
     __builtin_amdgcn_s_setprio(0);
-    __builtin_amdgcn_s_barrier();  // everyone done reading buf (t&1)
-    if (t + 2 < nK) stage((t + 2) * BK, t & 1);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // t+1 landed
+    __builtin_amdgcn_s_barrier();  // readers done AND next tile ready
   }
 
   // epilogue (same semantics as k_group_gemm_bf16)
