@@ -251,6 +251,8 @@ struct GemmArgs {
   int expertOffset;     // global expert id of blockIdx.z==0
   int nRows;            // packed-rows mode row count
   int H;                // row stride of x / O32 / moe_out
+  int splitK;           // K-split factor (PHASE 1 multi only: the fp32
+                        // atomicAdd combine makes split-K partials free)
 };
 
 // ---------------------------------------------------------------------------
@@ -643,7 +645,10 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
   const int xcd = lin % 8, pos = lin / 8;
   const int swz =
       (xcd < rx ? xcd * (qx + 1) : rx * (qx + 1) + (xcd - rx) * qx) + pos;
-  const int e = swz / (mT * nT);
+  const int sk = a.splitK > 0 ? a.splitK : 1;
+  const int eEff = swz / (mT * nT);  // in [0, E*splitK)
+  const int e = eEff % (gridDim.z / sk);
+  const int ksplit = eEff / (gridDim.z / sk);
   const int rem = swz % (mT * nT);
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -651,6 +656,8 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
   const int m0 = (rem % mT) * BM;
   const int n0 = (rem / mT) * BN;
   const int K = a.K, N = a.N;
+  const int kLen = K / sk;            // this split's K range
+  const int kStart = ksplit * kLen;
 
   const TPS* tpsE = a.tokenIds ? a.tokenIds + (size_t)e * a.pEC : nullptr;
   if (tid == 0)
@@ -683,12 +690,12 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
     const int row = (wave * GPW_A + i) * 8 + grow8;
     const size_t arow = (PHASE == 0) ? (size_t)sTps[row].tokenIdx
                                      : (size_t)min(m0 + row, mCap - 1);
-    aSrc[i] = Ag + aBase + arow * aRowStride + schunk * 8;
+    aSrc[i] = Ag + aBase + arow * aRowStride + kStart + schunk * 8;
   }
 #pragma unroll
   for (int i = 0; i < GPW_B; ++i) {
     const int row = (wave * GPW_B + i) * 8 + grow8;
-    bSrc[i] = Bg + (size_t)min(n0 + row, N - 1) * K + schunk * 8;
+    bSrc[i] = Bg + (size_t)min(n0 + row, N - 1) * K + kStart + schunk * 8;
   }
 
   auto stage = [&](int kt, int buf) {
@@ -715,7 +722,7 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
   // t+1 into the other buffer FIRST, compute tile t, then ONE
   // vmcnt(0)+barrier per tile (the drain is cheap: the stage had the
   // whole compute phase to land).
-  const int nK = K / BK;
+  const int nK = kLen / BK;
   stage(0, 0);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
@@ -794,10 +801,12 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
 #pragma unroll
   for (int ni = 0; ni < NF; ++ni) bv[ni] = 0.f;
   if constexpr (HAS_BIAS) {
+    if (ksplit == 0) {  // split partials must add the bias exactly once
 #pragma unroll
-    for (int ni = 0; ni < NF; ++ni) {
-      const int col = n0 + wc * (BN / 4) + ni * 16 + cl;
-      if (col < N) bv[ni] = toF(reinterpret_cast<const bf16*>(a.bias)[col]);
+      for (int ni = 0; ni < NF; ++ni) {
+        const int col = n0 + wc * (BN / 4) + ni * 16 + cl;
+        if (col < N) bv[ni] = toF(reinterpret_cast<const bf16*>(a.bias)[col]);
+      }
     }
   }
 #pragma unroll
@@ -1209,28 +1218,44 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
   const int N = a.N;
   if (g.esz == 2) {
     // tile selection: prefer the deep-pipelined 256-row kernel when the
-    // grid still fills the 256 CUs at 1 block/CU
+    // grid still fills the 256 CUs at 1 block/CU. When BN=256 alone
+    // cannot fill the chip (small N, e.g. the down GEMM at H=1024), the
+    // fp32-atomic combine epilogue makes K-split partials free, so
+    // split-K restores occupancy at BN=256's lower staging traffic.
     const int b256 = DIVUP(M, 256) * DIVUP(N, 256) * nE;
     const int b128n = DIVUP(M, 256) * DIVUP(N, 128) * nE;
     int mode;  // 0: big BN=256, 1: big BN=128, 2: small 128x128
-    if (M >= 256 && b256 >= 256) mode = 0;
-    else if (M >= 256 && b128n >= 256) mode = 1;
-    else mode = 2;
+    int skf = 1;
+    if (M >= 256 && b256 >= 256) {
+      mode = 0;
+    } else if (M >= 256) {
+      // NOTE: split-K via the atomic combine epilogue is wired up
+      // (a.splitK) but DISABLED by default: at config 2 it doubled the
+      // 8.3M fp32 atomics and cost more (+13 us) than the BN=256
+      // staging saving bought (profiles/r01). Revisit for huge-K shapes.
+      if (skf > 1) mode = 0;
+      else if (b128n >= 256) mode = 1;
+      else mode = 2;
+    } else {
+      mode = 2;
+    }
+    GemmArgs aa = a;
+    aa.splitK = skf;
     dim3 block(mode == 2 ? 256 : 512);
     dim3 grid(DIVUP(M, mode == 2 ? 128 : 256),
-              DIVUP(N, mode == 1 ? 128 : (mode == 0 ? 256 : 128)), nE);
+              DIVUP(N, mode == 1 ? 128 : (mode == 0 ? 256 : 128)), nE * skf);
     const int sel = phase * 4 + act * 2 + (hasBias ? 1 : 0);
 #define GG_CASE(PH, AC, HB)                                                   \
     do {                                                                      \
       if (mode == 0)                                                          \
         hipLaunchKernelGGL((k_group_gemm_bf16_big<PH, AC, HB, 256>), grid,    \
-                           block, 0, st, a);                                  \
+                           block, 0, st, aa);                                 \
       else if (mode == 1)                                                     \
         hipLaunchKernelGGL((k_group_gemm_bf16_big<PH, AC, HB, 128>), grid,    \
-                           block, 0, st, a);                                  \
+                           block, 0, st, aa);                                 \
       else                                                                    \
         hipLaunchKernelGGL((k_group_gemm_bf16<PH, AC, HB>), grid, block, 0,   \
-                           st, a);                                            \
+                           st, aa);                                           \
     } while (0)
     switch (sel) {
       case 0: GG_CASE(0, 0, false); break;
