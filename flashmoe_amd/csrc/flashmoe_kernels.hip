@@ -32,13 +32,36 @@ struct __align__(8) TPS {
 };
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(8))) _Float16 halfx8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 typedef __attribute__((ext_vector_type(4))) uint32_t u32x4;
 
+using fp16 = __half;
+
 __device__ __forceinline__ float toF(float v) { return v; }
 __device__ __forceinline__ float toF(bf16 v) { return __bfloat162float(v); }
+__device__ __forceinline__ float toF(fp16 v) { return __half2float(v); }
 __device__ __forceinline__ void fromF(float v, float& o) { o = v; }
 __device__ __forceinline__ void fromF(float v, bf16& o) { o = __float2bfloat16(v); }
+__device__ __forceinline__ void fromF(float v, fp16& o) { o = __float2half(v); }
+
+// element traits for the low-precision MFMA GEMM kernels (bf16 / fp16;
+// both K=32 16x16 MFMA shapes share operand/result layouts on gfx950)
+template <typename ET> struct ETr;
+template <> struct ETr<bf16> {
+  using vec8 = bf16x8;
+  static __device__ __forceinline__ f32x4 mfma(vec8 a, vec8 b, f32x4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+  }
+  static __device__ __forceinline__ bf16 fromf(float v) { return __float2bfloat16(v); }
+};
+template <> struct ETr<fp16> {
+  using vec8 = halfx8;
+  static __device__ __forceinline__ f32x4 mfma(vec8 a, vec8 b, f32x4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b, c, 0, 0, 0);
+  }
+  static __device__ __forceinline__ fp16 fromf(float v) { return __float2half(v); }
+};
 
 __device__ __forceinline__ float applyAct(float v, int act) {
   // hidden_act 0: ReLU, 1: exact-erf GELU (cutlass epilogue::thread::{ReLU,GELU},
@@ -47,213 +70,9 @@ __device__ __forceinline__ float applyAct(float v, int act) {
   return 0.5f * v * (1.0f + erff(v * 0.70710678118654752f));
 }
 
+template <typename T> constexpr bool cuda_false() { return false; }
+
 #define DIVUP(a, b) (((a) + (b) - 1) / (b))
-
-// ---------------------------------------------------------------------------
-// Gate kernel (reference: moe/gate.cuh:474-720 singleBlock semantics; the
-// multiBlock ring for E>64 computes the same values, gate.cuh:93-468).
-//
-// One block per 128-token tile, 256 threads.  Per tile:
-//   1. logits[128][E] fp32 via LDS-staged chunked dot products (2 thr/token)
-//   2. per token: online softmax over E (padded cols excluded,
-//      gate.cuh:578-585); iterative strict-> first-index-wins top-k on the
-//      LOGITS (monotone-equivalent to probs, gate.cuh:662-668); mCw = sum of
-//      selected probs (gate.cuh:669); write gate_out row (probs, Element)
-//   3. per expert: intra-tile count in token order + one atomicAdd(eC) base
-//      (gate.cuh:688-716 - inter-tile order nondeterministic, as reference)
-//   4. per token: tokenIds[e][base+local] = {token, mCw} if slot < EC
-//      (capacity, gate.cuh:711-715)
-// Supports E <= 128 this round (covers BASELINE configs 1-4; E=256 next).
-// ---------------------------------------------------------------------------
-
-template <typename T, int K>
-__global__ __launch_bounds__(256) void k_gate(
-    const T* __restrict__ x, const T* __restrict__ gate_w,
-    T* __restrict__ gate_out, TPS* __restrict__ tokenIds,
-    uint32_t* __restrict__ eC, int S, int H, int E, int PX, int EC, int pEC) {
-  constexpr int BM = 128;
-  constexpr int BK = 64;
-  constexpr int RPAD = 8;  // row pad (elements) to break LDS bank alignment
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  // arena layout (host mirrors gate_lds_bytes()):
-  float* logits = reinterpret_cast<float*>(smem);            // [BM][E+1]
-  T* aCh = reinterpret_cast<T*>(logits + BM * (E + 1));      // [BM][BK+RPAD]
-  T* gCh = aCh + BM * (BK + RPAD);                           // [E][BK+RPAD]
-  // post-logits arena reuses the A-chunk region:
-  uint16_t* sel = reinterpret_cast<uint16_t*>(aCh);          // [BM][K]
-  uint16_t* localIdx = sel + BM * K;                         // [BM][K]
-  uint32_t* base = reinterpret_cast<uint32_t*>(localIdx + BM * K);  // [E]
-
-  const int tid = threadIdx.x;
-  const int m0 = blockIdx.x * BM;
-  const int tok = tid & (BM - 1);
-  const int half = tid >> 7;              // 2 threads per token
-  const int E2 = (E + 1) / 2;
-  const int e0 = half * E2;
-  const int e1 = min(e0 + E2, E);
-
-  // zero the logits accumulator (accumulated in LDS across K chunks;
-  // a register array would be runtime-indexed -> scratch, guide rule 20)
-  for (int i = tid; i < BM * (E + 1); i += 256) logits[i] = 0.0f;
-
-  const int LDA = BK + RPAD;
-  for (int kc = 0; kc < H; kc += BK) {
-    // stage A chunk [BM][BK] and G chunk [E][BK], 16B units, coalesced
-    constexpr int EPU = 16 / sizeof(T);  // elements per 16B unit
-    {
-      const int unitsA = BM * BK / EPU;
-      for (int u = tid; u < unitsA; u += 256) {
-        const int row = u / (BK / EPU);
-        const int cu = u % (BK / EPU);
-        const T* src = x + (size_t)(m0 + row) * H + kc + cu * EPU;
-        T* dst = aCh + row * LDA + cu * EPU;
-        *reinterpret_cast<u32x4*>(dst) = *reinterpret_cast<const u32x4*>(src);
-      }
-      const int unitsG = E * BK / EPU;
-      for (int u = tid; u < unitsG; u += 256) {
-        const int row = u / (BK / EPU);
-        const int cu = u % (BK / EPU);
-        const T* src = gate_w + (size_t)row * H + kc + cu * EPU;
-        T* dst = gCh + row * LDA + cu * EPU;
-        *reinterpret_cast<u32x4*>(dst) = *reinterpret_cast<const u32x4*>(src);
-      }
-    }
-    __syncthreads();
-    // hoist this token's A chunk to registers as 16B vectors, then run
-    // every owned expert against it with vector LDS reads (scalar bf16
-    // LDS reads were the gate's dominant cost, profiles/r01)
-    const T* arow = aCh + tok * LDA;
-    u32x4 a8[BK * sizeof(T) / 16];
-#pragma unroll
-    for (int jb = 0; jb < BK * (int)sizeof(T) / 16; ++jb)
-      a8[jb] = *reinterpret_cast<const u32x4*>(&arow[jb * (16 / sizeof(T))]);
-    for (int e = e0; e < e1; ++e) {
-      const T* grow = gCh + e * LDA;
-      float s = 0.0f;
-#pragma unroll
-      for (int jb = 0; jb < BK * (int)sizeof(T) / 16; ++jb) {
-        const u32x4 g8 = *reinterpret_cast<const u32x4*>(&grow[jb * (16 / sizeof(T))]);
-#pragma unroll
-        for (int w = 0; w < 4; ++w) {
-          const uint32_t aw = a8[jb][w], gw = g8[w];
-          if constexpr (sizeof(T) == 2) {
-            const float2 av = __bfloat1622float2(
-                *reinterpret_cast<const __hip_bfloat162*>(&aw));
-            const float2 gv = __bfloat1622float2(
-                *reinterpret_cast<const __hip_bfloat162*>(&gw));
-            s = fmaf(av.x, gv.x, s);
-            s = fmaf(av.y, gv.y, s);
-          } else {
-            s = fmaf(__uint_as_float(aw), __uint_as_float(gw), s);
-          }
-        }
-      }
-      logits[tok * (E + 1) + e] += s;
-    }
-    __syncthreads();
-  }
-
-  // phase 2: softmax + top-k (thread t < 128 owns token t)
-  float mCw = 0.0f;
-  uint16_t mySel[K];
-  if (tid < BM) {
-    const float* lrow = logits + tid * (E + 1);
-    float m = -INFINITY;
-    for (int e = 0; e < E; ++e) m = fmaxf(m, lrow[e]);
-    float d = 0.0f;
-    for (int e = 0; e < E; ++e) d += __expf(lrow[e] - m);
-    const float inv_d = 1.0f / d;
-    T* grow = gate_out + (size_t)(m0 + tid) * PX;
-    for (int e = 0; e < E; ++e) fromF(__expf(lrow[e] - m) * inv_d, grow[e]);
-    for (int e = E; e < PX; ++e) fromF(0.0f, grow[e]);
-    // iterative argmax on logits, strict >, first index wins
-    uint64_t taken_lo = 0, taken_hi = 0;  // E <= 128
-#pragma unroll
-    for (int i = 0; i < K; ++i) {
-      float sV = -INFINITY;
-      int sIdx = 0;
-      for (int e = 0; e < E; ++e) {
-        const bool taken = (e < 64) ? ((taken_lo >> e) & 1)
-                                    : ((taken_hi >> (e - 64)) & 1);
-        if (!taken && lrow[e] > sV) { sV = lrow[e]; sIdx = e; }
-      }
-      if (sIdx < 64) taken_lo |= 1ull << sIdx; else taken_hi |= 1ull << (sIdx - 64);
-      mySel[i] = (uint16_t)sIdx;
-      mCw += __expf(sV - m) * inv_d;
-    }
-  }
-  __syncthreads();  // logits no longer needed; sel arena reuses A-chunk space
-  if (tid < BM) {
-#pragma unroll
-    for (int i = 0; i < K; ++i) sel[tid * K + i] = mySel[i];
-  }
-  __syncthreads();
-
-  // phase 3: per-expert intra-tile ordering + global base via atomicAdd
-  if (tid < E) {
-    uint32_t cnt = 0;
-    for (int mj = 0; mj < BM * K; ++mj) {
-      if (sel[mj] == tid) localIdx[mj] = (uint16_t)cnt++;
-    }
-    base[tid] = atomicAdd(eC + tid, cnt);
-  }
-  __syncthreads();
-
-  // phase 4: write token slots (capacity-clipped)
-  if (tid < BM) {
-#pragma unroll
-    for (int i = 0; i < K; ++i) {
-      const int e = mySel[i];
-      const uint32_t slot = base[e] + localIdx[tid * K + i];
-      if (slot < (uint32_t)EC) {
-        tokenIds[(size_t)e * pEC + slot] = TPS{(uint32_t)(m0 + tid), mCw};
-      }
-    }
-  }
-}
-
-// ---------------------------------------------------------------------------
-// Grouped expert GEMM, bf16 MFMA (reference: os/processor/processor.cuh
-// fGET preGEMM :339-468 / postGEMM :711-751, gemm.cuh FAA epilogue, combine
-// :44-205). 128x128 tile, 4 waves (2x2), 64x64 per wave as 4x4
-// v_mfma_f32_16x16x32_bf16 fragments, BK=64, LDS-staged NT operands.
-//
-// PHASE 0 (up):   A = x rows gathered via tokenIds[e]; B = Wup[e] [P,H];
-//                 epilogue act(acc + b_up) -> xM[e]
-// PHASE 1 (down): A = xM[e]; B = Wdn[e] viewed [H,P] (reference
-//                 reinterpretation, moe.cuh:114-116); epilogue
-//                 z = acc + b_dn; k>1: atomicAdd(O32, gate_out*z/probSum)
-//                 (processor.cuh:126-168); k==1: unscaled store to moe_out
-//                 (processor.cuh:173-204)
-// PHASE 2 (down-direct, EP path): like 1 but plain store to out_rows[m].
-// If tokenIds == nullptr, the row gather is identity and routed = n_rows
-// (packed-rows mode for fm_expert_ffn).
-// ---------------------------------------------------------------------------
-
-struct GemmArgs {
-  const void* A;        // phase0: x [S,H]; phase1/2: xM_e rows [pEC,K]
-  const void* B;        // weight rows [N,K] (K contiguous)
-  const void* bias;     // [N] or null
-  void* out;            // phase0: xM_e [pEC,P]; phase2: out_rows
-  float* O32;           // phase1 k>1
-  void* moe_out;        // phase1 k==1
-  const void* gate_out; // [S,PX]
-  const TPS* tokenIds;  // [E,pEC] (this expert's row = tokenIds + e*pEC)
-  const uint32_t* eC;
-  long long strideAExpert;  // elements between experts in A (phase1: pEC*K)
-  long long strideBExpert;  // elements between experts in B (2*P*H)
-  long long strideOExpert;  // elements between experts in out
-  int K;                // reduction dim
-  int N;                // output dim
-  int EC, pEC, PX;
-  int topk;
-  int act;
-  int expertOffset;     // global expert id of blockIdx.z==0
-  int nRows;            // packed-rows mode row count
-  int H;                // row stride of x / O32 / moe_out
-  int splitK;           // K-split factor (PHASE 1 multi only: the fp32
-                        // atomicAdd combine makes split-K partials free)
-};
 
 // ---------------------------------------------------------------------------
 // Split gate (v2): the single-kernel gate ran on only S/128 blocks (32 for
@@ -325,11 +144,16 @@ __global__ __launch_bounds__(256) void k_gate_logits(
 #pragma unroll
         for (int w = 0; w < 4; ++w) {
           const uint32_t aw = a8[jb][w], gw = g8[w];
-          if constexpr (sizeof(T) == 2) {
+          if constexpr (__is_same(T, bf16)) {
             const float2 av = __bfloat1622float2(
                 *reinterpret_cast<const __hip_bfloat162*>(&aw));
             const float2 gv = __bfloat1622float2(
                 *reinterpret_cast<const __hip_bfloat162*>(&gw));
+            sAcc = fmaf(av.x, gv.x, sAcc);
+            sAcc = fmaf(av.y, gv.y, sAcc);
+          } else if constexpr (__is_same(T, fp16)) {
+            const float2 av = __half22float2(*reinterpret_cast<const __half2*>(&aw));
+            const float2 gv = __half22float2(*reinterpret_cast<const __half2*>(&gw));
             sAcc = fmaf(av.x, gv.x, sAcc);
             sAcc = fmaf(av.y, gv.y, sAcc);
           } else {
@@ -436,8 +260,9 @@ __global__ __launch_bounds__(256) void k_gate_route(
 typedef __attribute__((address_space(1))) const uint32_t gas_u32;
 typedef __attribute__((address_space(3))) uint32_t las_u32;
 
-template <int PHASE, int ACT, bool HAS_BIAS>
+template <typename ET, int PHASE, int ACT, bool HAS_BIAS>
 __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
+  using vec8 = typename ETr<ET>::vec8;
   constexpr int BM = 128, BN = 128, BK = 64;
   // ONE shared arena (a second __shared__ object would force a vmcnt(0)
   // drain before every ds_read beside glds - guide par.5 trap 4a).
@@ -445,8 +270,8 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
   // bank swizzle lives on the SOURCE chunk index and the fragment-read
   // address (rule 21): chunk' = chunk ^ (row & 7).
   __shared__ __attribute__((aligned(16))) char smem[2 * BM * BK * 2 + BM * 8 + 16];
-  bf16* Alds = reinterpret_cast<bf16*>(smem);
-  bf16* Blds = Alds + BM * BK;
+  ET* Alds = reinterpret_cast<ET*>(smem);
+  ET* Blds = Alds + BM * BK;
   TPS* sTps = reinterpret_cast<TPS*>(Blds + BN * BK);
   uint32_t* sRouted = reinterpret_cast<uint32_t*>(sTps + BM);
 
@@ -475,9 +300,9 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
   }
   __syncthreads();
 
-  const bf16* __restrict__ Ag = reinterpret_cast<const bf16*>(a.A);
-  const bf16* __restrict__ Bg =
-      reinterpret_cast<const bf16*>(a.B) + (size_t)e * a.strideBExpert;
+  const ET* __restrict__ Ag = reinterpret_cast<const ET*>(a.A);
+  const ET* __restrict__ Bg =
+      reinterpret_cast<const ET*>(a.B) + (size_t)e * a.strideBExpert;
 
   // accumulators: wave (wr,wc) owns the 64x64 subtile at (wr*64, wc*64)
   const int wr = wave >> 1, wc = wave & 1;
@@ -501,38 +326,37 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
       const int row = grp * 8 + grow8;
       const size_t arow = (PHASE == 0) ? (size_t)sTps[row].tokenIdx
                                        : (size_t)(m0 + row);
-      const bf16* asrc = Ag + aBase + arow * aRowStride + kt + schunk * 8;
+      const ET* asrc = Ag + aBase + arow * aRowStride + kt + schunk * 8;
       __builtin_amdgcn_global_load_lds(
           (gas_u32*)asrc, (las_u32*)(Alds + grp * 512), 16, 0, 0);
       const int brow = min(n0 + row, N - 1);
-      const bf16* bsrc = Bg + (size_t)brow * K + kt + schunk * 8;
+      const ET* bsrc = Bg + (size_t)brow * K + kt + schunk * 8;
       __builtin_amdgcn_global_load_lds(
           (gas_u32*)bsrc, (las_u32*)(Blds + grp * 512), 16, 0, 0);
     }
     __syncthreads();  // carries the vmcnt(0) glds drain (guide par.5)
 #pragma unroll
     for (int s = 0; s < 2; ++s) {  // two K=32 MFMA steps per tile
-      bf16x8 af[4], bf[4];
+      vec8 af[4], bf[4];
       const int rl = lane & 15;
       const int cbase = 4 * s + (lane >> 4);  // 16B chunk before swizzle
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi) {
         const int R = wr * 64 + mi * 16 + rl;
-        af[mi] = *reinterpret_cast<const bf16x8*>(
+        af[mi] = *reinterpret_cast<const vec8*>(
             &Alds[R * BK + ((cbase ^ (R & 7)) * 8)]);
       }
 #pragma unroll
       for (int ni = 0; ni < 4; ++ni) {
         const int R = wc * 64 + ni * 16 + rl;
-        bf[ni] = *reinterpret_cast<const bf16x8*>(
+        bf[ni] = *reinterpret_cast<const vec8*>(
             &Blds[R * BK + ((cbase ^ (R & 7)) * 8)]);
       }
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
         for (int ni = 0; ni < 4; ++ni)
-          accv[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af[mi], bf[ni], accv[mi][ni], 0, 0, 0);
+          accv[mi][ni] = ETr<ET>::mfma(af[mi], bf[ni], accv[mi][ni]);
     }
     __syncthreads();
   }
@@ -552,7 +376,7 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
         const TPS tp = sTps[tid];
         float sc = 0.0f;
         if ((uint32_t)(m0 + tid) < routed)
-          sc = toF(reinterpret_cast<const bf16*>(
+          sc = toF(reinterpret_cast<const ET*>(
                    a.gate_out)[(size_t)tp.tokenIdx * a.PX +
                                a.expertOffset + e]) / tp.probSum;
         sScale[tid] = sc;
@@ -565,7 +389,7 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
 #pragma unroll
     for (int ni = 0; ni < 4; ++ni) {
       const int col = n0 + wc * 64 + ni * 16 + cl;
-      if (col < N) bv[ni] = toF(reinterpret_cast<const bf16*>(a.bias)[col]);
+      if (col < N) bv[ni] = toF(reinterpret_cast<const ET*>(a.bias)[col]);
     }
   }
 #pragma unroll
@@ -585,20 +409,20 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
         if constexpr (PHASE == 0) {
           v = (ACT == 0) ? fmaxf(v, 0.0f)
                          : 0.5f * v * (1.0f + erff(v * 0.70710678118654752f));
-          reinterpret_cast<bf16*>(a.out)[(size_t)e * a.strideOExpert +
+          reinterpret_cast<ET*>(a.out)[(size_t)e * a.strideOExpert +
                                          (size_t)m * N + col] =
-              __float2bfloat16(v);
+              ETr<ET>::fromf(v);
         } else if constexpr (PHASE == 1) {
           if (multi) {
             atomicAdd(&a.O32[(size_t)tp.tokenIdx * a.H + col], v * rowScale);
           } else {
-            reinterpret_cast<bf16*>(
+            reinterpret_cast<ET*>(
                 a.moe_out)[(size_t)tp.tokenIdx * a.H + col] =
-                __float2bfloat16(v);
+                ETr<ET>::fromf(v);
           }
         } else {  // PHASE 2: packed-rows direct output
-          reinterpret_cast<bf16*>(a.out)[(size_t)m * N + col] =
-              __float2bfloat16(v);
+          reinterpret_cast<ET*>(a.out)[(size_t)m * N + col] =
+              ETr<ET>::fromf(v);
         }
       }
     }
@@ -616,8 +440,9 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
 // Same phase semantics/epilogues as k_group_gemm_bf16.
 // ---------------------------------------------------------------------------
 
-template <int PHASE, int ACT, bool HAS_BIAS, int BN>
+template <typename ET, int PHASE, int ACT, bool HAS_BIAS, int BN>
 __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
+  using vec8 = typename ETr<ET>::vec8;
   constexpr int BM = 256, BK = 64;
   constexpr int NF = BN / 64;           // B fragments per wave (4 or 2)
   constexpr int AGRP = BM * BK * 2 / 1024;  // glds 1KiB groups per A tile (32)
@@ -626,8 +451,8 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
   constexpr int GPT = GPW_A + GPW_B;    // glds per wave per K-tile
   __shared__ __attribute__((aligned(16))) char smem[
       2 * BM * BK * 2 + 2 * BN * BK * 2 + BM * 8 + 16];
-  bf16* Abase = reinterpret_cast<bf16*>(smem);          // 2 x [BM][BK]
-  bf16* Bbase = Abase + 2 * BM * BK;                    // 2 x [BN][BK]
+  ET* Abase = reinterpret_cast<ET*>(smem);          // 2 x [BM][BK]
+  ET* Bbase = Abase + 2 * BM * BK;                    // 2 x [BN][BK]
   TPS* sTps = reinterpret_cast<TPS*>(Bbase + 2 * BN * BK);
   uint32_t* sRouted = reinterpret_cast<uint32_t*>(sTps + BM);
 
@@ -674,17 +499,17 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
   }
   __syncthreads();
 
-  const bf16* __restrict__ Ag = reinterpret_cast<const bf16*>(a.A);
-  const bf16* __restrict__ Bg =
-      reinterpret_cast<const bf16*>(a.B) + (size_t)e * a.strideBExpert;
+  const ET* __restrict__ Ag = reinterpret_cast<const ET*>(a.A);
+  const ET* __restrict__ Bg =
+      reinterpret_cast<const ET*>(a.B) + (size_t)e * a.strideBExpert;
 
   // per-lane glds source bases, hoisted out of the K loop
   const int grow8 = lane >> 3;
   const int schunk = (lane & 7) ^ grow8;
   const int aRowStride = (PHASE == 0) ? a.H : K;
   const size_t aBase = (PHASE == 0) ? 0 : (size_t)e * a.strideAExpert;
-  const bf16* aSrc[GPW_A];
-  const bf16* bSrc[GPW_B];
+  const ET* aSrc[GPW_A];
+  const ET* bSrc[GPW_B];
 #pragma unroll
   for (int i = 0; i < GPW_A; ++i) {
     const int row = (wave * GPW_A + i) * 8 + grow8;
@@ -736,8 +561,8 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
     // shorter MFMA halves lose more to the mid-stream insertion (measured)
     const bool late = (NF == 4) && wave >= 4;
     if (stageNow && !late) stage((t + 1) * BK, (t + 1) & 1);
-    const bf16* Al = Abase + (t & 1) * BM * BK;
-    const bf16* Bl = Bbase + (t & 1) * BN * BK;
+    const ET* Al = Abase + (t & 1) * BM * BK;
+    const ET* Bl = Bbase + (t & 1) * BN * BK;
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int s = 0; s < 2; ++s) {
@@ -746,7 +571,7 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
         stage((t + 1) * BK, (t + 1) & 1);
         __builtin_amdgcn_s_setprio(1);
       }
-      bf16x8 af[8], bfr[NF];
+      vec8 af[8], bfr[NF];
       const int rl = lane & 15;
       const int cbase = 4 * s + (lane >> 4);
       // read the B fragments FIRST: the first MFMA (af[0] x bfr[*])
@@ -756,13 +581,13 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
 #pragma unroll
       for (int ni = 0; ni < NF; ++ni) {
         const int R = wc * (BN / 4) + ni * 16 + rl;
-        bfr[ni] = *reinterpret_cast<const bf16x8*>(
+        bfr[ni] = *reinterpret_cast<const vec8*>(
             &Bl[R * BK + ((cbase ^ (R & 7)) * 8)]);
       }
 #pragma unroll
       for (int mi = 0; mi < 8; ++mi) {
         const int R = wr * 128 + mi * 16 + rl;
-        af[mi] = *reinterpret_cast<const bf16x8*>(
+        af[mi] = *reinterpret_cast<const vec8*>(
             &Al[R * BK + ((cbase ^ (R & 7)) * 8)]);
       }
 #pragma unroll
@@ -789,7 +614,7 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
         const TPS tp = sTps[tid];
         float sc = 0.0f;
         if ((uint32_t)(m0 + tid) < routed)
-          sc = toF(reinterpret_cast<const bf16*>(
+          sc = toF(reinterpret_cast<const ET*>(
                    a.gate_out)[(size_t)tp.tokenIdx * a.PX +
                                a.expertOffset + e]) / tp.probSum;
         sScale[tid] = sc;
@@ -805,7 +630,7 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
 #pragma unroll
       for (int ni = 0; ni < NF; ++ni) {
         const int col = n0 + wc * (BN / 4) + ni * 16 + cl;
-        if (col < N) bv[ni] = toF(reinterpret_cast<const bf16*>(a.bias)[col]);
+        if (col < N) bv[ni] = toF(reinterpret_cast<const ET*>(a.bias)[col]);
       }
     }
   }
@@ -826,20 +651,20 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
         if constexpr (PHASE == 0) {
           v = (ACT == 0) ? fmaxf(v, 0.0f)
                          : 0.5f * v * (1.0f + erff(v * 0.70710678118654752f));
-          reinterpret_cast<bf16*>(a.out)[(size_t)e * a.strideOExpert +
+          reinterpret_cast<ET*>(a.out)[(size_t)e * a.strideOExpert +
                                          (size_t)m * N + col] =
-              __float2bfloat16(v);
+              ETr<ET>::fromf(v);
         } else if constexpr (PHASE == 1) {
           if (multi) {
             atomicAdd(&a.O32[(size_t)tp.tokenIdx * a.H + col], v * rowScale);
           } else {
-            reinterpret_cast<bf16*>(
+            reinterpret_cast<ET*>(
                 a.moe_out)[(size_t)tp.tokenIdx * a.H + col] =
-                __float2bfloat16(v);
+                ETr<ET>::fromf(v);
           }
         } else {
-          reinterpret_cast<bf16*>(a.out)[(size_t)m * N + col] =
-              __float2bfloat16(v);
+          reinterpret_cast<ET*>(a.out)[(size_t)m * N + col] =
+              ETr<ET>::fromf(v);
         }
       }
     }
@@ -1124,6 +949,9 @@ int launch_gate(hipStream_t st, const void* x, const void* gate_w,
   if (g.cfg.dtype == 2) {
     GATE_LOGITS(bf16);
     GATE_K(bf16)
+  } else if (g.cfg.dtype == 3) {
+    GATE_LOGITS(fp16);
+    GATE_K(fp16)
   } else {
     GATE_LOGITS(float);
     GATE_K(float)
@@ -1163,8 +991,8 @@ int fm_initialize(const fm_config* cfg, int rank, int world_size) {
   g.pEC = DIVUP(g.EC, 128) * 128;
   switch (cfg->dtype) {
     case 0: case 1: g.esz = 4; break;
-    case 2: g.esz = 2; break;
-    default: setErr("dtype not supported this round (fp32/bf16)");
+    case 2: case 3: g.esz = 2; break;
+    default: setErr("unknown dtype");
              return FM_ERR_UNSUPPORTED;
   }
   if (g.E > 256) { setErr("E > 256 not supported this round"); return FM_ERR_UNSUPPORTED; }
@@ -1245,17 +1073,22 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
     dim3 grid(DIVUP(M, mode == 2 ? 128 : 256),
               DIVUP(N, mode == 1 ? 128 : (mode == 0 ? 256 : 128)), nE * skf);
     const int sel = phase * 4 + act * 2 + (hasBias ? 1 : 0);
-#define GG_CASE(PH, AC, HB)                                                   \
+#define GG_ET(ET, PH, AC, HB)                                                 \
     do {                                                                      \
       if (mode == 0)                                                          \
-        hipLaunchKernelGGL((k_group_gemm_bf16_big<PH, AC, HB, 256>), grid,    \
-                           block, 0, st, aa);                                 \
+        hipLaunchKernelGGL((k_group_gemm_bf16_big<ET, PH, AC, HB, 256>),      \
+                           grid, block, 0, st, aa);                           \
       else if (mode == 1)                                                     \
-        hipLaunchKernelGGL((k_group_gemm_bf16_big<PH, AC, HB, 128>), grid,    \
-                           block, 0, st, aa);                                 \
+        hipLaunchKernelGGL((k_group_gemm_bf16_big<ET, PH, AC, HB, 128>),      \
+                           grid, block, 0, st, aa);                           \
       else                                                                    \
-        hipLaunchKernelGGL((k_group_gemm_bf16<PH, AC, HB>), grid, block, 0,   \
-                           st, aa);                                           \
+        hipLaunchKernelGGL((k_group_gemm_bf16<ET, PH, AC, HB>), grid, block,  \
+                           0, st, aa);                                        \
+    } while (0)
+#define GG_CASE(PH, AC, HB)                                                   \
+    do {                                                                      \
+      if (g.cfg.dtype == 3) GG_ET(fp16, PH, AC, HB);                          \
+      else GG_ET(bf16, PH, AC, HB);                                           \
     } while (0)
     switch (sel) {
       case 0: GG_CASE(0, 0, false); break;
@@ -1344,7 +1177,10 @@ static int moe_forward_impl(hipStream_t st, const void* x, const void* gate_w,
   if (g.cfg.expert_top_k > 1) {
     const size_t n = (size_t)g.S * g.H;
     const int blocks = (int)std::min((size_t)2048, (size_t)DIVUP(n, (size_t)256 * 8));
-    if (g.esz == 2)
+    if (g.cfg.dtype == 3)
+      hipLaunchKernelGGL(k_cast_out<fp16>, dim3(blocks), dim3(256), 0, st,
+                         g.O32, reinterpret_cast<fp16*>(moe_out), n);
+    else if (g.esz == 2)
       hipLaunchKernelGGL(k_cast_out<bf16>, dim3(blocks), dim3(256), 0, st,
                          g.O32, reinterpret_cast<bf16*>(moe_out), n);
     else
@@ -1475,7 +1311,11 @@ int fm_combine(void* stream, const void* rows, const uint32_t* token_idx,
   if (n_rows <= 0) return FM_OK;
   // always accumulate into the fp32 buffer; caller passes scale=1.0 rows
   // for the k==1 unscaled semantics (each token appears at most once)
-  if (g.esz == 2) {
+  if (g.cfg.dtype == 3) {
+    hipLaunchKernelGGL((k_combine_rows<fp16, true>), dim3((int)n_rows),
+                       dim3(256), 0, st, reinterpret_cast<const fp16*>(rows),
+                       token_idx, scale, g.O32, (fp16*)nullptr, g.H);
+  } else if (g.esz == 2) {
     hipLaunchKernelGGL((k_combine_rows<bf16, true>), dim3((int)n_rows),
                        dim3(256), 0, st, reinterpret_cast<const bf16*>(rows),
                        token_idx, scale, g.O32, (bf16*)nullptr, g.H);
@@ -1494,7 +1334,10 @@ int fm_combine_finalize(void* stream, void* moe_out, int64_t S) {
   hipStream_t st = reinterpret_cast<hipStream_t>(stream);
   const size_t n = (size_t)g.S * g.H;
   const int blocks = (int)min((size_t)2048, DIVUP(n, (size_t)256 * 8));
-  if (g.esz == 2)
+  if (g.cfg.dtype == 3)
+    hipLaunchKernelGGL(k_cast_out<fp16>, dim3(blocks), dim3(256), 0, st, g.O32,
+                       reinterpret_cast<fp16*>(moe_out), n);
+  else if (g.esz == 2)
     hipLaunchKernelGGL(k_cast_out<bf16>, dim3(blocks), dim3(256), 0, st, g.O32,
                        reinterpret_cast<bf16*>(moe_out), n);
   else

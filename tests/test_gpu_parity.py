@@ -181,6 +181,15 @@ def test_single_expert_dense_fallback(fresh_moe):
     assert_values(out, ref["moe_out"], "bf16", "moe_out")
 
 
+def test_single_tile_fp16_top2(fresh_moe):
+    """torch_dtype 3 (fp16 element; mfma_f32_16x16x32_f16 path)."""
+    cfg, path = make_cfg(torch_dtype=3)
+    out, gate_out, ref, _ = run_pair(fresh_moe, cfg, path)
+    got = out.float().cpu().numpy()
+    scale = max(1.0, float(np.abs(ref["moe_out"]).max()))
+    assert np.allclose(got, ref["moe_out"], rtol=2e-2, atol=2e-3 * scale)
+
+
 def test_single_tile_gelu(fresh_moe):
     cfg, path = make_cfg(hidden_act=1)
     out, gate_out, ref, _ = run_pair(fresh_moe, cfg, path)
