@@ -256,6 +256,46 @@ __global__ __launch_bounds__(256) void k_gate_route(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Grouped expert GEMM (reference: os/processor/processor.cuh fGET preGEMM
+// :339-468 / postGEMM :711-751, gemm.cuh FAA epilogue, combine :44-205).
+// PHASE 0 (up):   A = x rows gathered via tokenIds[e]; B = Wup[e] [P,H];
+//                 epilogue act(acc + b_up) -> xM[e]
+// PHASE 1 (down): A = xM[e]; B = Wdn[e] viewed [H,P] (reference
+//                 reinterpretation, moe.cuh:114-116); epilogue
+//                 z = acc + b_dn; k>1: atomicAdd(O32, gate_out*z/probSum)
+//                 (processor.cuh:126-168); k==1: unscaled store to moe_out
+//                 (processor.cuh:173-204)
+// PHASE 2 (down-direct, EP path): like 1 but plain store to out_rows[m].
+// If tokenIds == nullptr the row gather is identity and routed = n_rows
+// (packed-rows mode for fm_expert_ffn).
+// ---------------------------------------------------------------------------
+
+struct GemmArgs {
+  const void* A;        // phase0: x [S,H]; phase1/2: xM_e rows [pEC,K]
+  const void* B;        // weight rows [N,K] (K contiguous)
+  const void* bias;     // [N] or null
+  void* out;            // phase0: xM_e [pEC,P]; phase2: out_rows
+  float* O32;           // phase1 k>1
+  void* moe_out;        // phase1 k==1
+  const void* gate_out; // [S,PX]
+  const TPS* tokenIds;  // [E,pEC] (this expert's row = tokenIds + e*pEC)
+  const uint32_t* eC;
+  long long strideAExpert;  // elements between experts in A (phase1: pEC*K)
+  long long strideBExpert;  // elements between experts in B (2*P*H)
+  long long strideOExpert;  // elements between experts in out
+  int K;                // reduction dim
+  int N;                // output dim
+  int EC, pEC, PX;
+  int topk;
+  int act;
+  int expertOffset;     // global expert id of blockIdx.z==0
+  int nRows;            // packed-rows mode row count
+  int H;                // row stride of x / O32 / moe_out
+  int splitK;           // K-split factor (PHASE 1 multi only: the fp32
+                        // atomicAdd combine makes split-K partials free)
+};
+
 // address-space helpers for global_load_lds (direct HBM->LDS DMA)
 typedef __attribute__((address_space(1))) const uint32_t gas_u32;
 typedef __attribute__((address_space(3))) uint32_t las_u32;
