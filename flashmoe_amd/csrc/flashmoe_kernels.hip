@@ -267,6 +267,11 @@ __global__ __launch_bounds__(256) void k_gate_route(
 //                 (processor.cuh:126-168); k==1: unscaled store to moe_out
 //                 (processor.cuh:173-204)
 // PHASE 2 (down-direct, EP path): like 1 but plain store to out_rows[m].
+// PHASE 3 (gate logits): A = x (identity rows), B = gate_w viewed [E,H]
+//                 (reference quirk, moe.cuh:107-109); epilogue: plain f32
+//                 store to out[m*N + col] - feeds k_gate_route. MFMA
+//                 replaces the VALU logits kernel (it was ~80 us at the
+//                 E=64/H=1024 8-GPU shape).
 // If tokenIds == nullptr the row gather is identity and routed = n_rows
 // (packed-rows mode for fm_expert_ffn).
 // ---------------------------------------------------------------------------
@@ -460,9 +465,11 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
                 a.moe_out)[(size_t)tp.tokenIdx * a.H + col] =
                 ETr<ET>::fromf(v);
           }
-        } else {  // PHASE 2: packed-rows direct output
+        } else if constexpr (PHASE == 2) {  // packed-rows direct output
           reinterpret_cast<ET*>(a.out)[(size_t)m * N + col] =
               ETr<ET>::fromf(v);
+        } else {  // PHASE 3: gate logits, fp32
+          reinterpret_cast<float*>(a.out)[(size_t)m * N + col] = v;
         }
       }
     }
@@ -896,6 +903,9 @@ __global__ void k_mfma_probe(const bf16* A, const bf16* B, float* D) {
 // Host side: state, workspace, ABI
 // ===========================================================================
 
+static int launch_group_gemm(hipStream_t st, int phase, const struct GemmArgs& a,
+                             int M, int nE);
+
 namespace {
 
 struct State {
@@ -947,7 +957,9 @@ int launch_gate(hipStream_t st, const void* x, const void* gate_w,
   const int chunks = DIVUP(g.H, Hc);
   const int eChunks = DIVUP(g.E, 128);
   const int Ec = g.E < 128 ? g.E : 128;
-  FM_HIP_CHECK(hipMemsetAsync(g.logits32, 0, (size_t)g.S * g.E * sizeof(float), st));
+  const bool mfmaLogits = (g.esz == 2);  // bf16/fp16: MFMA logits GEMM
+  if (!mfmaLogits)
+    FM_HIP_CHECK(hipMemsetAsync(g.logits32, 0, (size_t)g.S * g.E * sizeof(float), st));
   const size_t ldsL = gate_lds_bytes(Ec, g.esz);
   const size_t ldsR = 128 * (g.E + 1) * sizeof(float) +
                       128 * 8 * 2 * sizeof(uint16_t) + g.E * sizeof(uint32_t) +
@@ -986,11 +998,26 @@ int launch_gate(hipStream_t st, const void* x, const void* gate_w,
     case 8: GATE_ROUTE(T, 8); break;                                          \
     default: setErr("unsupported expert_top_k (1,2,4,8)"); return FM_ERR_UNSUPPORTED; \
   }
+  if (mfmaLogits) {
+    GemmArgs ga{};
+    ga.A = x;
+    ga.B = gate_w;
+    ga.out = g.logits32;
+    ga.tokenIds = nullptr;
+    ga.eC = nullptr;
+    ga.K = g.H;
+    ga.N = g.E;
+    ga.nRows = (int)S;
+    ga.H = g.H;
+    ga.splitK = 1;
+    int rc = launch_group_gemm(st, 3, ga, (int)S, 1);
+    if (rc != FM_OK) return rc;
+  }
   if (g.cfg.dtype == 2) {
-    GATE_LOGITS(bf16);
+    if (!mfmaLogits) GATE_LOGITS(bf16);
     GATE_K(bf16)
   } else if (g.cfg.dtype == 3) {
-    GATE_LOGITS(fp16);
+    if (!mfmaLogits) GATE_LOGITS(fp16);
     GATE_K(fp16)
   } else {
     GATE_LOGITS(float);
@@ -1112,6 +1139,17 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
     dim3 block(mode == 2 ? 256 : 512);
     dim3 grid(DIVUP(M, mode == 2 ? 128 : 256),
               DIVUP(N, mode == 1 ? 128 : (mode == 0 ? 256 : 128)), nE * skf);
+    if (phase == 3) {  // gate logits: small kernel, no bias/act
+      dim3 g3(DIVUP(M, 128), DIVUP(N, 128), 1);
+      if (g.cfg.dtype == 3)
+        hipLaunchKernelGGL((k_group_gemm_bf16<fp16, 3, 0, false>), g3,
+                           dim3(256), 0, st, a);
+      else
+        hipLaunchKernelGGL((k_group_gemm_bf16<bf16, 3, 0, false>), g3,
+                           dim3(256), 0, st, a);
+      FM_HIP_CHECK(hipGetLastError());
+      return FM_OK;
+    }
     const int sel = phase * 4 + act * 2 + (hasBias ? 1 : 0);
 #define GG_ET(ET, PH, AC, HB)                                                 \
     do {                                                                      \
