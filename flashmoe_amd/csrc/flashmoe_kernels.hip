@@ -1380,6 +1380,30 @@ int fm_expert_ffn(void* stream, const void* rows, const void* expert_w,
   return launch_group_gemm(st, 2, dn, (int)n_rows, 1);
 }
 
+int fm_expert_ffn_grouped(void* stream, const void* rows,
+                          const int64_t* offsets, int32_t n_experts,
+                          const void* expert_w, const void* b_up,
+                          const void* b_dn, void* out_rows) {
+  if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
+  for (int le = 0; le < n_experts; ++le) {
+    const int64_t n = offsets[le + 1] - offsets[le];
+    if (n <= 0) continue;
+    const char* r = reinterpret_cast<const char*>(rows) +
+                    (size_t)offsets[le] * g.H * g.esz;
+    char* o = reinterpret_cast<char*>(out_rows) +
+              (size_t)offsets[le] * g.H * g.esz;
+    const void* bu = b_up ? reinterpret_cast<const char*>(b_up) +
+                                (size_t)le * g.P * g.esz
+                          : nullptr;
+    const void* bd = b_dn ? reinterpret_cast<const char*>(b_dn) +
+                                (size_t)le * g.H * g.esz
+                          : nullptr;
+    int rc = fm_expert_ffn(stream, r, expert_w, bu, bd, o, n, le);
+    if (rc != FM_OK) return rc;
+  }
+  return FM_OK;
+}
+
 int fm_combine(void* stream, const void* rows, const uint32_t* token_idx,
                const float* scale, int64_t n_rows, int32_t zero_first) {
   if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }

@@ -156,11 +156,18 @@ def exchange_rows(x_rows: torch.Tensor, send_counts: torch.Tensor,
     return out
 
 
-def moe_forward_ep(input, gate_weights, expert_weights, group=None):
+def moe_forward_ep(input, gate_weights, expert_weights, group=None,
+                   _stub_exchange=False):
     """The EP forward on GPU ranks (called from moe.moe_forward when
     world > 1). Planning runs on-GPU from fm_export_routing; the only
     host syncs are the [world]-sized split counts the RCCL all_to_all
-    API needs as python ints."""
+    API needs as python ints.
+
+    _stub_exchange=True replaces every all_to_all with an identity
+    (tokens processed as if all their experts were local) - the
+    compute-only arm of the overlap-efficiency metric
+    (t_compute_only / t_full, reference README.md:33-35 concept);
+    results are NOT the DMoE output in that mode."""
     import torch.distributed as dist
 
     from . import _ext, moe
@@ -211,25 +218,26 @@ def moe_forward_ep(input, gate_weights, expert_weights, group=None):
     perm_d = perm.to(dev)
     grouped = recv_rows.index_select(0, perm_d) if perm.numel() else recv_rows
     out_grouped = torch.empty_like(grouped)
-    off = 0
+    offsets = (ctypes.c_int64 * (nLx + 1))()
+    acc = 0
     for le in range(nLx):
-        n = int(per_expert[le])
-        if n == 0:
-            continue
-        _ext.check(lib.fm_expert_ffn(
-            ctypes.c_void_p(stream),
-            ctypes.c_void_p(grouped[off:off + n].data_ptr()),
-            ctypes.c_void_p(expert_weights.data_ptr()),
-            None, None,
-            ctypes.c_void_p(out_grouped[off:off + n].data_ptr()),
-            n, le), "fm_expert_ffn")
-        off += n
+        offsets[le] = acc
+        acc += int(per_expert[le])
+    offsets[nLx] = acc
+    _ext.check(lib.fm_expert_ffn_grouped(
+        ctypes.c_void_p(stream),
+        ctypes.c_void_p(grouped.data_ptr()), offsets, nLx,
+        ctypes.c_void_p(expert_weights.data_ptr()), None, None,
+        ctypes.c_void_p(out_grouped.data_ptr())), "fm_expert_ffn_grouped")
     result_rows = torch.empty_like(recv_rows)
     if perm.numel():
         result_rows.index_copy_(0, perm_d, out_grouped)
 
     # 5. return exchange (reverse splits), combine at source
-    returned = exchange_rows(result_rows, recv_counts, send_counts_cpu, group)
+    if _stub_exchange:
+        returned = result_rows
+    else:
+        returned = exchange_rows(result_rows, recv_counts, send_counts_cpu, group)
     n = returned.shape[0]
     tok_d = plan.token_idx.to(torch.int32)
     if k > 1:

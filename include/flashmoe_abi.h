@@ -114,6 +114,15 @@ int fm_expert_ffn(void* stream, const void* rows, const void* expert_w,
                   const void* b_up, const void* b_dn, void* out_rows,
                   int64_t n_rows, int32_t local_e);
 
+/* Batched fm_expert_ffn over all local experts: rows are packed
+ * expert-major; offsets = host array of n_experts+1 row offsets
+ * (offsets[le]..offsets[le+1] belong to local expert le). One call per
+ * EP step instead of nLx ctypes crossings. */
+int fm_expert_ffn_grouped(void* stream, const void* rows,
+                          const int64_t* offsets, int32_t n_experts,
+                          const void* expert_w, const void* b_up,
+                          const void* b_dn, void* out_rows);
+
 /* Combine pre-scaled return rows into moe_out: for i < n_rows,
  * moe_out[token_idx[i]] += scale[i] * rows[i] (k>1 path,
  * processor.cuh:126-168); k==1: unscaled overwrite. Caller passes
