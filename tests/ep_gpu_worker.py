@@ -50,6 +50,11 @@ def main():
     out_ep = ep.moe_forward_ep(x, gw, ew_local)
     torch.cuda.synchronize()
 
+    # exercise the compute-only overlap-probe arm (results unchecked -
+    # it is a timing stub; just must run)
+    ep.moe_forward_ep(x, gw, ew_local, _stub_exchange=True)
+    torch.cuda.synchronize()
+
     # reference: the single-rank HIP path on this rank's tokens with ALL
     # experts (requires a world-1 re-init)
     moe.finalize()
