@@ -54,6 +54,12 @@ def _bind(lib: ctypes.CDLL) -> ctypes.CDLL:
     lib.fm_read_routing.restype = ctypes.c_int
     lib.fm_expert_ffn.argtypes = [p, p, p, p, p, p, i64, ctypes.c_int32]
     lib.fm_expert_ffn.restype = ctypes.c_int
+    lib.fm_pack_dispatch.argtypes = [p, p, p]
+    lib.fm_pack_dispatch.restype = ctypes.c_int
+    lib.fm_expert_ffn_segments.argtypes = [p, p, p, ctypes.c_int32, p, p]
+    lib.fm_expert_ffn_segments.restype = ctypes.c_int
+    lib.fm_combine_padded.argtypes = [p, p, p, p, i64]
+    lib.fm_combine_padded.restype = ctypes.c_int
     lib.fm_expert_ffn_grouped.argtypes = [p, p, p, ctypes.c_int32, p, p, p, p]
     lib.fm_expert_ffn_grouped.restype = ctypes.c_int
     lib.fm_combine.argtypes = [p, p, p, p, i64, ctypes.c_int32]

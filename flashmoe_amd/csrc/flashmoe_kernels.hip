@@ -299,6 +299,8 @@ struct GemmArgs {
   int H;                // row stride of x / O32 / moe_out
   int splitK;           // K-split factor (PHASE 1 multi only: the fp32
                         // atomicAdd combine makes split-K partials free)
+  const int32_t* segExpert;  // optional [gridZ] device map: z -> weight
+                             // expert (padded-EP segments; null = identity)
 };
 
 // address-space helpers for global_load_lds (direct HBM->LDS DMA)
@@ -328,6 +330,7 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
   const int n0 = blockIdx.y * BN;
   const int K = a.K, N = a.N;
 
+  const int we = a.segExpert ? a.segExpert[e] : e;  // weight/bias expert
   const TPS* tpsE = a.tokenIds ? a.tokenIds + (size_t)e * a.pEC : nullptr;
   if (tid == 0) {
     *sRouted = tpsE ? min(a.eC[e], (uint32_t)a.EC) : (uint32_t)a.nRows;
@@ -347,7 +350,7 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
 
   const ET* __restrict__ Ag = reinterpret_cast<const ET*>(a.A);
   const ET* __restrict__ Bg =
-      reinterpret_cast<const ET*>(a.B) + (size_t)e * a.strideBExpert;
+      reinterpret_cast<const ET*>(a.B) + (size_t)we * a.strideBExpert;
 
   // accumulators: wave (wr,wc) owns the 64x64 subtile at (wr*64, wc*64)
   const int wr = wave >> 1, wc = wave & 1;
@@ -466,7 +469,8 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
                 ETr<ET>::fromf(v);
           }
         } else if constexpr (PHASE == 2) {  // packed-rows direct output
-          reinterpret_cast<ET*>(a.out)[(size_t)m * N + col] =
+          reinterpret_cast<ET*>(a.out)[(size_t)e * a.strideOExpert +
+                                       (size_t)m * N + col] =
               ETr<ET>::fromf(v);
         } else {  // PHASE 3: gate logits, fp32
           reinterpret_cast<float*>(a.out)[(size_t)m * N + col] = v;
@@ -538,6 +542,7 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
   const uint32_t routed = *sRouted;
   if ((uint32_t)m0 >= routed) return;
   const int mCap = a.tokenIds ? a.pEC : a.nRows;  // A-row clamp bound
+  const int we = a.segExpert ? a.segExpert[e] : e;  // weight/bias expert
   if (tid < BM) {
     TPS t{0u, 1.0f};
     if ((uint32_t)(m0 + tid) < routed)
@@ -548,7 +553,7 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
 
   const ET* __restrict__ Ag = reinterpret_cast<const ET*>(a.A);
   const ET* __restrict__ Bg =
-      reinterpret_cast<const ET*>(a.B) + (size_t)e * a.strideBExpert;
+      reinterpret_cast<const ET*>(a.B) + (size_t)we * a.strideBExpert;
 
   // per-lane glds source bases, hoisted out of the K loop
   const int grow8 = lane >> 3;
@@ -710,7 +715,8 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
                 ETr<ET>::fromf(v);
           }
         } else {
-          reinterpret_cast<ET*>(a.out)[(size_t)m * N + col] =
+          reinterpret_cast<ET*>(a.out)[(size_t)e * a.strideOExpert +
+                                       (size_t)m * N + col] =
               ETr<ET>::fromf(v);
         }
       }
@@ -739,6 +745,7 @@ __global__ __launch_bounds__(256) void k_group_gemm_f32(GemmArgs a) {
   const int n0 = blockIdx.y * BN;
   const int K = a.K, N = a.N;
   const TPS* tpsE = a.tokenIds ? a.tokenIds + (size_t)e * a.pEC : nullptr;
+  const int we = a.segExpert ? a.segExpert[e] : e;
   if (tid == 0)
     sRouted = tpsE ? min(a.eC[e], (uint32_t)a.EC) : (uint32_t)a.nRows;
   __syncthreads();
@@ -754,7 +761,7 @@ __global__ __launch_bounds__(256) void k_group_gemm_f32(GemmArgs a) {
 
   const float* __restrict__ Ag = reinterpret_cast<const float*>(a.A);
   const float* __restrict__ Bg =
-      reinterpret_cast<const float*>(a.B) + (size_t)e * a.strideBExpert;
+      reinterpret_cast<const float*>(a.B) + (size_t)we * a.strideBExpert;
 
   const int tr = (tid / 16) * 4;  // thread rows [tr, tr+4)
   const int tc = (tid % 16) * 4;
@@ -825,7 +832,8 @@ __global__ __launch_bounds__(256) void k_group_gemm_f32(GemmArgs a) {
                                               col] = v;
         }
       } else {
-        reinterpret_cast<float*>(a.out)[(size_t)m * N + col] = v;
+        reinterpret_cast<float*>(a.out)[(size_t)e * a.strideOExpert +
+                                        (size_t)m * N + col] = v;
       }
     }
   }
@@ -861,6 +869,57 @@ __global__ void k_combine_rows(const T* __restrict__ rows,
       atomicAdd(&O32[(size_t)t * H + h], s * v);
     else
       moeOut[(size_t)t * H + h] = rows[(size_t)i * H + h];
+  }
+}
+
+// pack the dispatch send buffer: send[e][i][:] = x[tokenIds[e][i]][:]
+// for i < min(eC[e], EC); rows past the routed count keep whatever is
+// there (they are dropped at the source on combine). This is the
+// reference's symmetric-heap cell layout (peer-major expert-slot cells,
+// types.cuh:1014-1032) built for a capacity-padded static all_to_all.
+template <typename T>
+__global__ void k_pack_dispatch(const T* __restrict__ x,
+                                const TPS* __restrict__ tokenIds,
+                                const uint32_t* __restrict__ eC,
+                                T* __restrict__ sendbuf, int H, int EC,
+                                int pEC) {
+  const int e = blockIdx.y;
+  const int i = blockIdx.x;
+  const uint32_t r = min(eC[e], (uint32_t)EC);
+  if ((uint32_t)i >= r) return;
+  const uint32_t tok = tokenIds[(size_t)e * pEC + i].tokenIdx;
+  const T* src = x + (size_t)tok * H;
+  T* dst = sendbuf + ((size_t)e * EC + i) * H;
+  for (int h = threadIdx.x * 8; h < H; h += blockDim.x * 8) {
+    *reinterpret_cast<u32x4*>(dst + h) =
+        *reinterpret_cast<const u32x4*>(src + h);
+  }
+}
+
+// combine the capacity-padded RETURNED buffer at the source: for each
+// (e, i < routed), out[token] += gate_out[token, e]/probSum * row
+// (k>1; k==1 unscaled overwrite semantics via scale=1 on a zero
+// accumulator). All metadata is source-local (tokenIds/eC/gate_out) -
+// no host involvement (processor.cuh:44-205 combine semantics).
+template <typename T>
+__global__ void k_combine_padded(const T* __restrict__ rows,
+                                 const TPS* __restrict__ tokenIds,
+                                 const uint32_t* __restrict__ eC,
+                                 const T* __restrict__ gate_out,
+                                 float* __restrict__ O32, int H, int EC,
+                                 int pEC, int PX, int topk) {
+  const int e = blockIdx.y;
+  const int i = blockIdx.x;
+  const uint32_t r = min(eC[e], (uint32_t)EC);
+  if ((uint32_t)i >= r) return;
+  const TPS tp = tokenIds[(size_t)e * pEC + i];
+  const float sc = (topk > 1)
+      ? toF(gate_out[(size_t)tp.tokenIdx * PX + e]) / tp.probSum
+      : 1.0f;
+  const T* src = rows + ((size_t)e * EC + i) * H;
+  float* dst = O32 + (size_t)tp.tokenIdx * H;
+  for (int h = threadIdx.x; h < H; h += blockDim.x) {
+    atomicAdd(dst + h, sc * toF(src[h]));
   }
 }
 
@@ -1378,6 +1437,100 @@ int fm_expert_ffn(void* stream, const void* rows, const void* expert_w,
   dn.out = out_rows;
   dn.K = g.P; dn.N = g.H;
   return launch_group_gemm(st, 2, dn, (int)n_rows, 1);
+}
+
+int fm_pack_dispatch(void* stream, const void* x, void* sendbuf) {
+  if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
+  hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+  dim3 grid(g.EC, g.E);
+  if (g.cfg.dtype == 3)
+    hipLaunchKernelGGL(k_pack_dispatch<fp16>, grid, dim3(128), 0, st,
+                       reinterpret_cast<const fp16*>(x), g.tokenIds, g.eC,
+                       reinterpret_cast<fp16*>(sendbuf), g.H, g.EC, g.pEC);
+  else if (g.esz == 2)
+    hipLaunchKernelGGL(k_pack_dispatch<bf16>, grid, dim3(128), 0, st,
+                       reinterpret_cast<const bf16*>(x), g.tokenIds, g.eC,
+                       reinterpret_cast<bf16*>(sendbuf), g.H, g.EC, g.pEC);
+  else
+    hipLaunchKernelGGL(k_pack_dispatch<float>, grid, dim3(128), 0, st,
+                       reinterpret_cast<const float*>(x), g.tokenIds, g.eC,
+                       reinterpret_cast<float*>(sendbuf), g.H, g.EC, g.pEC);
+  FM_HIP_CHECK(hipGetLastError());
+  return FM_OK;
+}
+
+int fm_combine_padded(void* stream, const void* returned, const void* gate_out,
+                      void* moe_out, int64_t S) {
+  if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
+  if (S != g.S) { setErr("S mismatch"); return FM_ERR_SHAPE; }
+  hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+  dim3 grid(g.EC, g.E);
+  const int tk = g.cfg.expert_top_k;
+  if (g.cfg.dtype == 3)
+    hipLaunchKernelGGL(k_combine_padded<fp16>, grid, dim3(256), 0, st,
+                       reinterpret_cast<const fp16*>(returned), g.tokenIds,
+                       g.eC, reinterpret_cast<const fp16*>(gate_out), g.O32,
+                       g.H, g.EC, g.pEC, g.PX, tk);
+  else if (g.esz == 2)
+    hipLaunchKernelGGL(k_combine_padded<bf16>, grid, dim3(256), 0, st,
+                       reinterpret_cast<const bf16*>(returned), g.tokenIds,
+                       g.eC, reinterpret_cast<const bf16*>(gate_out), g.O32,
+                       g.H, g.EC, g.pEC, g.PX, tk);
+  else
+    hipLaunchKernelGGL(k_combine_padded<float>, grid, dim3(256), 0, st,
+                       reinterpret_cast<const float*>(returned), g.tokenIds,
+                       g.eC, reinterpret_cast<const float*>(gate_out), g.O32,
+                       g.H, g.EC, g.pEC, g.PX, tk);
+  FM_HIP_CHECK(hipGetLastError());
+  return fm_combine_finalize(stream, moe_out, S);
+}
+
+/* Padded-EP grouped FFN: rows = [n_segs, EC, H] (fixed EC rows per
+ * segment), seg_expert_dev = device int32[n_segs] mapping segment ->
+ * local expert. TWO kernel launches total (grid.z = segments via the
+ * GemmArgs segExpert indirection). Rows past each segment's real count
+ * are garbage-in/garbage-out by design (dropped at the source). */
+int fm_expert_ffn_segments(void* stream, const void* rows,
+                           const void* seg_expert_dev, int32_t n_segs,
+                           const void* expert_w, void* out_rows) {
+  if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
+  if ((int64_t)n_segs * g.EC > (int64_t)g.nLxAlloc * g.pEC) {
+    setErr("segments exceed xM workspace"); return FM_ERR_SHAPE;
+  }
+  hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+  GemmArgs up{};
+  up.A = rows;
+  up.B = expert_w;
+  up.bias = nullptr;
+  up.out = g.xM;
+  up.tokenIds = nullptr;
+  up.eC = nullptr;
+  up.strideAExpert = (long long)g.EC * g.H;
+  up.strideBExpert = 2LL * g.P * g.H;
+  up.strideOExpert = (long long)g.EC * g.P;
+  up.K = g.H;
+  up.N = g.P;
+  up.EC = 0;
+  up.pEC = 0;
+  up.PX = g.PX;
+  up.topk = 1;
+  up.act = g.cfg.hidden_act;
+  up.expertOffset = 0;
+  up.nRows = g.EC;
+  up.H = g.H;
+  up.splitK = 1;
+  up.segExpert = reinterpret_cast<const int32_t*>(seg_expert_dev);
+  int rc = launch_group_gemm(st, 0, up, g.EC, n_segs);
+  if (rc != FM_OK) return rc;
+  GemmArgs dn = up;
+  dn.A = g.xM;
+  dn.B = reinterpret_cast<const char*>(expert_w) + (size_t)g.P * g.H * g.esz;
+  dn.out = out_rows;
+  dn.strideAExpert = (long long)g.EC * g.P;
+  dn.strideOExpert = (long long)g.EC * g.H;
+  dn.K = g.P;
+  dn.N = g.H;
+  return launch_group_gemm(st, 2, dn, g.EC, n_segs);
 }
 
 int fm_expert_ffn_grouped(void* stream, const void* rows,

@@ -156,18 +156,39 @@ def exchange_rows(x_rows: torch.Tensor, send_counts: torch.Tensor,
     return out
 
 
+def _padded_buffers(moe, world, E, EC, H, dtype, dev):
+    """Cached static exchange buffers for the padded EP pipeline."""
+    import torch
+
+    key = (world, E, EC, H, str(dtype))
+    cache = moe._state.setdefault("ep_buffers", {})
+    if cache.get("key") != key:
+        nLx = E // world
+        cache.clear()
+        cache["key"] = key
+        cache["send"] = torch.empty(E * EC, H, dtype=dtype, device=dev)
+        cache["recv"] = torch.empty(E * EC, H, dtype=dtype, device=dev)
+        cache["ffn_out"] = torch.empty(E * EC, H, dtype=dtype, device=dev)
+        cache["ret"] = torch.empty(E * EC, H, dtype=dtype, device=dev)
+        # segment s = (source rank, local expert): expert = s % nLx
+        cache["seg_expert"] = (torch.arange(world * nLx, device=dev,
+                                            dtype=torch.int32) % nLx).contiguous()
+    return cache
+
+
 def moe_forward_ep(input, gate_weights, expert_weights, group=None,
                    _stub_exchange=False):
-    """The EP forward on GPU ranks (called from moe.moe_forward when
-    world > 1). Planning runs on-GPU from fm_export_routing; the only
-    host syncs are the [world]-sized split counts the RCCL all_to_all
-    API needs as python ints.
+    """Capacity-padded EP forward (the product multi-GPU path): the
+    exchange unit is a fixed [E, EC, H] expert-major buffer (the
+    reference's symmetric-heap cell layout, types.cuh:1014-1032), so the
+    all_to_all has STATIC equal splits and the step performs NO host
+    synchronization. Rows past each expert's routed count are
+    garbage-in/garbage-out and dropped by the source-side combine.
 
-    _stub_exchange=True replaces every all_to_all with an identity
-    (tokens processed as if all their experts were local) - the
-    compute-only arm of the overlap-efficiency metric
-    (t_compute_only / t_full, reference README.md:33-35 concept);
-    results are NOT the DMoE output in that mode."""
+    _stub_exchange=True skips the exchanges (compute-only arm of the
+    overlap-efficiency metric; output is not the DMoE result).
+    """
+    import torch
     import torch.distributed as dist
 
     from . import _ext, moe
@@ -176,7 +197,6 @@ def moe_forward_ep(input, gate_weights, expert_weights, group=None,
     world = dist.get_world_size(group)
     cc = moe.get_compiled_config()
     S, H, E = cc["S"], cc["H"], cc["E"]
-    nLx = E // world
     cfg = moe._state["cfg"]
     k = cfg["expert_top_k"]
     base = -(-S // E) if cfg["drop_tokens"] else S
@@ -185,73 +205,42 @@ def moe_forward_ep(input, gate_weights, expert_weights, group=None,
     stream = torch.cuda.current_stream().cuda_stream
     dev = input.device
     x2d = input.view(S, H)
+    nLx = E // world
+    buf = _padded_buffers(moe, world, E, EC, H, input.dtype, dev)
 
-    # 1. local gate (fills the library's tokenIds/eC)
+    # 1. local gate (fills tokenIds/eC) + pack the dispatch cells
     _ext.check(lib.fm_gate_forward(
         ctypes.c_void_p(stream),
         ctypes.c_void_p(x2d.data_ptr()),
         ctypes.c_void_p(gate_weights.data_ptr()),
         ctypes.c_void_p(gate_out.data_ptr()), S), "fm_gate_forward")
+    _ext.check(lib.fm_pack_dispatch(
+        ctypes.c_void_p(stream), ctypes.c_void_p(x2d.data_ptr()),
+        ctypes.c_void_p(buf["send"].data_ptr())), "fm_pack_dispatch")
 
-    # 2. GPU-resident routing export + vectorized plan
-    routed = torch.empty(E, dtype=torch.int32, device=dev)
-    tps = torch.empty(E, EC, 2, dtype=torch.int32, device=dev)
-    _ext.check(lib.fm_export_routing(
-        ctypes.c_void_p(stream), ctypes.c_void_p(routed.data_ptr()),
-        ctypes.c_void_p(tps.data_ptr())), "fm_export_routing")
-    tps_f = tps.view(torch.float32)  # reinterpret bits; [E, EC, 2]
-    plan = plan_dispatch_vec(routed.long(), tps[..., 0].long(),
-                             tps_f[..., 1], world)
-
-    # 3. exchange per-(dst, expert) counts, then rows
-    all_expert_counts = exchange_rows(
-        plan.send_expert_counts.contiguous(),
-        torch.ones(world, dtype=torch.long),
-        torch.ones(world, dtype=torch.long), group).cpu()  # [world, nLx]
-    recv_counts = all_expert_counts.sum(1)
-    send_counts_cpu = plan.send_counts.cpu()  # host sync: world ints
-    send_rows = x2d.index_select(0, plan.token_idx)
-    recv_rows = exchange_rows(send_rows, send_counts_cpu, recv_counts, group)
-
-    # 4. group by local expert, FFN, restore order
-    perm, per_expert = segment_recv(all_expert_counts)
-    perm_d = perm.to(dev)
-    grouped = recv_rows.index_select(0, perm_d) if perm.numel() else recv_rows
-    out_grouped = torch.empty_like(grouped)
-    offsets = (ctypes.c_int64 * (nLx + 1))()
-    acc = 0
-    for le in range(nLx):
-        offsets[le] = acc
-        acc += int(per_expert[le])
-    offsets[nLx] = acc
-    _ext.check(lib.fm_expert_ffn_grouped(
-        ctypes.c_void_p(stream),
-        ctypes.c_void_p(grouped.data_ptr()), offsets, nLx,
-        ctypes.c_void_p(expert_weights.data_ptr()), None, None,
-        ctypes.c_void_p(out_grouped.data_ptr())), "fm_expert_ffn_grouped")
-    result_rows = torch.empty_like(recv_rows)
-    if perm.numel():
-        result_rows.index_copy_(0, perm_d, out_grouped)
-
-    # 5. return exchange (reverse splits), combine at source
-    if _stub_exchange:
-        returned = result_rows
+    # 2. dispatch all_to_all: equal static chunks of nLx*EC rows
+    if _stub_exchange or world == 1:
+        recv = buf["send"]
     else:
-        returned = exchange_rows(result_rows, recv_counts, send_counts_cpu, group)
-    n = returned.shape[0]
-    tok_d = plan.token_idx.to(torch.int32)
-    if k > 1:
-        probs = gate_out[plan.token_idx, plan.expert_of_row].float()
-        scale = (probs / plan.prob_sum).contiguous()
+        dist.all_to_all_single(buf["recv"], buf["send"], group=group)
+        recv = buf["recv"]
+
+    # 3. grouped FFN over [world, nLx, EC, H] segments (2 launches)
+    _ext.check(lib.fm_expert_ffn_segments(
+        ctypes.c_void_p(stream), ctypes.c_void_p(recv.data_ptr()),
+        ctypes.c_void_p(buf["seg_expert"].data_ptr()), world * nLx,
+        ctypes.c_void_p(expert_weights.data_ptr()),
+        ctypes.c_void_p(buf["ffn_out"].data_ptr())), "fm_expert_ffn_segments")
+
+    # 4. return all_to_all (same static splits), combine at the source
+    if _stub_exchange or world == 1:
+        returned = buf["ffn_out"]
     else:
-        scale = torch.ones(n, dtype=torch.float32, device=dev)
-    _ext.check(lib.fm_combine(
-        ctypes.c_void_p(stream),
-        ctypes.c_void_p(returned.data_ptr()),
-        ctypes.c_void_p(tok_d.data_ptr()),
-        ctypes.c_void_p(scale.data_ptr()), n, 1), "fm_combine")
+        dist.all_to_all_single(buf["ret"], buf["ffn_out"], group=group)
+        returned = buf["ret"]
     out = torch.empty_like(input)
-    _ext.check(lib.fm_combine_finalize(
-        ctypes.c_void_p(stream), ctypes.c_void_p(out.data_ptr()), S),
-        "fm_combine_finalize")
+    _ext.check(lib.fm_combine_padded(
+        ctypes.c_void_p(stream), ctypes.c_void_p(returned.data_ptr()),
+        ctypes.c_void_p(gate_out.data_ptr()),
+        ctypes.c_void_p(out.data_ptr()), S), "fm_combine_padded")
     return out

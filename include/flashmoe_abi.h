@@ -114,6 +114,30 @@ int fm_expert_ffn(void* stream, const void* rows, const void* expert_w,
                   const void* b_up, const void* b_dn, void* out_rows,
                   int64_t n_rows, int32_t local_e);
 
+/* --- Capacity-padded EP pipeline (the product multi-GPU path). The
+ * exchange unit is the reference's symmetric-heap cell layout
+ * (types.cuh:1014-1032): a fixed [E, EC, H] buffer, expert-major,
+ * capacity-padded - so the all_to_all has STATIC equal splits and the
+ * step needs no host synchronization at all. Rows past each expert's
+ * routed count carry garbage and are dropped at the source. --- */
+
+/* Gather x rows into the [E, EC, H] dispatch buffer per the last
+ * fm_gate_forward's routing (os/packet.cuh:20-286 dispatch semantics). */
+int fm_pack_dispatch(void* stream, const void* x, void* sendbuf);
+
+/* FFN over the received padded buffer: rows = [n_segs, EC, H] with
+ * seg_expert_dev = device int32[n_segs] mapping segment -> local expert
+ * (canonical order: source-rank major). Two kernel launches total. */
+int fm_expert_ffn_segments(void* stream, const void* rows,
+                           const void* seg_expert_dev, int32_t n_segs,
+                           const void* expert_w, void* out_rows);
+
+/* Combine the returned [E, EC, H] buffer at the source using the local
+ * routing metadata (scale = gate_out/probSum; k==1 unscaled) and write
+ * moe_out (processor.cuh:44-205). */
+int fm_combine_padded(void* stream, const void* returned,
+                      const void* gate_out, void* moe_out, int64_t S);
+
 /* Batched fm_expert_ffn over all local experts: rows are packed
  * expert-major; offsets = host array of n_experts+1 row offsets
  * (offsets[le]..offsets[le+1] belong to local expert le). One call per
