@@ -366,7 +366,7 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
   const int grow8 = lane >> 3;           // row within the 8-row group
   const int schunk = (lane & 7) ^ grow8;  // swizzled 16B chunk index
   const int aRowStride = (PHASE == 0) ? a.H : K;
-  const size_t aBase = (PHASE == 0) ? 0 : (size_t)e * a.strideAExpert;
+  const size_t aBase = (size_t)e * a.strideAExpert;  // 0 for the x-gather up phase
   for (int kt = 0; kt < K; kt += BK) {
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
@@ -559,7 +559,7 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
   const int grow8 = lane >> 3;
   const int schunk = (lane & 7) ^ grow8;
   const int aRowStride = (PHASE == 0) ? a.H : K;
-  const size_t aBase = (PHASE == 0) ? 0 : (size_t)e * a.strideAExpert;
+  const size_t aBase = (size_t)e * a.strideAExpert;  // 0 for the x-gather up phase
   const ET* aSrc[GPW_A];
   const ET* bSrc[GPW_B];
 #pragma unroll
@@ -888,9 +888,10 @@ __global__ void k_pack_dispatch(const T* __restrict__ x,
   const uint32_t r = min(eC[e], (uint32_t)EC);
   if ((uint32_t)i >= r) return;
   const uint32_t tok = tokenIds[(size_t)e * pEC + i].tokenIdx;
+  constexpr int EPU = 16 / sizeof(T);  // elements per 16B unit
   const T* src = x + (size_t)tok * H;
   T* dst = sendbuf + ((size_t)e * EC + i) * H;
-  for (int h = threadIdx.x * 8; h < H; h += blockDim.x * 8) {
+  for (int h = threadIdx.x * EPU; h < H; h += blockDim.x * EPU) {
     *reinterpret_cast<u32x4*>(dst + h) =
         *reinterpret_cast<const u32x4*>(src + h);
   }
