@@ -10,111 +10,15 @@ results are identical to the single-rank path on that rank's tokens
 is what the gloo CPU tests check. Round 2 replaces the exchange with
 in-kernel xGMI stores.
 
-The pure-torch helpers (plan_dispatch / segment_recv) are device-agnostic
-so the CPU (gloo) tests cover the exchange logic exactly as the GPU path
-runs it.
+`exchange_rows` doubles as the gloo (CPU test) emulation of the
+all_to_all, so tests/test_ep_padded_gloo.py exercises the same exchange
+semantics the RCCL path uses.
 """
 from __future__ import annotations
 
 import ctypes
-from dataclasses import dataclass
 
 import torch
-
-
-@dataclass
-class DispatchPlan:
-    """Packing of this rank's routed tokens for the all-to-all.
-
-    order: indices into the (expert-major, slot-minor) routed list such
-    that rows are grouped by destination rank (= expert // nLx), expert
-    ascending within a rank, slot (arrival order) ascending within an
-    expert — the canonical order both sides agree on.
-    """
-
-    send_counts: torch.Tensor  # [world] rows sent to each rank
-    send_expert_counts: torch.Tensor  # [world, nLx] rows per (dst, local expert)
-    token_idx: torch.Tensor  # [n] source-local token index, canonical order
-    prob_sum: torch.Tensor  # [n] float32
-    expert_of_row: torch.Tensor  # [n] global expert id
-
-
-def plan_dispatch(routed_counts: torch.Tensor, token_idx_2d: torch.Tensor,
-                  prob_sum_2d: torch.Tensor, world: int) -> DispatchPlan:
-    """routed_counts: [E]; token_idx_2d/prob_sum_2d: [E, EC] (slots beyond
-    routed_counts[e] are junk). Experts are owned contiguously:
-    rank r owns experts [r*nLx, (r+1)*nLx) (uniform split,
-    bootstrap.cuh:36-52)."""
-    E = routed_counts.numel()
-    nLx = E // world
-    toks, probs, experts = [], [], []
-    send_counts = torch.zeros(world, dtype=torch.long)
-    send_expert_counts = torch.zeros(world, nLx, dtype=torch.long)
-    for e in range(E):
-        n = int(routed_counts[e])
-        if n == 0:
-            continue
-        toks.append(token_idx_2d[e, :n])
-        probs.append(prob_sum_2d[e, :n])
-        experts.append(torch.full((n,), e, dtype=torch.long))
-        send_counts[e // nLx] += n
-        send_expert_counts[e // nLx, e % nLx] += n
-    cat = (lambda lst, dt: torch.cat(lst) if lst else torch.empty(0, dtype=dt))
-    return DispatchPlan(
-        send_counts=send_counts,
-        send_expert_counts=send_expert_counts,
-        token_idx=cat(toks, torch.int64).long(),
-        prob_sum=cat(probs, torch.float32).float(),
-        expert_of_row=cat(experts, torch.int64),
-    )
-
-
-def plan_dispatch_vec(counts: torch.Tensor, tok2d: torch.Tensor,
-                      ps2d: torch.Tensor, world: int) -> DispatchPlan:
-    """Vectorized plan_dispatch (no python loop over experts): same
-    canonical order — expert-major, slot-minor; experts are owned
-    contiguously so destination-rank grouping is automatic. All tensor
-    ops, so it runs on-GPU in the EP hot path (tested equivalent to
-    plan_dispatch in tests/test_ep_gloo.py)."""
-    E, ECap = tok2d.shape
-    nLx = E // world
-    slot = torch.arange(ECap, device=counts.device)
-    mask = slot[None, :] < counts[:, None]            # [E, EC]
-    token_idx = tok2d[mask]                           # row-major -> canonical
-    prob_sum = ps2d[mask]
-    e_ids = torch.arange(E, device=counts.device)[:, None].expand(E, ECap)
-    expert_of_row = e_ids[mask]
-    sec = counts.reshape(world, nLx)
-    return DispatchPlan(
-        send_counts=sec.sum(1),
-        send_expert_counts=sec,
-        token_idx=token_idx.long(),
-        prob_sum=prob_sum.float(),
-        expert_of_row=expert_of_row,
-    )
-
-
-def segment_recv(recv_expert_counts: torch.Tensor):
-    """recv_expert_counts: [world, nLx] rows per (source, local expert) in
-    the canonical receive order (source-major, expert-minor). Returns a
-    permutation grouping the received rows by local expert (expert-major,
-    source-minor, slot order preserved) and the per-expert row counts —
-    the order the grouped-FFN consumes, inverse applied before the
-    return exchange."""
-    world, nLx = recv_expert_counts.shape
-    counts = recv_expert_counts
-    start = torch.cat([torch.zeros(1, dtype=torch.long), counts.flatten().cumsum(0)[:-1]])
-    offs = start.reshape(world, nLx)
-    perm = []
-    per_expert = torch.zeros(nLx, dtype=torch.long)
-    for le in range(nLx):
-        for r in range(world):
-            n = int(counts[r, le])
-            if n:
-                perm.append(torch.arange(offs[r, le], offs[r, le] + n))
-                per_expert[le] += n
-    perm_t = torch.cat(perm) if perm else torch.empty(0, dtype=torch.long)
-    return perm_t, per_expert
 
 
 def exchange_rows(x_rows: torch.Tensor, send_counts: torch.Tensor,
