@@ -626,28 +626,37 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
       vec8 af[8], bfr[NF];
       const int rl = lane & 15;
       const int cbase = 4 * s + (lane >> 4);
-      // read the B fragments FIRST: the first MFMA (af[0] x bfr[*])
-      // then depends only on the B reads + af[0], so the compiler can
-      // ladder partial lgkmcnt waits instead of draining all 12 reads
-      // before the first MFMA (seen in the .s with af-first order)
+      auto aread = [&](int mi) {
+        const int R = wr * 128 + mi * 16 + rl;
+        af[mi] = *reinterpret_cast<const vec8*>(
+            &Al[R * BK + ((cbase ^ (R & 7)) * 8)]);
+      };
+      // Round-robin reads and MFMAs, pinned with sched_barrier(0): each
+      // MFMA pair then waits (partial lgkmcnt) only on fragments read a
+      // group earlier, instead of the scheduler's whole-segment
+      // lgkmcnt(0) drain before the first MFMA (seen in the .s).
 #pragma unroll
       for (int ni = 0; ni < NF; ++ni) {
         const int R = wc * (BN / 4) + ni * 16 + rl;
         bfr[ni] = *reinterpret_cast<const vec8*>(
             &Bl[R * BK + ((cbase ^ (R & 7)) * 8)]);
       }
+      aread(0);
+      aread(1);
+      __builtin_amdgcn_sched_barrier(0);
 #pragma unroll
-      for (int mi = 0; mi < 8; ++mi) {
-        const int R = wr * 128 + mi * 16 + rl;
-        af[mi] = *reinterpret_cast<const vec8*>(
-            &Al[R * BK + ((cbase ^ (R & 7)) * 8)]);
+      for (int g2 = 0; g2 < 4; ++g2) {
+        if (g2 < 3) {
+          aread(2 * g2 + 2);
+          aread(2 * g2 + 3);
+        }
+#pragma unroll
+        for (int mi = 2 * g2; mi < 2 * g2 + 2; ++mi)
+#pragma unroll
+          for (int ni = 0; ni < NF; ++ni)
+            accv[mi][ni] = ETr<ET>::mfma(af[mi], bfr[ni], accv[mi][ni]);
+        __builtin_amdgcn_sched_barrier(0);
       }
-#pragma unroll
-      for (int mi = 0; mi < 8; ++mi)
-#pragma unroll
-        for (int ni = 0; ni < NF; ++ni)
-          accv[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af[mi], bfr[ni], accv[mi][ni], 0, 0, 0);
     }
     __builtin_amdgcn_s_setprio(0);
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // t+1 landed

@@ -190,6 +190,17 @@ def test_single_tile_fp16_top2(fresh_moe):
     assert np.allclose(got, ref["moe_out"], rtol=2e-2, atol=2e-3 * scale)
 
 
+def test_multi_tile_fp16_big_kernel(fresh_moe):
+    """fp16 through the 256-row kernel (S=2048 -> pEC >= 256), CF=2 so
+    capacity cannot overflow (deterministic routing)."""
+    cfg, path = make_cfg(torch_dtype=3, sequence_len=2048, capacity_factor=2,
+                         hidden_size=256, intermediate_size=512)
+    out, gate_out, ref, _ = run_pair(fresh_moe, cfg, path)
+    got = out.float().cpu().numpy()
+    scale = max(1.0, float(np.abs(ref["moe_out"]).max()))
+    assert np.allclose(got, ref["moe_out"], rtol=2e-2, atol=2e-3 * scale)
+
+
 def test_single_tile_gelu(fresh_moe):
     cfg, path = make_cfg(hidden_act=1)
     out, gate_out, ref, _ = run_pair(fresh_moe, cfg, path)
