@@ -54,6 +54,18 @@ def _bind(lib: ctypes.CDLL) -> ctypes.CDLL:
     lib.fm_read_routing.restype = ctypes.c_int
     lib.fm_expert_ffn.argtypes = [p, p, p, p, p, p, i64, ctypes.c_int32]
     lib.fm_expert_ffn.restype = ctypes.c_int
+    lib.fm_heap_init.argtypes = []
+    lib.fm_heap_init.restype = ctypes.c_int
+    lib.fm_heap_handle.argtypes = [p]
+    lib.fm_heap_handle.restype = ctypes.c_int
+    lib.fm_heap_connect.argtypes = [p]
+    lib.fm_heap_connect.restype = ctypes.c_int
+    lib.fm_heap_ptrs.argtypes = [ctypes.POINTER(ctypes.c_void_p)] * 2
+    lib.fm_heap_ptrs.restype = ctypes.c_int
+    lib.fm_dispatch_p2p.argtypes = [p, p]
+    lib.fm_dispatch_p2p.restype = ctypes.c_int
+    lib.fm_return_p2p.argtypes = [p, p]
+    lib.fm_return_p2p.restype = ctypes.c_int
     lib.fm_pack_dispatch.argtypes = [p, p, p]
     lib.fm_pack_dispatch.restype = ctypes.c_int
     lib.fm_expert_ffn_segments.argtypes = [p, p, p, ctypes.c_int32, p, p]

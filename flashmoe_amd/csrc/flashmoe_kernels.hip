@@ -933,6 +933,129 @@ __global__ void k_combine_padded(const T* __restrict__ rows,
   }
 }
 
+// ---------------------------------------------------------------------------
+// One-sided P2P dispatch/return over xGMI (the reference's intra-node
+// mode: direct stores into peers' symmetric-heap cells + system-scope
+// 8-byte signals, os/packet.cuh:214-258 / bootstrap.cuh:442-443
+// nvshmem_ptr resolution). Heap cell layout identical to the padded
+// all_to_all path ([source, local expert, slot<EC, H]); flags are
+// seq-tagged so they never need re-zeroing (seqBit concept,
+// types.cuh:1045-1063). Visibility protocol per the CDNA4 guide
+// (Guideline 16, system scope for cross-device): payload stores ->
+// per-wave vmcnt drain -> one-lane system-scope release fence + asm
+// vmcnt drain -> relaxed system-scope flag store; consumer polls
+// relaxed, then one system-scope acquire fence.
+// ---------------------------------------------------------------------------
+
+// write this rank's routed rows for expert e straight into the OWNER
+// rank's recv heap cell [myRank][e%nLx][i]; the last arriving block of
+// each expert signals the owner's dispatch flag with `seq`.
+template <typename T>
+__global__ void k_dispatch_p2p(const T* __restrict__ x,
+                               const TPS* __restrict__ tokenIds,
+                               const uint32_t* __restrict__ eC,
+                               const uint64_t* __restrict__ peerRecv,
+                               const uint64_t* __restrict__ peerDispFlags,
+                               uint32_t* __restrict__ arrive, int H, int EC,
+                               int pEC, int nLx, int myRank,
+                               unsigned long long seq) {
+  const int e = blockIdx.y;
+  const int i = blockIdx.x;
+  const uint32_t r = min(eC[e], (uint32_t)EC);
+  const uint32_t participants = max(r, 1u);
+  if ((uint32_t)i >= participants) return;
+  const int owner = e / nLx;
+  const int le = e % nLx;
+  T* heap = reinterpret_cast<T*>(peerRecv[owner]);
+  if ((uint32_t)i < r) {
+    const uint32_t tok = tokenIds[(size_t)e * pEC + i].tokenIdx;
+    const T* src = x + (size_t)tok * H;
+    T* dst = heap + (((size_t)myRank * nLx + le) * EC + i) * H;
+    constexpr int EPU = 16 / sizeof(T);
+    for (int h = threadIdx.x * EPU; h < H; h += blockDim.x * EPU) {
+      *reinterpret_cast<u32x4*>(dst + h) =
+          *reinterpret_cast<const u32x4*>(src + h);
+    }
+  }
+  // every wave drains its own stores before the block arrives
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    const uint32_t got = __hip_atomic_fetch_add(
+        arrive + e, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    if (got + 1 == participants) {
+      // last arriver publishes to the owner: system release + flag
+      __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      unsigned long long* flag =
+          reinterpret_cast<unsigned long long*>(peerDispFlags[owner]) +
+          (size_t)myRank * nLx + le;
+      __hip_atomic_store(flag, seq, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_SYSTEM);
+    }
+  }
+}
+
+// wait until every (source, local expert) dispatch flag equals seq
+__global__ void k_await_flags(const unsigned long long* __restrict__ flags,
+                              int n, unsigned long long seq) {
+  bool done = false;
+  for (long long spin = 0; !done && spin < (1ll << 26); ++spin) {
+    done = true;
+    for (int i = threadIdx.x; i < n; i += blockDim.x) {
+      done &= (__hip_atomic_load(flags + i, __ATOMIC_RELAXED,
+                                 __HIP_MEMORY_SCOPE_SYSTEM) >= seq);
+    }
+    done = __syncthreads_and(done);
+    if (!done) __builtin_amdgcn_s_sleep(16);
+  }
+  if (threadIdx.x == 0 && !done) {
+    // bounded spin gave up: poison is preferable to a hang
+    printf("flashmoe: k_await_flags timeout (seq %llu)\n", seq);
+  }
+  __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
+  __syncthreads();
+}
+
+// return the FFN results: segment (s, le) rows go back to source rank
+// s's return heap at cell [global expert][i]; last block per segment
+// signals the source's return flag for that global expert.
+template <typename T>
+__global__ void k_return_p2p(const T* __restrict__ ffn_out,
+                             const uint64_t* __restrict__ peerRet,
+                             const uint64_t* __restrict__ peerRetFlags,
+                             uint32_t* __restrict__ arrive, int H, int EC,
+                             int nLx, int myRank,
+                             unsigned long long seq) {
+  const int seg = blockIdx.y;  // (source rank, local expert)
+  const int i = blockIdx.x;
+  const int src = seg / nLx;
+  const int le = seg % nLx;
+  const int ge = myRank * nLx + le;  // global expert id (I am the owner)
+  T* ret = reinterpret_cast<T*>(peerRet[src]);
+  const T* row = ffn_out + ((size_t)seg * EC + i) * H;
+  T* dst = ret + ((size_t)ge * EC + i) * H;
+  constexpr int EPU = 16 / sizeof(T);
+  for (int h = threadIdx.x * EPU; h < H; h += blockDim.x * EPU) {
+    *reinterpret_cast<u32x4*>(dst + h) =
+        *reinterpret_cast<const u32x4*>(row + h);
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    const uint32_t got = __hip_atomic_fetch_add(
+        arrive + seg, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    if (got + 1 == (uint32_t)EC) {
+      __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      unsigned long long* flag =
+          reinterpret_cast<unsigned long long*>(peerRetFlags[src]) + ge;
+      __hip_atomic_store(flag, seq, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_SYSTEM);
+    }
+  }
+}
+
 // export clipped routing into caller tensors: routed[e] = min(eC, EC),
 // out[e][i] = {tokenIdx, probSum-bits} for i < routed (EP host pipeline)
 __global__ void k_export_routing(const TPS* __restrict__ tokenIds,
@@ -989,6 +1112,17 @@ struct State {
   float* logits32 = nullptr; // [S, E] gate logit accumulator
   float* gML = nullptr;      // [E] training aux: mean gate prob
   float* gMeC = nullptr;     // [E] training aux: routed fraction
+  // one-sided P2P heap (opt-in EP transport)
+  char* heap = nullptr;      // local block: recv | ret | dispFlags | retFlags
+  size_t heapRecvOff = 0, heapRetOff = 0, heapDFlagOff = 0, heapRFlagOff = 0;
+  size_t heapBytes = 0;
+  void* peerBase[64] = {};   // opened peer heap bases (self = heap)
+  uint64_t* dPeerRecv = nullptr;   // device tables [world]
+  uint64_t* dPeerRet = nullptr;
+  uint64_t* dPeerDFlag = nullptr;
+  uint64_t* dPeerRFlag = nullptr;
+  uint32_t* dArrive = nullptr;     // [2*E] dispatch/return arrival counters
+  unsigned long long seq = 0;
   void* xM = nullptr;        // [nLx_alloc, pEC, P] Element
   float* O32 = nullptr;      // [S, H]
   int nLxAlloc = 0;
@@ -1157,6 +1291,12 @@ int fm_finalize(void) {
   if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
   hipFree(g.tokenIds); hipFree(g.eC); hipFree(g.xM); hipFree(g.O32);
   hipFree(g.logits32); hipFree(g.gML);
+  if (g.heap) {
+    for (int p = 0; p < g.world; ++p) {
+      if (p != g.rank && g.peerBase[p]) hipIpcCloseMemHandle(g.peerBase[p]);
+    }
+    hipFree(g.heap); hipFree(g.dPeerRecv); hipFree(g.dArrive);
+  }
   g = State{};
   return FM_OK;
 }
@@ -1447,6 +1587,122 @@ int fm_expert_ffn(void* stream, const void* rows, const void* expert_w,
   dn.out = out_rows;
   dn.K = g.P; dn.N = g.H;
   return launch_group_gemm(st, 2, dn, (int)n_rows, 1);
+}
+
+static size_t alignUp(size_t v, size_t a) { return (v + a - 1) / a * a; }
+
+int fm_heap_init(void) {
+  if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
+  if (g.heap) return FM_OK;
+  const size_t cells = (size_t)g.E * g.EC * g.H * g.esz;
+  g.heapRecvOff = 0;
+  g.heapRetOff = alignUp(cells, 256);
+  g.heapDFlagOff = alignUp(g.heapRetOff + cells, 256);
+  g.heapRFlagOff = alignUp(g.heapDFlagOff + (size_t)g.E * 8, 256);
+  g.heapBytes = alignUp(g.heapRFlagOff + (size_t)g.E * 8, 256);
+  FM_HIP_CHECK(hipMalloc(&g.heap, g.heapBytes));
+  FM_HIP_CHECK(hipMemset(g.heap, 0, g.heapBytes));
+  FM_HIP_CHECK(hipMalloc(&g.dPeerRecv, 4 * (size_t)g.world * 8));
+  g.dPeerRet = g.dPeerRecv + g.world;
+  g.dPeerDFlag = g.dPeerRet + g.world;
+  g.dPeerRFlag = g.dPeerDFlag + g.world;
+  FM_HIP_CHECK(hipMalloc(&g.dArrive, 2 * (size_t)g.E * 4));
+  return FM_OK;
+}
+
+int fm_heap_handle(void* out64) {
+  if (!g.heap) { setErr("fm_heap_init first"); return FM_ERR_STATE; }
+  hipIpcMemHandle_t h;
+  FM_HIP_CHECK(hipIpcGetMemHandle(&h, g.heap));
+  memcpy(out64, &h, sizeof(h));
+  return FM_OK;
+}
+
+int fm_heap_connect(const void* handles /* world x 64B, null = local only */) {
+  if (!g.heap) { setErr("fm_heap_init first"); return FM_ERR_STATE; }
+  uint64_t hRecv[64], hRet[64], hDF[64], hRF[64];
+  for (int p = 0; p < g.world; ++p) {
+    void* base = nullptr;
+    if (p == g.rank || !handles) {
+      base = g.heap;
+    } else {
+      hipIpcMemHandle_t h;
+      memcpy(&h, reinterpret_cast<const char*>(handles) + (size_t)p * 64,
+             sizeof(h));
+      FM_HIP_CHECK(hipIpcOpenMemHandle(&base, h,
+                                       hipIpcMemLazyEnablePeerAccess));
+    }
+    g.peerBase[p] = base;
+    hRecv[p] = (uint64_t)(uintptr_t)base + g.heapRecvOff;
+    hRet[p] = (uint64_t)(uintptr_t)base + g.heapRetOff;
+    hDF[p] = (uint64_t)(uintptr_t)base + g.heapDFlagOff;
+    hRF[p] = (uint64_t)(uintptr_t)base + g.heapRFlagOff;
+  }
+  FM_HIP_CHECK(hipMemcpy(g.dPeerRecv, hRecv, (size_t)g.world * 8,
+                         hipMemcpyHostToDevice));
+  FM_HIP_CHECK(hipMemcpy(g.dPeerRet, hRet, (size_t)g.world * 8,
+                         hipMemcpyHostToDevice));
+  FM_HIP_CHECK(hipMemcpy(g.dPeerDFlag, hDF, (size_t)g.world * 8,
+                         hipMemcpyHostToDevice));
+  FM_HIP_CHECK(hipMemcpy(g.dPeerRFlag, hRF, (size_t)g.world * 8,
+                         hipMemcpyHostToDevice));
+  return FM_OK;
+}
+
+int fm_heap_ptrs(void** recv, void** ret) {
+  if (!g.heap) { setErr("fm_heap_init first"); return FM_ERR_STATE; }
+  if (recv) *recv = g.heap + g.heapRecvOff;
+  if (ret) *ret = g.heap + g.heapRetOff;
+  return FM_OK;
+}
+
+int fm_dispatch_p2p(void* stream, const void* x) {
+  if (!g.heap) { setErr("fm_heap_init/connect first"); return FM_ERR_STATE; }
+  hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+  g.seq += 1;
+  FM_HIP_CHECK(hipMemsetAsync(g.dArrive, 0, 2 * (size_t)g.E * 4, st));
+  const int nLx = g.E / g.world;
+  dim3 grid(g.EC, g.E);
+#define DP2P(T)                                                               \
+  hipLaunchKernelGGL(k_dispatch_p2p<T>, grid, dim3(128), 0, st,               \
+                     reinterpret_cast<const T*>(x), g.tokenIds, g.eC,         \
+                     g.dPeerRecv, g.dPeerDFlag, g.dArrive, g.H, g.EC, g.pEC,  \
+                     nLx, g.rank, g.seq)
+  if (g.cfg.dtype == 3) DP2P(fp16);
+  else if (g.esz == 2) DP2P(bf16);
+  else DP2P(float);
+#undef DP2P
+  FM_HIP_CHECK(hipGetLastError());
+  // wait for every source's cells for MY experts
+  hipLaunchKernelGGL(
+      k_await_flags, dim3(1), dim3(256), 0, st,
+      reinterpret_cast<const unsigned long long*>(g.heap + g.heapDFlagOff),
+      g.E, g.seq);
+  FM_HIP_CHECK(hipGetLastError());
+  return FM_OK;
+}
+
+int fm_return_p2p(void* stream, const void* ffn_out) {
+  if (!g.heap) { setErr("fm_heap_init/connect first"); return FM_ERR_STATE; }
+  hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+  const int nLx = g.E / g.world;
+  dim3 grid(g.EC, g.world * nLx);
+#define RP2P(T)                                                               \
+  hipLaunchKernelGGL(k_return_p2p<T>, grid, dim3(128), 0, st,                 \
+                     reinterpret_cast<const T*>(ffn_out), g.dPeerRet,         \
+                     g.dPeerRFlag, g.dArrive + g.E, g.H, g.EC, nLx, g.rank,   \
+                     g.seq)
+  if (g.cfg.dtype == 3) RP2P(fp16);
+  else if (g.esz == 2) RP2P(bf16);
+  else RP2P(float);
+#undef RP2P
+  FM_HIP_CHECK(hipGetLastError());
+  hipLaunchKernelGGL(
+      k_await_flags, dim3(1), dim3(256), 0, st,
+      reinterpret_cast<const unsigned long long*>(g.heap + g.heapRFlagOff),
+      g.E, g.seq);
+  FM_HIP_CHECK(hipGetLastError());
+  return FM_OK;
 }
 
 int fm_pack_dispatch(void* stream, const void* x, void* sendbuf) {

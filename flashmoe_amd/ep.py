@@ -17,6 +17,7 @@ semantics the RCCL path uses.
 from __future__ import annotations
 
 import ctypes
+import os
 
 import torch
 
@@ -80,6 +81,80 @@ def _padded_buffers(moe, world, E, EC, H, dtype, dev):
     return cache
 
 
+def _p2p_setup(moe, lib, world, group):
+    """One-time hipIpc heap exchange for the one-sided transport."""
+    import torch.distributed as dist
+
+    from . import _ext
+
+    _ext.check(lib.fm_heap_init(), "fm_heap_init")
+    if world > 1:
+        h = (ctypes.c_char * 64)()
+        _ext.check(lib.fm_heap_handle(h), "fm_heap_handle")
+        gathered = [None] * world
+        dist.all_gather_object(gathered, bytes(h), group=group)
+        blob = (ctypes.c_char * (64 * world)).from_buffer_copy(
+            b"".join(gathered))
+        _ext.check(lib.fm_heap_connect(blob), "fm_heap_connect")
+    else:
+        _ext.check(lib.fm_heap_connect(None), "fm_heap_connect")
+    recv = ctypes.c_void_p()
+    ret = ctypes.c_void_p()
+    _ext.check(lib.fm_heap_ptrs(ctypes.byref(recv), ctypes.byref(ret)),
+               "fm_heap_ptrs")
+    moe._state["p2p"] = {"recv": recv.value, "ret": ret.value}
+    return moe._state["p2p"]
+
+
+def moe_forward_ep_p2p(input, gate_weights, expert_weights, group=None):
+    """EP forward over the one-sided xGMI transport (FLASHMOE_P2P=1):
+    in-kernel stores into peers' heap cells + system-scope signals
+    replace both all_to_alls (os/packet.cuh:214-258 semantics). Heap cell
+    layout is identical to the padded path, so the FFN/combine stages are
+    shared."""
+    import torch.distributed as dist
+
+    from . import _ext, moe
+
+    lib = _ext.load()
+    world = dist.get_world_size(group) if dist.is_initialized() else 1
+    cc = moe.get_compiled_config()
+    S, H, E = cc["S"], cc["H"], cc["E"]
+    cfg = moe._state["cfg"]
+    base = -(-S // E) if cfg["drop_tokens"] else S
+    EC = base * cfg["capacity_factor"] * cfg["expert_top_k"]
+    nLx = E // world
+    gate_out = moe._state["gate_out"]
+    stream = torch.cuda.current_stream().cuda_stream
+    dev = input.device
+    x2d = input.view(S, H)
+    p2p = moe._state.get("p2p") or _p2p_setup(moe, lib, world, group)
+    buf = _padded_buffers(moe, world, E, EC, H, input.dtype, dev)
+
+    _ext.check(lib.fm_gate_forward(
+        ctypes.c_void_p(stream), ctypes.c_void_p(x2d.data_ptr()),
+        ctypes.c_void_p(gate_weights.data_ptr()),
+        ctypes.c_void_p(gate_out.data_ptr()), S), "fm_gate_forward")
+    # one-sided dispatch + in-kernel wait for my experts' cells
+    _ext.check(lib.fm_dispatch_p2p(
+        ctypes.c_void_p(stream), ctypes.c_void_p(x2d.data_ptr())),
+        "fm_dispatch_p2p")
+    _ext.check(lib.fm_expert_ffn_segments(
+        ctypes.c_void_p(stream), ctypes.c_void_p(p2p["recv"]),
+        ctypes.c_void_p(buf["seg_expert"].data_ptr()), world * nLx,
+        ctypes.c_void_p(expert_weights.data_ptr()),
+        ctypes.c_void_p(buf["ffn_out"].data_ptr())), "fm_expert_ffn_segments")
+    _ext.check(lib.fm_return_p2p(
+        ctypes.c_void_p(stream),
+        ctypes.c_void_p(buf["ffn_out"].data_ptr())), "fm_return_p2p")
+    out = torch.empty_like(input)
+    _ext.check(lib.fm_combine_padded(
+        ctypes.c_void_p(stream), ctypes.c_void_p(p2p["ret"]),
+        ctypes.c_void_p(gate_out.data_ptr()),
+        ctypes.c_void_p(out.data_ptr()), S), "fm_combine_padded")
+    return out
+
+
 def moe_forward_ep(input, gate_weights, expert_weights, group=None,
                    _stub_exchange=False):
     """Capacity-padded EP forward (the product multi-GPU path): the
@@ -97,6 +172,8 @@ def moe_forward_ep(input, gate_weights, expert_weights, group=None,
 
     from . import _ext, moe
 
+    if os.environ.get("FLASHMOE_P2P") == "1" and not _stub_exchange:
+        return moe_forward_ep_p2p(input, gate_weights, expert_weights, group)
     lib = _ext.load()
     world = dist.get_world_size(group)
     cc = moe.get_compiled_config()

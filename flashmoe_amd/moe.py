@@ -78,6 +78,8 @@ def finalize():
         raise RuntimeError("finalize() before initialize()")
     _ext.check(_ext.load().fm_finalize(), "fm_finalize")
     _state.update(initialized=False, cfg=None, gate_out=None)
+    _state.pop("p2p", None)        # heap freed by fm_finalize
+    _state.pop("ep_buffers", None)
 
 
 def get_compiled_config() -> dict:

@@ -175,6 +175,23 @@ int fm_moe_forward_phased(void* stream, const void* x, const void* gate_w,
  * (replaces the reference's in-kernel decode, os/packet.cuh:288-455). */
 int fm_export_routing(void* stream, void* routed_dev, void* tps_dev);
 
+/* --- Opt-in one-sided P2P transport for the EP exchange (the
+ * reference's intra-node mode: direct stores into hipIpc-mapped peer
+ * heap cells + system-scope 8-byte seq-tagged signals,
+ * os/packet.cuh:214-258; heap cell layout = the padded pipeline's).
+ * Enable from Python with FLASHMOE_P2P=1; the all_to_all path stays the
+ * default until multi-GPU validation. --- */
+int fm_heap_init(void);
+int fm_heap_handle(void* out64 /* hipIpcMemHandle_t, 64 B */);
+int fm_heap_connect(const void* handles /* world x 64 B; NULL = self only */);
+int fm_heap_ptrs(void** recv /* [world,nLx,EC,H] */, void** ret /* [E,EC,H] */);
+/* write routed rows into owners' heaps + signal; then wait for all
+ * sources' cells for MY experts (bounded spin; prints on timeout) */
+int fm_dispatch_p2p(void* stream, const void* x);
+/* write FFN results back into sources' return heaps + signal; wait for
+ * my own return cells */
+int fm_return_p2p(void* stream, const void* ffn_out);
+
 /* Training-mode auxiliary-loss accumulators (gate.cuh:273-299,763-773;
  * types.cuh:936-958): gML[e] = mean softmax prob of expert e over the
  * last forward's tokens, gMeC[e] = fraction routed to e (pre-capacity).

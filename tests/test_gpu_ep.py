@@ -104,6 +104,48 @@ def test_staged_entry_points_match_oracle():
         moe.finalize()
 
 
+def test_p2p_transport_world1_matches_single_rank():
+    """The one-sided heap transport (FLASHMOE_P2P path) at world 1:
+    in-kernel store + system-scope signal + bounded-spin wait kernels,
+    peer = self (hipIpc mapping itself needs >1 GPU; the driver's scale
+    run covers that with FLASHMOE_P2P=1)."""
+    from flashmoe_amd import ep, moe
+
+    cfg = {
+        "capacity_factor": 2, "drop_tokens": 1, "expert_top_k": 2,
+        "global_batch": 256, "is_training": 0, "hidden_act": 0,
+        "hidden_size": 128, "intermediate_size": 256, "mini_batch": 1,
+        "moe_frequency": 1, "num_experts": 8, "num_layers": 1,
+        "sequence_len": 512, "torch_dtype": 2, "vocab_size": 32000,
+    }
+    with tempfile.NamedTemporaryFile("w", suffix=".json", delete=False) as f:
+        json.dump(cfg, f)
+        path = f.name
+    moe.initialize(path, rank=0, world_size=1)
+    try:
+        S, H, P, E = 512, 128, 256, 8
+        torch.manual_seed(5)
+        x = torch.randn(1, S, H, dtype=torch.bfloat16, device="cuda")
+        gw = torch.randn(H, E, dtype=torch.bfloat16, device="cuda")
+        ew = torch.randn(E, 2, P, H, dtype=torch.bfloat16, device="cuda")
+        want = moe.moe_forward(x, gw, ew)
+        got = ep.moe_forward_ep_p2p(x, gw, ew)
+        torch.cuda.synchronize()
+        a = got.float().cpu().numpy()
+        b = want.float().cpu().numpy()
+        scale = max(1.0, float(np.abs(b).max()))
+        assert np.allclose(a, b, rtol=2e-2, atol=2e-3 * scale), (
+            float(np.abs(a - b).max()))
+        # run it twice more: seq-tagged flags must keep working without
+        # re-zeroing across calls
+        for _ in range(2):
+            got2 = ep.moe_forward_ep_p2p(x, gw, ew)
+        torch.cuda.synchronize()
+        assert torch.equal(got2, got)
+    finally:
+        moe.finalize()
+
+
 def test_ep_pipeline_under_torchrun():
     """Full EP pipeline over torch.distributed (RCCL) at world = #GPUs."""
     n = torch.cuda.device_count()
