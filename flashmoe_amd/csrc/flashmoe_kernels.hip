@@ -322,13 +322,17 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
   TPS* sTps = reinterpret_cast<TPS*>(Blds + BN * BK);
   uint32_t* sRouted = reinterpret_cast<uint32_t*>(sTps + BM);
 
-  const int e = blockIdx.z;
+  const int skS = a.splitK > 0 ? a.splitK : 1;
+  const int e = blockIdx.z % (gridDim.z / skS);
+  const int ksplitS = blockIdx.z / (gridDim.z / skS);
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
   const int m0 = blockIdx.x * BM;
   const int n0 = blockIdx.y * BN;
   const int K = a.K, N = a.N;
+  const int kLenS = K / skS;
+  const int kStartS = ksplitS * kLenS;
 
   const int we = a.segExpert ? a.segExpert[e] : e;  // weight/bias expert
   const TPS* tpsE = a.tokenIds ? a.tokenIds + (size_t)e * a.pEC : nullptr;
@@ -367,7 +371,7 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
   const int schunk = (lane & 7) ^ grow8;  // swizzled 16B chunk index
   const int aRowStride = (PHASE == 0) ? a.H : K;
   const size_t aBase = (size_t)e * a.strideAExpert;  // 0 for the x-gather up phase
-  for (int kt = 0; kt < K; kt += BK) {
+  for (int kt = kStartS; kt < kStartS + kLenS; kt += BK) {
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       const int grp = wave * 4 + i;
@@ -472,8 +476,11 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
           reinterpret_cast<ET*>(a.out)[(size_t)e * a.strideOExpert +
                                        (size_t)m * N + col] =
               ETr<ET>::fromf(v);
-        } else {  // PHASE 3: gate logits, fp32
-          reinterpret_cast<float*>(a.out)[(size_t)m * N + col] = v;
+        } else {  // PHASE 3: gate logits, fp32 (atomic when K-split)
+          if (skS > 1)
+            atomicAdd(&reinterpret_cast<float*>(a.out)[(size_t)m * N + col], v);
+          else
+            reinterpret_cast<float*>(a.out)[(size_t)m * N + col] = v;
         }
       }
     }
@@ -1348,14 +1355,24 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
     dim3 block(mode == 2 ? 256 : 512);
     dim3 grid(DIVUP(M, mode == 2 ? 128 : 256),
               DIVUP(N, mode == 1 ? 128 : (mode == 0 ? 256 : 128)), nE * skf);
-    if (phase == 3) {  // gate logits: small kernel, no bias/act
-      dim3 g3(DIVUP(M, 128), DIVUP(N, 128), 1);
+    if (phase == 3) {  // gate logits: small kernel, no bias/act; K-split
+      // to fill the chip (the logits grid is only S/128 x E/128 blocks)
+      const int b3 = DIVUP(M, 128) * DIVUP(N, 128);
+      int sk3 = 1;
+      while (sk3 < 8 && b3 * sk3 * 2 <= 512 && (a.K / 64) % (sk3 * 2) == 0)
+        sk3 *= 2;
+      GemmArgs a3 = a;
+      a3.splitK = sk3;
+      if (sk3 > 1)
+        FM_HIP_CHECK(hipMemsetAsync(a3.out, 0,
+                                    (size_t)M * a.N * sizeof(float), st));
+      dim3 g3(DIVUP(M, 128), DIVUP(N, 128), sk3);
       if (g.cfg.dtype == 3)
         hipLaunchKernelGGL((k_group_gemm_bf16<fp16, 3, 0, false>), g3,
-                           dim3(256), 0, st, a);
+                           dim3(256), 0, st, a3);
       else
         hipLaunchKernelGGL((k_group_gemm_bf16<bf16, 3, 0, false>), g3,
-                           dim3(256), 0, st, a);
+                           dim3(256), 0, st, a3);
       FM_HIP_CHECK(hipGetLastError());
       return FM_OK;
     }
