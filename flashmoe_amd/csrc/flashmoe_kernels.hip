@@ -498,13 +498,14 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
 // Same phase semantics/epilogues as k_group_gemm_bf16.
 // ---------------------------------------------------------------------------
 
-template <typename ET, int PHASE, int ACT, bool HAS_BIAS, int BN>
+template <typename ET, int PHASE, int ACT, bool HAS_BIAS, int BN, int BM = 256>
 __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
   using vec8 = typename ETr<ET>::vec8;
-  constexpr int BM = 256, BK = 64;
+  constexpr int BK = 64;
   constexpr int NF = BN / 64;           // B fragments per wave (4 or 2)
-  constexpr int AGRP = BM * BK * 2 / 1024;  // glds 1KiB groups per A tile (32)
-  constexpr int BGRP = BN * BK * 2 / 1024;  // (16 or 8)
+  constexpr int MI = BM / 32;           // A fragments per wave (8 or 4)
+  constexpr int AGRP = BM * BK * 2 / 1024;  // glds 1KiB groups per A tile
+  constexpr int BGRP = BN * BK * 2 / 1024;
   constexpr int GPW_A = AGRP / 8, GPW_B = BGRP / 8;  // per wave
   constexpr int GPT = GPW_A + GPW_B;    // glds per wave per K-tile
   __shared__ __attribute__((aligned(16))) char smem[
@@ -596,9 +597,9 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
   };
 
   const int wr = wave >> 2, wc = wave & 3;  // 2M x 4N wave grid
-  f32x4 accv[8][NF];
+  f32x4 accv[MI][NF];
 #pragma unroll
-  for (int i = 0; i < 8; ++i)
+  for (int i = 0; i < MI; ++i)
 #pragma unroll
     for (int j = 0; j < NF; ++j) accv[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
@@ -630,11 +631,11 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
         stage((t + 1) * BK, (t + 1) & 1);
         __builtin_amdgcn_s_setprio(1);
       }
-      vec8 af[8], bfr[NF];
+      vec8 af[MI], bfr[NF];
       const int rl = lane & 15;
       const int cbase = 4 * s + (lane >> 4);
       auto aread = [&](int mi) {
-        const int R = wr * 128 + mi * 16 + rl;
+        const int R = wr * (BM / 2) + mi * 16 + rl;
         af[mi] = *reinterpret_cast<const vec8*>(
             &Al[R * BK + ((cbase ^ (R & 7)) * 8)]);
       };
@@ -652,8 +653,8 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
       aread(1);
       __builtin_amdgcn_sched_barrier(0);
 #pragma unroll
-      for (int g2 = 0; g2 < 4; ++g2) {
-        if (g2 < 3) {
+      for (int g2 = 0; g2 < MI / 2; ++g2) {
+        if (g2 < MI / 2 - 1) {
           aread(2 * g2 + 2);
           aread(2 * g2 + 3);
         }
@@ -703,10 +704,10 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
     }
   }
 #pragma unroll
-  for (int mi = 0; mi < 8; ++mi) {
+  for (int mi = 0; mi < MI; ++mi) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int row = wr * 128 + mi * 16 + r0 + r;
+      const int row = wr * (BM / 2) + mi * 16 + r0 + r;
       const int m = m0 + row;
       if ((uint32_t)m >= routed) continue;
       const TPS tp = sTps[row];
@@ -1335,26 +1336,31 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
     // split-K restores occupancy at BN=256's lower staging traffic.
     const int b256 = DIVUP(M, 256) * DIVUP(N, 256) * nE;
     const int b128n = DIVUP(M, 256) * DIVUP(N, 128) * nE;
-    int mode;  // 0: big BN=256, 1: big BN=128, 2: small 128x128
+    const int h256 = DIVUP(M, 128) * DIVUP(N, 256) * nE;  // BM=128 rows
+    const int h128 = DIVUP(M, 128) * DIVUP(N, 128) * nE;
+    // modes: 0 big(BMxBN 256x256), 1 big(256x128), 3 big(128x256),
+    // 4 big(128x128), 2 small synchronous 128x128. Prefer the largest
+    // tile whose grid still fills the chip; the BM=128 pipelined tiles
+    // cover the many-experts shapes where per-expert M == pEC == 128
+    // (configs 3-5 shapes and the EP segments).
+    // NOTE: split-K via the atomic combine epilogue is wired up
+    // (a.splitK) but DISABLED by default: at config 2 it doubled the
+    // 8.3M fp32 atomics and cost more (+13 us) than the BN=256 staging
+    // saving bought (profiles/r01). Revisit for huge-K shapes.
+    int mode;
     int skf = 1;
-    if (M >= 256 && b256 >= 256) {
-      mode = 0;
-    } else if (M >= 256) {
-      // NOTE: split-K via the atomic combine epilogue is wired up
-      // (a.splitK) but DISABLED by default: at config 2 it doubled the
-      // 8.3M fp32 atomics and cost more (+13 us) than the BN=256
-      // staging saving bought (profiles/r01). Revisit for huge-K shapes.
-      if (skf > 1) mode = 0;
-      else if (b128n >= 256) mode = 1;
-      else mode = 2;
-    } else {
-      mode = 2;
-    }
+    if (M >= 256 && b256 >= 256) mode = 0;
+    else if (M >= 256 && b128n >= 256) mode = 1;
+    else if (M >= 128 && h256 >= 256) mode = 3;
+    else if (M >= 128 && h128 >= 256) mode = 4;
+    else mode = 2;
     GemmArgs aa = a;
     aa.splitK = skf;
+    const int bmSel = (mode == 0 || mode == 1) ? 256 : 128;
+    const int bnSel = (mode == 0 || mode == 3) ? 256 : 128;
     dim3 block(mode == 2 ? 256 : 512);
-    dim3 grid(DIVUP(M, mode == 2 ? 128 : 256),
-              DIVUP(N, mode == 1 ? 128 : (mode == 0 ? 256 : 128)), nE * skf);
+    dim3 grid(DIVUP(M, mode == 2 ? 128 : bmSel),
+              DIVUP(N, mode == 2 ? 128 : bnSel), nE * skf);
     if (phase == 3) {  // gate logits: small kernel, no bias/act; K-split
       // to fill the chip (the logits grid is only S/128 x E/128 blocks)
       const int b3 = DIVUP(M, 128) * DIVUP(N, 128);
@@ -1380,10 +1386,16 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
 #define GG_ET(ET, PH, AC, HB)                                                 \
     do {                                                                      \
       if (mode == 0)                                                          \
-        hipLaunchKernelGGL((k_group_gemm_bf16_big<ET, PH, AC, HB, 256>),      \
+        hipLaunchKernelGGL((k_group_gemm_bf16_big<ET, PH, AC, HB, 256, 256>), \
                            grid, block, 0, st, aa);                           \
       else if (mode == 1)                                                     \
-        hipLaunchKernelGGL((k_group_gemm_bf16_big<ET, PH, AC, HB, 128>),      \
+        hipLaunchKernelGGL((k_group_gemm_bf16_big<ET, PH, AC, HB, 128, 256>), \
+                           grid, block, 0, st, aa);                           \
+      else if (mode == 3)                                                     \
+        hipLaunchKernelGGL((k_group_gemm_bf16_big<ET, PH, AC, HB, 256, 128>), \
+                           grid, block, 0, st, aa);                           \
+      else if (mode == 4)                                                     \
+        hipLaunchKernelGGL((k_group_gemm_bf16_big<ET, PH, AC, HB, 128, 128>), \
                            grid, block, 0, st, aa);                           \
       else                                                                    \
         hipLaunchKernelGGL((k_group_gemm_bf16<ET, PH, AC, HB>), grid, block,  \
