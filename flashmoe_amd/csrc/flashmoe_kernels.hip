@@ -1351,6 +1351,11 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
     int skf = 1;
     if (M >= 256 && b256 >= 256) mode = 0;
     else if (M >= 256 && b128n >= 256) mode = 1;
+    // 128x128 runs 2 blocks/CU (65 KB LDS, 71 VGPR) so block-level
+    // overlap hides the per-tile sync; prefer it when the grid can
+    // fill 2/CU AND K is not huge (at K=14336 the 1.33x staging of the
+    // smaller tile outweighs the overlap - measured on the cfg4 shape)
+    else if (M >= 128 && h128 >= 512 && a.K <= 8192) mode = 4;
     else if (M >= 128 && h256 >= 256) mode = 3;
     else if (M >= 128 && h128 >= 256) mode = 4;
     else mode = 2;
