@@ -25,11 +25,18 @@
 using bf16 = __hip_bfloat16;
 
 // token slot descriptor (reference TPS, types.cuh:299-312): token index +
-// sum of the token's selected top-k probabilities (gate.cuh:669,711-715)
+// sum of the token's selected top-k probabilities (gate.cuh:669,711-715).
+// The top 4 bits of tokenIdx carry the assignment's position j within the
+// token's top-k (tokens <= 2^28; lets the combine write non-atomic
+// per-(token, j) slots instead of fp32 atomics).
 struct __align__(8) TPS {
   uint32_t tokenIdx;
   float probSum;
 };
+__device__ __host__ __forceinline__ uint32_t tpsTok(uint32_t v) {
+  return v & 0x0FFFFFFFu;
+}
+__device__ __host__ __forceinline__ uint32_t tpsJ(uint32_t v) { return v >> 28; }
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(8))) _Float16 halfx8;
@@ -176,7 +183,7 @@ __global__ __launch_bounds__(256) void k_gate_route(
     const float* __restrict__ logits32, T* __restrict__ gate_out,
     TPS* __restrict__ tokenIds, uint32_t* __restrict__ eC, int S, int E,
     int PX, int EC, int pEC, float* __restrict__ gML,
-    float* __restrict__ gMeC) {
+    float* __restrict__ gMeC, uint8_t* __restrict__ kept) {
   // gML/gMeC: training-mode aux-loss accumulators (gate.cuh:273-299,
   // 763-773; types.cuh:936-958): gML[e] += column-mean of softmax probs,
   // gMeC[e] += routed fraction. Null in inference mode.
@@ -250,8 +257,12 @@ __global__ __launch_bounds__(256) void k_gate_route(
     for (int i = 0; i < K; ++i) {
       const int e = mySel[i];
       const uint32_t slot = base[e] + localIdx[tid * K + i];
-      if (slot < (uint32_t)EC)
-        tokenIds[(size_t)e * pEC + slot] = TPS{(uint32_t)(m0 + tid), mCw};
+      const bool keep = slot < (uint32_t)EC;
+      if (keep) {
+        tokenIds[(size_t)e * pEC + slot] =
+            TPS{(uint32_t)(m0 + tid) | ((uint32_t)i << 28), mCw};
+      }
+      if (kept) kept[(size_t)(m0 + tid) * K + i] = keep ? 1 : 0;
     }
   }
 }
@@ -376,7 +387,7 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
     for (int i = 0; i < 4; ++i) {
       const int grp = wave * 4 + i;
       const int row = grp * 8 + grow8;
-      const size_t arow = (PHASE == 0) ? (size_t)sTps[row].tokenIdx
+      const size_t arow = (PHASE == 0) ? (size_t)tpsTok(sTps[row].tokenIdx)
                                        : (size_t)(m0 + row);
       const ET* asrc = Ag + aBase + arow * aRowStride + kt + schunk * 8;
       __builtin_amdgcn_global_load_lds(
@@ -429,7 +440,7 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
         float sc = 0.0f;
         if ((uint32_t)(m0 + tid) < routed)
           sc = toF(reinterpret_cast<const ET*>(
-                   a.gate_out)[(size_t)tp.tokenIdx * a.PX +
+                   a.gate_out)[(size_t)tpsTok(tp.tokenIdx) * a.PX +
                                a.expertOffset + e]) / tp.probSum;
         sScale[tid] = sc;
       }
@@ -466,10 +477,14 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
               ETr<ET>::fromf(v);
         } else if constexpr (PHASE == 1) {
           if (multi) {
-            atomicAdd(&a.O32[(size_t)tp.tokenIdx * a.H + col], v * rowScale);
+            // non-atomic per-(token, j) combine slot (summed with the
+            // kept mask in k_cast_combine; replaces fp32 atomics)
+            reinterpret_cast<ET*>(a.O32)[
+                ((size_t)tpsTok(tp.tokenIdx) * a.topk + tpsJ(tp.tokenIdx)) *
+                    a.H + col] = ETr<ET>::fromf(v * rowScale);
           } else {
             reinterpret_cast<ET*>(
-                a.moe_out)[(size_t)tp.tokenIdx * a.H + col] =
+                a.moe_out)[(size_t)tpsTok(tp.tokenIdx) * a.H + col] =
                 ETr<ET>::fromf(v);
           }
         } else if constexpr (PHASE == 2) {  // packed-rows direct output
@@ -573,7 +588,7 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
 #pragma unroll
   for (int i = 0; i < GPW_A; ++i) {
     const int row = (wave * GPW_A + i) * 8 + grow8;
-    const size_t arow = (PHASE == 0) ? (size_t)sTps[row].tokenIdx
+    const size_t arow = (PHASE == 0) ? (size_t)tpsTok(sTps[row].tokenIdx)
                                      : (size_t)min(m0 + row, mCap - 1);
     aSrc[i] = Ag + aBase + arow * aRowStride + kStart + schunk * 8;
   }
@@ -684,7 +699,7 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
         float sc = 0.0f;
         if ((uint32_t)(m0 + tid) < routed)
           sc = toF(reinterpret_cast<const ET*>(
-                   a.gate_out)[(size_t)tp.tokenIdx * a.PX +
+                   a.gate_out)[(size_t)tpsTok(tp.tokenIdx) * a.PX +
                                a.expertOffset + e]) / tp.probSum;
         sScale[tid] = sc;
       }
@@ -725,10 +740,14 @@ __global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
               ETr<ET>::fromf(v);
         } else if constexpr (PHASE == 1) {
           if (multi) {
-            atomicAdd(&a.O32[(size_t)tp.tokenIdx * a.H + col], v * rowScale);
+            // non-atomic per-(token, j) combine slot (summed with the
+            // kept mask in k_cast_combine; replaces fp32 atomics)
+            reinterpret_cast<ET*>(a.O32)[
+                ((size_t)tpsTok(tp.tokenIdx) * a.topk + tpsJ(tp.tokenIdx)) *
+                    a.H + col] = ETr<ET>::fromf(v * rowScale);
           } else {
             reinterpret_cast<ET*>(
-                a.moe_out)[(size_t)tp.tokenIdx * a.H + col] =
+                a.moe_out)[(size_t)tpsTok(tp.tokenIdx) * a.H + col] =
                 ETr<ET>::fromf(v);
           }
         } else {
@@ -790,7 +809,7 @@ __global__ __launch_bounds__(256) void k_group_gemm_f32(GemmArgs a) {
       const int u = tid + u4 * 256;
       const int row = u / BK, cc = u % BK;
       const float* src = (PHASE == 0)
-          ? Ag + (size_t)sTps[row].tokenIdx * a.H + kt + cc
+          ? Ag + (size_t)tpsTok(sTps[row].tokenIdx) * a.H + kt + cc
           : Ag + (size_t)e * a.strideAExpert + (size_t)(m0 + row) * K + kt + cc;
       Alds[row * LDT + cc] = *src;
       const int brow = min(n0 + row, N - 1);
@@ -829,7 +848,7 @@ __global__ __launch_bounds__(256) void k_group_gemm_f32(GemmArgs a) {
     float rowScale = 1.0f;
     if (multi)
       rowScale = reinterpret_cast<const float*>(
-                     a.gate_out)[(size_t)tp.tokenIdx * a.PX +
+                     a.gate_out)[(size_t)tpsTok(tp.tokenIdx) * a.PX +
                                  a.expertOffset + e] / tp.probSum;
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
@@ -843,16 +862,39 @@ __global__ __launch_bounds__(256) void k_group_gemm_f32(GemmArgs a) {
                                         (size_t)m * N + col] = v;
       } else if constexpr (PHASE == 1) {
         if (multi) {
-          atomicAdd(&a.O32[(size_t)tp.tokenIdx * a.H + col], v * rowScale);
+          reinterpret_cast<float*>(a.O32)[
+              ((size_t)tpsTok(tp.tokenIdx) * a.topk + tpsJ(tp.tokenIdx)) *
+                  a.H + col] = v * rowScale;
         } else {
-          reinterpret_cast<float*>(a.moe_out)[(size_t)tp.tokenIdx * a.H +
-                                              col] = v;
+          reinterpret_cast<float*>(
+              a.moe_out)[(size_t)tpsTok(tp.tokenIdx) * a.H + col] = v;
         }
       } else {
         reinterpret_cast<float*>(a.out)[(size_t)e * a.strideOExpert +
                                         (size_t)m * N + col] = v;
       }
     }
+  }
+}
+
+// sum the k non-atomic combine slots (kept-masked) into the output
+// (k>1 path; replaces the fp32-atomic accumulator + cast)
+template <typename T, int K>
+__global__ void k_cast_combine(const T* __restrict__ cbuf,
+                               const uint8_t* __restrict__ kept,
+                               T* __restrict__ out, int S, int H) {
+  const size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  const size_t n = (size_t)S * H;
+  for (size_t x = i; x < n; x += stride) {
+    const size_t t = x / H, h = x % H;
+    float acc = 0.0f;
+#pragma unroll
+    for (int j = 0; j < K; ++j) {
+      if (kept[t * K + j])
+        acc += toF(cbuf[((size_t)t * K + j) * H + h]);
+    }
+    fromF(acc, out[x]);
   }
 }
 
@@ -904,7 +946,7 @@ __global__ void k_pack_dispatch(const T* __restrict__ x,
   const int i = blockIdx.x;
   const uint32_t r = min(eC[e], (uint32_t)EC);
   if ((uint32_t)i >= r) return;
-  const uint32_t tok = tokenIds[(size_t)e * pEC + i].tokenIdx;
+  const uint32_t tok = tpsTok(tokenIds[(size_t)e * pEC + i].tokenIdx);
   constexpr int EPU = 16 / sizeof(T);  // elements per 16B unit
   const T* src = x + (size_t)tok * H;
   T* dst = sendbuf + ((size_t)e * EC + i) * H;
@@ -924,20 +966,23 @@ __global__ void k_combine_padded(const T* __restrict__ rows,
                                  const TPS* __restrict__ tokenIds,
                                  const uint32_t* __restrict__ eC,
                                  const T* __restrict__ gate_out,
-                                 float* __restrict__ O32, int H, int EC,
+                                 T* __restrict__ cbuf, int H, int EC,
                                  int pEC, int PX, int topk) {
   const int e = blockIdx.y;
   const int i = blockIdx.x;
   const uint32_t r = min(eC[e], (uint32_t)EC);
   if ((uint32_t)i >= r) return;
   const TPS tp = tokenIds[(size_t)e * pEC + i];
+  const uint32_t tok = tpsTok(tp.tokenIdx);
+  const uint32_t j = tpsJ(tp.tokenIdx);
   const float sc = (topk > 1)
-      ? toF(gate_out[(size_t)tp.tokenIdx * PX + e]) / tp.probSum
+      ? toF(gate_out[(size_t)tok * PX + e]) / tp.probSum
       : 1.0f;
   const T* src = rows + ((size_t)e * EC + i) * H;
-  float* dst = O32 + (size_t)tp.tokenIdx * H;
+  // non-atomic per-(token, j) combine slot (k_cast_combine sums them)
+  T* dst = cbuf + ((size_t)tok * topk + j) * H;
   for (int h = threadIdx.x; h < H; h += blockDim.x) {
-    atomicAdd(dst + h, sc * toF(src[h]));
+    fromF(sc * toF(src[h]), dst[h]);
   }
 }
 
@@ -976,7 +1021,7 @@ __global__ void k_dispatch_p2p(const T* __restrict__ x,
   const int le = e % nLx;
   T* heap = reinterpret_cast<T*>(peerRecv[owner]);
   if ((uint32_t)i < r) {
-    const uint32_t tok = tokenIds[(size_t)e * pEC + i].tokenIdx;
+    const uint32_t tok = tpsTok(tokenIds[(size_t)e * pEC + i].tokenIdx);
     const T* src = x + (size_t)tok * H;
     T* dst = heap + (((size_t)myRank * nLx + le) * EC + i) * H;
     constexpr int EPU = 16 / sizeof(T);
@@ -1076,7 +1121,7 @@ __global__ void k_export_routing(const TPS* __restrict__ tokenIds,
   if (threadIdx.x == 0) routed[e] = r;
   for (int i = threadIdx.x; i < EC; i += blockDim.x) {
     TPS t = (i < (int)r) ? tokenIds[(size_t)e * pEC + i] : TPS{0u, 0.0f};
-    out[((size_t)e * EC + i) * 2] = t.tokenIdx;
+    out[((size_t)e * EC + i) * 2] = tpsTok(t.tokenIdx);
     out[((size_t)e * EC + i) * 2 + 1] = __float_as_uint(t.probSum);
   }
 }
@@ -1132,7 +1177,9 @@ struct State {
   uint32_t* dArrive = nullptr;     // [2*E] dispatch/return arrival counters
   unsigned long long seq = 0;
   void* xM = nullptr;        // [nLx_alloc, pEC, P] Element
-  float* O32 = nullptr;      // [S, H]
+  float* O32 = nullptr;      // [S, H] (staged-API combine accumulator)
+  void* cbuf = nullptr;      // [S, k, H] Element: non-atomic combine slots
+  uint8_t* kept = nullptr;   // [S, k] capacity-kept mask (gate-written)
   int nLxAlloc = 0;
 };
 State g;
@@ -1199,7 +1246,7 @@ int launch_gate(hipStream_t st, const void* x, const void* gate_w,
     hipLaunchKernelGGL((k_gate_route<T, KK>), dim3(tiles), dim3(256), ldsR,   \
                        st, g.logits32, reinterpret_cast<T*>(gate_out),        \
                        g.tokenIds, g.eC, (int)S, g.E, g.PX, g.EC, g.pEC,      \
-                       gMLp, gMeCp);                                          \
+                       gMLp, gMeCp, g.kept);                                  \
   } while (0)
 #define GATE_K(T)                                                             \
   switch (g.cfg.expert_top_k) {                                               \
@@ -1291,6 +1338,9 @@ int fm_initialize(const fm_config* cfg, int rank, int world_size) {
   FM_HIP_CHECK(hipMalloc(&g.xM, (size_t)g.nLxAlloc * g.pEC * g.P * g.esz));
   FM_HIP_CHECK(hipMalloc(&g.O32, (size_t)g.S * g.H * sizeof(float)));
   FM_HIP_CHECK(hipMemset(g.O32, 0, (size_t)g.S * g.H * sizeof(float)));
+  FM_HIP_CHECK(hipMalloc(&g.cbuf,
+                         (size_t)g.S * g.cfg.expert_top_k * g.H * g.esz));
+  FM_HIP_CHECK(hipMalloc(&g.kept, (size_t)g.S * g.cfg.expert_top_k));
   g.initialized = true;
   return FM_OK;
 }
@@ -1298,6 +1348,7 @@ int fm_initialize(const fm_config* cfg, int rank, int world_size) {
 int fm_finalize(void) {
   if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
   hipFree(g.tokenIds); hipFree(g.eC); hipFree(g.xM); hipFree(g.O32);
+  hipFree(g.cbuf); hipFree(g.kept);
   hipFree(g.logits32); hipFree(g.gML);
   if (g.heap) {
     for (int p = 0; p < g.world; ++p) {
@@ -1445,6 +1496,29 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
   return FM_OK;
 }
 
+static int launch_cast_combine(hipStream_t st, void* moe_out) {
+  const size_t n = (size_t)g.S * g.H;
+  const int blocks = (int)std::min((size_t)4096, (size_t)DIVUP(n, (size_t)256 * 4));
+#define CC(T, KK)                                                             \
+  hipLaunchKernelGGL((k_cast_combine<T, KK>), dim3(blocks), dim3(256), 0, st, \
+                     reinterpret_cast<const T*>(g.cbuf), g.kept,              \
+                     reinterpret_cast<T*>(moe_out), g.S, g.H)
+#define CC_K(T)                                                               \
+  switch (g.cfg.expert_top_k) {                                               \
+    case 2: CC(T, 2); break;                                                  \
+    case 4: CC(T, 4); break;                                                  \
+    case 8: CC(T, 8); break;                                                  \
+    default: setErr("bad topk"); return FM_ERR_UNSUPPORTED;                   \
+  }
+  if (g.cfg.dtype == 3) { CC_K(fp16) }
+  else if (g.esz == 2) { CC_K(bf16) }
+  else { CC_K(float) }
+#undef CC_K
+#undef CC
+  FM_HIP_CHECK(hipGetLastError());
+  return FM_OK;
+}
+
 static int moe_forward_impl(hipStream_t st, const void* x, const void* gate_w,
                             const void* expert_w, const void* b_up,
                             const void* b_dn, void* gate_out, void* moe_out,
@@ -1487,7 +1561,7 @@ static int moe_forward_impl(hipStream_t st, const void* x, const void* gate_w,
   dn.B = reinterpret_cast<const char*>(expert_w) + (size_t)g.P * g.H * g.esz;
   dn.bias = b_dn;
   dn.out = nullptr;
-  dn.O32 = g.O32;
+  dn.O32 = reinterpret_cast<float*>(g.cbuf);  // non-atomic combine slots
   dn.moe_out = moe_out;
   dn.strideAExpert = (long long)g.pEC * g.P;
   dn.K = g.P; dn.N = g.H;
@@ -1496,18 +1570,8 @@ static int moe_forward_impl(hipStream_t st, const void* x, const void* gate_w,
   if (evs) FM_HIP_CHECK(hipEventRecord(evs[4], st));
 
   if (g.cfg.expert_top_k > 1) {
-    const size_t n = (size_t)g.S * g.H;
-    const int blocks = (int)std::min((size_t)2048, (size_t)DIVUP(n, (size_t)256 * 8));
-    if (g.cfg.dtype == 3)
-      hipLaunchKernelGGL(k_cast_out<fp16>, dim3(blocks), dim3(256), 0, st,
-                         g.O32, reinterpret_cast<fp16*>(moe_out), n);
-    else if (g.esz == 2)
-      hipLaunchKernelGGL(k_cast_out<bf16>, dim3(blocks), dim3(256), 0, st,
-                         g.O32, reinterpret_cast<bf16*>(moe_out), n);
-    else
-      hipLaunchKernelGGL(k_cast_out<float>, dim3(blocks), dim3(256), 0, st,
-                         g.O32, reinterpret_cast<float*>(moe_out), n);
-    FM_HIP_CHECK(hipGetLastError());
+    int rc2 = launch_cast_combine(st, moe_out);
+    if (rc2 != FM_OK) return rc2;
   }
   return FM_OK;
 }
@@ -1582,7 +1646,7 @@ int fm_read_routing(void* stream, uint32_t* routed_counts, uint32_t* token_idx,
     const uint32_t r = hEC[e] < (uint32_t)g.EC ? hEC[e] : (uint32_t)g.EC;
     routed_counts[e] = r;
     for (uint32_t i = 0; i < r; ++i) {
-      token_idx[(size_t)e * g.EC + i] = hT[(size_t)e * g.pEC + i].tokenIdx;
+      token_idx[(size_t)e * g.EC + i] = tpsTok(hT[(size_t)e * g.pEC + i].tokenIdx);
       prob_sum[(size_t)e * g.EC + i] = hT[(size_t)e * g.pEC + i].probSum;
     }
   }
@@ -1766,30 +1830,38 @@ int fm_combine_padded(void* stream, const void* returned, const void* gate_out,
   hipStream_t st = reinterpret_cast<hipStream_t>(stream);
   dim3 grid(g.EC, g.E);
   const int tk = g.cfg.expert_top_k;
-  if (g.cfg.dtype == 3)
-    hipLaunchKernelGGL(k_combine_padded<fp16>, grid, dim3(256), 0, st,
-                       reinterpret_cast<const fp16*>(returned), g.tokenIds,
-                       g.eC, reinterpret_cast<const fp16*>(gate_out), g.O32,
-                       g.H, g.EC, g.pEC, g.PX, tk);
-  else if (g.esz == 2)
-    hipLaunchKernelGGL(k_combine_padded<bf16>, grid, dim3(256), 0, st,
-                       reinterpret_cast<const bf16*>(returned), g.tokenIds,
-                       g.eC, reinterpret_cast<const bf16*>(gate_out), g.O32,
-                       g.H, g.EC, g.pEC, g.PX, tk);
-  else
-    hipLaunchKernelGGL(k_combine_padded<float>, grid, dim3(256), 0, st,
-                       reinterpret_cast<const float*>(returned), g.tokenIds,
-                       g.eC, reinterpret_cast<const float*>(gate_out), g.O32,
-                       g.H, g.EC, g.pEC, g.PX, tk);
+  if (tk == 1) {
+    // k==1: unscaled direct write path via the slot buffer would be
+    // redundant; write slots with scale 1 and reduce with the mask
+    // (kept mask is exact, so semantics match the reference's
+    // CombineMode::single overwrite of a zeroed output)
+  }
+#define KCP(T)                                                                \
+  hipLaunchKernelGGL(k_combine_padded<T>, grid, dim3(256), 0, st,             \
+                     reinterpret_cast<const T*>(returned), g.tokenIds, g.eC,  \
+                     reinterpret_cast<const T*>(gate_out),                    \
+                     reinterpret_cast<T*>(g.cbuf), g.H, g.EC, g.pEC, g.PX, tk)
+  if (g.cfg.dtype == 3) KCP(fp16);
+  else if (g.esz == 2) KCP(bf16);
+  else KCP(float);
+#undef KCP
   FM_HIP_CHECK(hipGetLastError());
-  return fm_combine_finalize(stream, moe_out, S);
+  if (tk > 1) return launch_cast_combine(st, moe_out);
+  // k==1: each token has exactly one slot; reduce with K=1 semantics
+  const size_t n = (size_t)g.S * g.H;
+  const int blocks = (int)std::min((size_t)4096, (size_t)DIVUP(n, (size_t)256 * 4));
+#define CC1(T)                                                                \
+  hipLaunchKernelGGL((k_cast_combine<T, 1>), dim3(blocks), dim3(256), 0, st,  \
+                     reinterpret_cast<const T*>(g.cbuf), g.kept,              \
+                     reinterpret_cast<T*>(moe_out), g.S, g.H)
+  if (g.cfg.dtype == 3) CC1(fp16);
+  else if (g.esz == 2) CC1(bf16);
+  else CC1(float);
+#undef CC1
+  FM_HIP_CHECK(hipGetLastError());
+  return FM_OK;
 }
 
-/* Padded-EP grouped FFN: rows = [n_segs, EC, H] (fixed EC rows per
- * segment), seg_expert_dev = device int32[n_segs] mapping segment ->
- * local expert. TWO kernel launches total (grid.z = segments via the
- * GemmArgs segExpert indirection). Rows past each segment's real count
- * are garbage-in/garbage-out by design (dropped at the source). */
 int fm_expert_ffn_segments(void* stream, const void* rows,
                            const void* seg_expert_dev, int32_t n_segs,
                            const void* expert_w, void* out_rows) {
