@@ -878,23 +878,51 @@ __global__ __launch_bounds__(256) void k_group_gemm_f32(GemmArgs a) {
 }
 
 // sum the k non-atomic combine slots (kept-masked) into the output
-// (k>1 path; replaces the fp32-atomic accumulator + cast)
+// (k>1 path; replaces the fp32-atomic accumulator + cast). One block
+// per token chunk, 16-B vector loads per slot (a scalar grid-stride
+// version with a per-element div ran at 1.35 TB/s on the cfg5 shape).
 template <typename T, int K>
 __global__ void k_cast_combine(const T* __restrict__ cbuf,
                                const uint8_t* __restrict__ kept,
                                T* __restrict__ out, int S, int H) {
-  const size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
-  const size_t stride = (size_t)gridDim.x * blockDim.x;
-  const size_t n = (size_t)S * H;
-  for (size_t x = i; x < n; x += stride) {
-    const size_t t = x / H, h = x % H;
-    float acc = 0.0f;
+  constexpr int EPU = 16 / sizeof(T);
+  for (int t = blockIdx.x; t < S; t += gridDim.x) {
+    bool kp[K];
 #pragma unroll
-    for (int j = 0; j < K; ++j) {
-      if (kept[t * K + j])
-        acc += toF(cbuf[((size_t)t * K + j) * H + h]);
+    for (int j = 0; j < K; ++j) kp[j] = kept[(size_t)t * K + j] != 0;
+    for (int h = threadIdx.x * EPU; h < H; h += blockDim.x * EPU) {
+      float acc[EPU];
+#pragma unroll
+      for (int q = 0; q < EPU; ++q) acc[q] = 0.0f;
+#pragma unroll
+      for (int j = 0; j < K; ++j) {
+        if (!kp[j]) continue;
+        const u32x4 v = *reinterpret_cast<const u32x4*>(
+            cbuf + ((size_t)t * K + j) * H + h);
+#pragma unroll
+        for (int w = 0; w < 4; ++w) {
+          const uint32_t vw = v[w];
+          if constexpr (__is_same(T, bf16)) {
+            const float2 f = __bfloat1622float2(
+                *reinterpret_cast<const __hip_bfloat162*>(&vw));
+            acc[2 * w] += f.x;
+            acc[2 * w + 1] += f.y;
+          } else if constexpr (__is_same(T, fp16)) {
+            const float2 f = __half22float2(
+                *reinterpret_cast<const __half2*>(&vw));
+            acc[2 * w] += f.x;
+            acc[2 * w + 1] += f.y;
+          } else {
+            acc[w] += __uint_as_float(vw);
+          }
+        }
+      }
+      T ov[EPU];
+#pragma unroll
+      for (int q = 0; q < EPU; ++q) fromF(acc[q], ov[q]);
+      *reinterpret_cast<u32x4*>(out + (size_t)t * H + h) =
+          *reinterpret_cast<const u32x4*>(ov);
     }
-    fromF(acc, out[x]);
   }
 }
 
@@ -1497,8 +1525,7 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
 }
 
 static int launch_cast_combine(hipStream_t st, void* moe_out) {
-  const size_t n = (size_t)g.S * g.H;
-  const int blocks = (int)std::min((size_t)4096, (size_t)DIVUP(n, (size_t)256 * 4));
+  const int blocks = std::min(g.S, 4096);
 #define CC(T, KK)                                                             \
   hipLaunchKernelGGL((k_cast_combine<T, KK>), dim3(blocks), dim3(256), 0, st, \
                      reinterpret_cast<const T*>(g.cbuf), g.kept,              \
