@@ -1257,7 +1257,7 @@ int launch_gate(hipStream_t st, const void* x, const void* gate_w,
 #define GATE_LOGITS(T)                                                        \
   do {                                                                        \
     if (ldsL > 64 * 1024)                                                     \
-      hipFuncSetAttribute(reinterpret_cast<const void*>(&k_gate_logits<T>),   \
+      (void)hipFuncSetAttribute(reinterpret_cast<const void*>(&k_gate_logits<T>),   \
                           hipFuncAttributeMaxDynamicSharedMemorySize,         \
                           (int)ldsL);                                         \
     hipLaunchKernelGGL((k_gate_logits<T>), dim3(tiles, chunks, eChunks),     \
@@ -1268,7 +1268,7 @@ int launch_gate(hipStream_t st, const void* x, const void* gate_w,
 #define GATE_ROUTE(T, KK)                                                     \
   do {                                                                        \
     if (ldsR > 64 * 1024)                                                     \
-      hipFuncSetAttribute(                                                    \
+      (void)hipFuncSetAttribute(                                                    \
           reinterpret_cast<const void*>(&k_gate_route<T, KK>),                \
           hipFuncAttributeMaxDynamicSharedMemorySize, (int)ldsR);             \
     hipLaunchKernelGGL((k_gate_route<T, KK>), dim3(tiles), dim3(256), ldsR,   \
@@ -1375,14 +1375,15 @@ int fm_initialize(const fm_config* cfg, int rank, int world_size) {
 
 int fm_finalize(void) {
   if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
-  hipFree(g.tokenIds); hipFree(g.eC); hipFree(g.xM); hipFree(g.O32);
-  hipFree(g.cbuf); hipFree(g.kept);
-  hipFree(g.logits32); hipFree(g.gML);
+  (void)hipFree(g.tokenIds); (void)hipFree(g.eC); (void)hipFree(g.xM);
+  (void)hipFree(g.O32);
+  (void)hipFree(g.cbuf); (void)hipFree(g.kept);
+  (void)hipFree(g.logits32); (void)hipFree(g.gML);
   if (g.heap) {
     for (int p = 0; p < g.world; ++p) {
-      if (p != g.rank && g.peerBase[p]) hipIpcCloseMemHandle(g.peerBase[p]);
+      if (p != g.rank && g.peerBase[p]) (void)hipIpcCloseMemHandle(g.peerBase[p]);
     }
-    hipFree(g.heap); hipFree(g.dPeerRecv); hipFree(g.dArrive);
+    (void)hipFree(g.heap); (void)hipFree(g.dPeerRecv); (void)hipFree(g.dArrive);
   }
   g = State{};
   return FM_OK;
@@ -1633,14 +1634,14 @@ int fm_moe_forward_phased(void* stream, const void* x, const void* gate_w,
     FM_HIP_CHECK(hipEventRecord(evs[5], st));
     FM_HIP_CHECK(hipStreamSynchronize(st));
     float pre = 0, gate = 0, upT = 0, dnT = 0, post = 0;
-    hipEventElapsedTime(&pre, evs[0], evs[1]);
-    hipEventElapsedTime(&gate, evs[1], evs[2]);
-    hipEventElapsedTime(&upT, evs[2], evs[3]);
-    hipEventElapsedTime(&dnT, evs[3], evs[4]);
-    hipEventElapsedTime(&post, evs[4], evs[5]);
+    (void)hipEventElapsedTime(&pre, evs[0], evs[1]);
+    (void)hipEventElapsedTime(&gate, evs[1], evs[2]);
+    (void)hipEventElapsedTime(&upT, evs[2], evs[3]);
+    (void)hipEventElapsedTime(&dnT, evs[3], evs[4]);
+    (void)hipEventElapsedTime(&post, evs[4], evs[5]);
     ms[0] = gate; ms[1] = upT; ms[2] = dnT; ms[3] = pre + post;
   }
-  for (auto& e : evs) hipEventDestroy(e);
+  for (auto& e : evs) (void)hipEventDestroy(e);
   return rc;
 }
 
