@@ -514,7 +514,7 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
 // ---------------------------------------------------------------------------
 
 template <typename ET, int PHASE, int ACT, bool HAS_BIAS, int BN, int BM = 256>
-__global__ __launch_bounds__(512) void k_group_gemm_bf16_big(GemmArgs a) {
+__global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
   using vec8 = typename ETr<ET>::vec8;
   constexpr int BK = 64;
   constexpr int NF = BN / 64;           // B fragments per wave (4 or 2)
