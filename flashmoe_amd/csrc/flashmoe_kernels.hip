@@ -93,7 +93,13 @@ template <typename T> constexpr bool cuda_false() { return false; }
 template <typename T>
 __global__ __launch_bounds__(256) void k_gate_logits(
     const T* __restrict__ x, const T* __restrict__ gate_w,
-    float* __restrict__ logits32, int S, int H, int E, int Hc) {
+    float* __restrict__ logits32, uint32_t* __restrict__ eC, int S, int H,
+    int E, int Hc) {
+  // fold the eC zeroing into the first block (route runs after this
+  // kernel; saves a 4-us launch-bound memset per forward)
+  if (blockIdx.x == 0 && blockIdx.y == 0 && blockIdx.z == 0) {
+    for (int i = threadIdx.x; i < E; i += blockDim.x) eC[i] = 0;
+  }
   // blockIdx.z selects a chunk of <=128 experts (E up to 256, config 5)
   const int eBase = blockIdx.z * 128;
   const int Ec = min(E - eBase, 128);
@@ -197,10 +203,14 @@ __global__ __launch_bounds__(256) void k_gate_route(
   float* sMax = sInv + BM;                                         // [BM]
   const int tid = threadIdx.x;
   const int m0 = blockIdx.x * BM;
-  // cooperative coalesced copy of the tile's logits into LDS
+  // cooperative coalesced copy of the tile's logits into LDS, then
+  // re-zero the global tile (keeps logits32 zero between forwards so
+  // the split-K logits GEMM can atomically accumulate without a memset)
   for (int i = tid; i < BM * E; i += 256) {
     const int row = i / E, col = i % E;
-    logits[row * (E + 1) + col] = logits32[(size_t)(m0 + row) * E + col];
+    float* src = const_cast<float*>(&logits32[(size_t)(m0 + row) * E + col]);
+    logits[row * (E + 1) + col] = *src;
+    *src = 0.0f;
   }
   __syncthreads();
 
@@ -346,6 +356,15 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
   const int kStartS = ksplitS * kLenS;
 
   const int we = a.segExpert ? a.segExpert[e] : e;  // weight/bias expert
+  if constexpr (PHASE == 3) {
+    // gate-logits phase: block (0,0,0) zeroes eC for the route kernel
+    // that follows (saves a launch-bound memset per forward); EC field
+    // carries the expert count here
+    if (a.eC && blockIdx.x == 0 && blockIdx.y == 0 && blockIdx.z == 0) {
+      uint32_t* ec = const_cast<uint32_t*>(a.eC);
+      for (int i = tid; i < a.EC; i += 256) ec[i] = 0;
+    }
+  }
   const TPS* tpsE = a.tokenIds ? a.tokenIds + (size_t)e * a.pEC : nullptr;
   if (tid == 0) {
     *sRouted = tpsE ? min(a.eC[e], (uint32_t)a.EC) : (uint32_t)a.nRows;
@@ -1244,8 +1263,6 @@ int launch_gate(hipStream_t st, const void* x, const void* gate_w,
   const int eChunks = DIVUP(g.E, 128);
   const int Ec = g.E < 128 ? g.E : 128;
   const bool mfmaLogits = (g.esz == 2);  // bf16/fp16: MFMA logits GEMM
-  if (!mfmaLogits)
-    FM_HIP_CHECK(hipMemsetAsync(g.logits32, 0, (size_t)g.S * g.E * sizeof(float), st));
   const size_t ldsL = gate_lds_bytes(Ec, g.esz);
   const size_t ldsR = 128 * (g.E + 1) * sizeof(float) +
                       128 * 8 * 2 * sizeof(uint16_t) + g.E * sizeof(uint32_t) +
@@ -1262,7 +1279,7 @@ int launch_gate(hipStream_t st, const void* x, const void* gate_w,
                           (int)ldsL);                                         \
     hipLaunchKernelGGL((k_gate_logits<T>), dim3(tiles, chunks, eChunks),     \
                        dim3(256), ldsL, st, reinterpret_cast<const T*>(x),    \
-                       reinterpret_cast<const T*>(gate_w), g.logits32,        \
+                       reinterpret_cast<const T*>(gate_w), g.logits32, g.eC,  \
                        (int)S, g.H, g.E, Hc);                                 \
   } while (0)
 #define GATE_ROUTE(T, KK)                                                     \
@@ -1290,7 +1307,8 @@ int launch_gate(hipStream_t st, const void* x, const void* gate_w,
     ga.B = gate_w;
     ga.out = g.logits32;
     ga.tokenIds = nullptr;
-    ga.eC = nullptr;
+    ga.eC = g.eC;      // PHASE 3 zeroes it (block 0) for the route kernel
+    ga.EC = g.E;
     ga.K = g.H;
     ga.N = g.E;
     ga.nRows = (int)S;
@@ -1360,6 +1378,7 @@ int fm_initialize(const fm_config* cfg, int rank, int world_size) {
                                  : g.nLx * DIVUP(world_size * g.EC, g.pEC);
   FM_HIP_CHECK(hipMalloc(&g.tokenIds, (size_t)g.E * g.pEC * sizeof(TPS)));
   FM_HIP_CHECK(hipMalloc(&g.logits32, (size_t)g.S * g.E * sizeof(float)));
+  FM_HIP_CHECK(hipMemset(g.logits32, 0, (size_t)g.S * g.E * sizeof(float)));
   FM_HIP_CHECK(hipMalloc(&g.gML, 2 * (size_t)g.E * sizeof(float)));
   g.gMeC = g.gML + g.E;
   FM_HIP_CHECK(hipMalloc(&g.eC, (size_t)g.E * sizeof(uint32_t)));
@@ -1454,9 +1473,8 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
         sk3 *= 2;
       GemmArgs a3 = a;
       a3.splitK = sk3;
-      if (sk3 > 1)
-        FM_HIP_CHECK(hipMemsetAsync(a3.out, 0,
-                                    (size_t)M * a.N * sizeof(float), st));
+      // logits32 is zero on entry (zeroed at initialize and re-zeroed by
+      // k_gate_route after each read)
       dim3 g3(DIVUP(M, 128), DIVUP(N, 128), sk3);
       if (g.cfg.dtype == 3)
         hipLaunchKernelGGL((k_group_gemm_bf16<fp16, 3, 0, false>), g3,
@@ -1551,7 +1569,7 @@ static int moe_forward_impl(hipStream_t st, const void* x, const void* gate_w,
                             const void* expert_w, const void* b_up,
                             const void* b_dn, void* gate_out, void* moe_out,
                             int64_t S, hipEvent_t* evs) {
-  FM_HIP_CHECK(hipMemsetAsync(g.eC, 0, (size_t)g.E * sizeof(uint32_t), st));
+  // eC is zeroed by the gate logits kernel (block 0)
   // O32 is kept zero across calls (zeroed at initialize; k_cast_out
   // re-zeroes after reading)
   // k==1: zero moe_out so dropped tokens keep zeros (clearState,
@@ -1650,8 +1668,7 @@ int fm_gate_forward(void* stream, const void* x, const void* gate_w,
   if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
   if (S != g.S) { setErr("S mismatch vs frozen config"); return FM_ERR_SHAPE; }
   hipStream_t st = reinterpret_cast<hipStream_t>(stream);
-  FM_HIP_CHECK(hipMemsetAsync(g.eC, 0, (size_t)g.E * sizeof(uint32_t), st));
-  return launch_gate(st, x, gate_w, gate_out, S);
+  return launch_gate(st, x, gate_w, gate_out, S);  // gate zeroes eC itself
 }
 
 int fm_read_routing(void* stream, uint32_t* routed_counts, uint32_t* token_idx,
