@@ -1231,6 +1231,11 @@ struct State {
   uint64_t* dPeerRFlag = nullptr;
   uint32_t* dArrive = nullptr;     // [2*E] dispatch/return arrival counters
   unsigned long long seq = 0;
+  // transparent hipGraph cache for repeated-pointer forwards
+  hipGraphExec_t graphExec = nullptr;
+  void* graphKey[8] = {};
+  void* lastKey[8] = {};
+  bool graphValid = false;
   void* xM = nullptr;        // [nLx_alloc, pEC, P] Element
   float* O32 = nullptr;      // [S, H] (staged-API combine accumulator)
   void* cbuf = nullptr;      // [S, k, H] Element: non-atomic combine slots
@@ -1402,6 +1407,7 @@ int fm_initialize(const fm_config* cfg, int rank, int world_size) {
 
 int fm_finalize(void) {
   if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
+  if (g.graphExec) { (void)hipGraphExecDestroy(g.graphExec); g.graphExec = nullptr; }
   (void)hipFree(g.tokenIds); (void)hipFree(g.eC); (void)hipFree(g.xM);
   (void)hipFree(g.O32);
   (void)hipFree(g.cbuf); (void)hipFree(g.kept);
@@ -1639,8 +1645,53 @@ int fm_moe_forward(void* stream, const void* x, const void* gate_w,
     return FM_ERR_STATE;
   }
   if (S != g.S) { setErr("S mismatch vs frozen config"); return FM_ERR_SHAPE; }
-  return moe_forward_impl(reinterpret_cast<hipStream_t>(stream), x, gate_w,
-                          expert_w, b_up, b_dn, gate_out, moe_out, S, nullptr);
+  hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+
+  // hipGraph fast path: torch's caching allocator hands back the same
+  // buffers on steady-state iterations, so after two identical calls we
+  // capture the 6-launch sequence once and replay it (saves the
+  // per-launch host+boundary overhead; falls back to eager whenever any
+  // pointer changes or capture is unavailable).
+  void* key[8] = {const_cast<void*>(x), const_cast<void*>(gate_w),
+                  const_cast<void*>(expert_w), const_cast<void*>(b_up),
+                  const_cast<void*>(b_dn), gate_out, moe_out,
+                  reinterpret_cast<void*>(static_cast<uintptr_t>(S))};
+  if (g.graphValid && memcmp(key, g.graphKey, sizeof(key)) == 0) {
+    if (hipGraphLaunch(g.graphExec, st) == hipSuccess) return FM_OK;
+    g.graphValid = false;  // replay failed: rebuild next time
+  }
+  if (st != nullptr && memcmp(key, g.lastKey, sizeof(key)) == 0) {
+    // second consecutive identical call: try to capture
+    if (g.graphExec) { (void)hipGraphExecDestroy(g.graphExec); g.graphExec = nullptr; }
+    if (hipStreamBeginCapture(st, hipStreamCaptureModeThreadLocal) ==
+        hipSuccess) {
+      int rc = moe_forward_impl(st, x, gate_w, expert_w, b_up, b_dn, gate_out,
+                                moe_out, S, nullptr);
+      hipGraph_t graph = nullptr;
+      hipError_t ec = hipStreamEndCapture(st, &graph);
+      if (rc == FM_OK && ec == hipSuccess && graph) {
+        if (hipGraphInstantiate(&g.graphExec, graph, nullptr, nullptr, 0) ==
+            hipSuccess) {
+          memcpy(g.graphKey, key, sizeof(key));
+          g.graphValid = true;
+          (void)hipGraphDestroy(graph);
+          // the capture did not execute; launch the graph now
+          if (hipGraphLaunch(g.graphExec, st) == hipSuccess) return FM_OK;
+          g.graphValid = false;
+        } else {
+          (void)hipGraphDestroy(graph);
+        }
+      } else if (graph) {
+        (void)hipGraphDestroy(graph);
+      }
+      if (rc != FM_OK) return rc;
+      // capture succeeded as a recording but instantiate/launch failed:
+      // nothing ran yet - fall through to an eager execution
+    }
+  }
+  memcpy(g.lastKey, key, sizeof(key));
+  return moe_forward_impl(st, x, gate_w, expert_w, b_up, b_dn, gate_out,
+                          moe_out, S, nullptr);
 }
 
 int fm_moe_forward_phased(void* stream, const void* x, const void* gate_w,
