@@ -1,0 +1,83 @@
+"""Randomized parity sweep: many random (shape, E, k, act, dtype) configs
+vs the CPU oracle on one GPU. Exact-output comparisons use single-tile
+or overflow-free capacities (routing set determinism); run via
+  python tools/parity_sweep.py [n_cases] [seed]
+Exits nonzero on the first failure.
+"""
+import json
+import os
+import random
+import sys
+import tempfile
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import numpy as np
+import torch
+
+from oracle.moe_oracle import OracleConfig, moe_forward as oracle_forward
+from flashmoe_amd import moe
+from flashmoe_amd.config import torch_dtype_of
+
+n_cases = int(sys.argv[1]) if len(sys.argv) > 1 else 20
+rng = random.Random(int(sys.argv[2]) if len(sys.argv) > 2 else 1234)
+
+fails = 0
+for case in range(n_cases):
+    E = rng.choice([1, 2, 4, 8, 16, 32, 64, 96, 128, 192, 256])
+    k = rng.choice([kk for kk in (1, 2, 4, 8) if kk <= E])
+    H = rng.choice([128, 256, 512, 1024])
+    P = rng.choice([128, 256, 512, 1024, 2048])
+    act = rng.choice([0, 1])
+    dtype = rng.choice([0, 2, 2, 3])  # fp32, bf16 (x2 weight), fp16
+    if dtype != 0 and (H % 128 or P % 128):
+        H = max(128, H // 128 * 128)
+        P = max(128, P // 128 * 128)
+    single_tile = rng.random() < 0.5
+    if single_tile:
+        S, cf, drop = 128, rng.choice([1, 2]), 1
+    else:
+        S = rng.choice([256, 512, 1024])
+        cf, drop = rng.choice([(4, 1), (1, 0)])  # overflow-free
+    cfg = {"capacity_factor": cf, "drop_tokens": drop, "expert_top_k": k,
+           "global_batch": 256, "is_training": rng.choice([0, 1]),
+           "hidden_act": act, "hidden_size": H, "intermediate_size": P,
+           "mini_batch": 1, "moe_frequency": 1, "num_experts": E,
+           "num_layers": 1, "sequence_len": S, "torch_dtype": dtype,
+           "vocab_size": 32000}
+    f = tempfile.NamedTemporaryFile("w", suffix=".json", delete=False)
+    json.dump(cfg, f)
+    f.close()
+    desc = f"S={S} H={H} P={P} E={E} k={k} act={act} dt={dtype} cf={cf} drop={drop}"
+    try:
+        moe.initialize(f.name, rank=0, world_size=1)
+        dt = torch_dtype_of(dtype)
+        gten = torch.Generator().manual_seed(1000 + case)
+        x = torch.randn(1, S, H, generator=gten).to(dt).cuda()
+        gw = torch.randn(H, E, generator=gten).to(dt).cuda()
+        ew = torch.randn(E, 2, P, H, generator=gten).to(dt).cuda()
+        out = moe.moe_forward(x, gw, ew)
+        torch.cuda.synchronize()
+        element = {0: "fp32", 2: "bf16", 3: "fp16"}[dtype]
+        ocfg = OracleConfig(num_experts=E, expert_top_k=k, capacity_factor=cf,
+                            drop_tokens=drop, hidden_act=act, element=element)
+        ref = oracle_forward(x.view(S, H).float().cpu().numpy(),
+                             gw.float().cpu().numpy().reshape(-1),
+                             ew.float().cpu().numpy(), ocfg)
+        got = out.view(S, H).float().cpu().numpy()
+        want = ref["moe_out"]
+        scale = max(1.0, float(np.abs(want).max()))
+        tol = (1e-5, 1e-5 * scale) if element == "fp32" else (2e-2, 2e-3 * scale)
+        ok = np.allclose(got, want, rtol=tol[0], atol=tol[1])
+        err = float(np.abs(got - want).max())
+        print(f"case {case:3d} [{desc}] -> {'OK' if ok else 'FAIL'} (err {err:.4f})",
+              flush=True)
+        fails += (not ok)
+    finally:
+        try:
+            moe.finalize()
+        except Exception:
+            pass
+if fails:
+    print(f"{fails} FAILURES")
+    sys.exit(1)
+print("sweep clean")
