@@ -19,15 +19,28 @@ REQUIRED_KEYS = [
 ]
 
 # torch_dtype enum (schema): 0 float, 1 tf32 (== fp32 on CDNA4: no xf32),
-# 2 bf16, 3 fp16
+# 2 bf16, 3 fp16, 4 fp8e4m3 expert weights with bf16 activations (the
+# config-5 "fp8 weights / bf16 accumulate" regime; extension code - the
+# reference schema stops at 3). torch_dtype_of() is the ACTIVATION dtype;
+# weight_dtype_of() the expert-weight storage dtype.
 def torch_dtype_of(code: int):
     import torch
 
-    return {0: torch.float32, 1: torch.float32, 2: torch.bfloat16, 3: torch.float16}[code]
+    return {0: torch.float32, 1: torch.float32, 2: torch.bfloat16,
+            3: torch.float16, 4: torch.bfloat16}[code]
+
+
+def weight_dtype_of(code: int):
+    import torch
+
+    if code == 4:
+        return torch.float8_e4m3fn
+    return torch_dtype_of(code)
 
 
 def element_size_of(code: int) -> int:
-    return {0: 4, 1: 4, 2: 2, 3: 2}[code]
+    # activation element size (workspace sizing); dtype-4 weights are 1B
+    return {0: 4, 1: 4, 2: 2, 3: 2, 4: 2}[code]
 
 
 def load_config(path: str | None = None) -> dict:

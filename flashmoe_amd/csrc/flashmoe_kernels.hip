@@ -44,6 +44,24 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 typedef __attribute__((ext_vector_type(4))) uint32_t u32x4;
 
 using fp16 = __half;
+struct fp8e4m3 { uint8_t v; };  // storage-only; dequant at fragment read
+
+// convert 8 packed fp8-e4m3 (as 2 dwords) to 8 bf16 lanes
+typedef float f32x2n __attribute__((ext_vector_type(2)));
+__device__ __forceinline__ bf16x8 dequant_fp8x8_bf16(uint32_t lo, uint32_t hi) {
+  bf16x8 r;
+#pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    const uint32_t w = p ? hi : lo;
+    const f32x2n f0 = __builtin_amdgcn_cvt_pk_f32_fp8(w, false);  // bytes 0,1
+    const f32x2n f1 = __builtin_amdgcn_cvt_pk_f32_fp8(w, true);   // bytes 2,3
+    r[p * 4 + 0] = (__bf16)__float2bfloat16(f0.x);
+    r[p * 4 + 1] = (__bf16)__float2bfloat16(f0.y);
+    r[p * 4 + 2] = (__bf16)__float2bfloat16(f1.x);
+    r[p * 4 + 3] = (__bf16)__float2bfloat16(f1.y);
+  }
+  return r;
+}
 
 __device__ __forceinline__ float toF(float v) { return v; }
 __device__ __forceinline__ float toF(bf16 v) { return __bfloat162float(v); }
@@ -336,19 +354,29 @@ struct GemmArgs {
 typedef __attribute__((address_space(1))) const uint32_t gas_u32;
 typedef __attribute__((address_space(3))) uint32_t las_u32;
 
-template <typename ET, int PHASE, int ACT, bool HAS_BIAS>
+template <typename ET, int PHASE, int ACT, bool HAS_BIAS, typename WET = ET>
 __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
   using vec8 = typename ETr<ET>::vec8;
   constexpr int BM = 128, BN = 128, BK = 64;
+  // WET is the WEIGHT (B operand) storage element: == ET normally, or
+  // fp8e4m3 for the config-5 fp8-weight path (W8A16: B staged through
+  // glds as raw fp8 bytes, dequantized to bf16 at fragment-read time so
+  // the MFMA stream and the all-glds staging pipeline are unchanged and
+  // B's HBM/LDS traffic halves).
+  constexpr int BEZ = (int)sizeof(WET);
+  constexpr int BCH = BK * BEZ / 16;  // 16B chunks per B row (8 or 4)
   // ONE shared arena (a second __shared__ object would force a vmcnt(0)
   // drain before every ds_read beside glds - guide par.5 trap 4a).
   // A/B tiles are LINEAR [128][64] bf16 (glds writes lane-linearly); the
   // bank swizzle lives on the SOURCE chunk index and the fragment-read
-  // address (rule 21): chunk' = chunk ^ (row & 7).
-  __shared__ __attribute__((aligned(16))) char smem[2 * BM * BK * 2 + BM * 8 + 16];
+  // address (rule 21): chunk' = chunk ^ (row & 7) (bf16 B); for fp8 B
+  // (64B rows) it is chunk16' = chunk16 ^ ((row >> 2) & 3), which makes
+  // the b64 fragment reads bank-conflict-free.
+  __shared__ __attribute__((aligned(16))) char smem[
+      BM * BK * 2 + BN * BK * BEZ + BM * 8 + 16];
   ET* Alds = reinterpret_cast<ET*>(smem);
-  ET* Blds = Alds + BM * BK;
-  TPS* sTps = reinterpret_cast<TPS*>(Blds + BN * BK);
+  WET* Blds = reinterpret_cast<WET*>(smem + BM * BK * 2);
+  TPS* sTps = reinterpret_cast<TPS*>(smem + BM * BK * 2 + BN * BK * BEZ);
   uint32_t* sRouted = reinterpret_cast<uint32_t*>(sTps + BM);
 
   const int skS = a.splitK > 0 ? a.splitK : 1;
@@ -391,8 +419,8 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
   __syncthreads();
 
   const ET* __restrict__ Ag = reinterpret_cast<const ET*>(a.A);
-  const ET* __restrict__ Bg =
-      reinterpret_cast<const ET*>(a.B) + (size_t)we * a.strideBExpert;
+  const WET* __restrict__ Bg =
+      reinterpret_cast<const WET*>(a.B) + (size_t)we * a.strideBExpert;
 
   // accumulators: wave (wr,wc) owns the 64x64 subtile at (wr*64, wc*64)
   const int wr = wave >> 1, wc = wave & 1;
@@ -407,6 +435,11 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
   // Source chunk is XOR-swizzled: sc = (l%8) ^ (l/8) (row&7 == l/8).
   const int grow8 = lane >> 3;           // row within the 8-row group
   const int schunk = (lane & 7) ^ grow8;  // swizzled 16B chunk index
+  // B staging geometry in WET units: 1KiB glds group covers 1024/(BK*BEZ)
+  // rows; lane -> (row-in-group, chunk) with the WET-specific swizzle.
+  constexpr int BGW = BN * BK * BEZ / 1024 / 4;  // B groups per wave (4 or 2)
+  const int bgrow = lane / BCH;
+  const int bsc = (lane % BCH) ^ (BEZ == 1 ? ((bgrow >> 2) & 3) : bgrow);
   const int aRowStride = (PHASE == 0) ? a.H : K;
   const size_t aBase = (size_t)e * a.strideAExpert;  // 0 for the x-gather up phase
   for (int kt = kStartS; kt < kStartS + kLenS; kt += BK) {
@@ -419,10 +452,15 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
       const ET* asrc = Ag + aBase + arow * aRowStride + kt + schunk * 8;
       __builtin_amdgcn_global_load_lds(
           (gas_u32*)asrc, (las_u32*)(Alds + grp * 512), 16, 0, 0);
+    }
+#pragma unroll
+    for (int i = 0; i < BGW; ++i) {
+      const int grp = wave * BGW + i;
+      const int row = grp * (1024 / (BK * BEZ)) + bgrow;
       const int brow = min(n0 + row, N - 1);
-      const ET* bsrc = Bg + (size_t)brow * K + kt + schunk * 8;
+      const WET* bsrc = Bg + (size_t)brow * K + kt + bsc * (16 / BEZ);
       __builtin_amdgcn_global_load_lds(
-          (gas_u32*)bsrc, (las_u32*)(Blds + grp * 512), 16, 0, 0);
+          (gas_u32*)bsrc, (las_u32*)((char*)Blds + grp * 1024), 16, 0, 0);
     }
     __syncthreads();  // carries the vmcnt(0) glds drain (guide par.5)
 #pragma unroll
@@ -439,8 +477,18 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
 #pragma unroll
       for (int ni = 0; ni < 4; ++ni) {
         const int R = wc * 64 + ni * 16 + rl;
-        bf[ni] = *reinterpret_cast<const vec8*>(
-            &Blds[R * BK + ((cbase ^ (R & 7)) * 8)]);
+        if constexpr (BEZ == 1) {
+          // fp8 B: 8-byte read (the (R>>2)&3 chunk16 swizzle makes the 16
+          // lanes of each quarter-wave hit 16 distinct banks), dequant to
+          // bf16 in regs before the MFMA
+          const int c16 = (cbase >> 1) ^ ((R >> 2) & 3);
+          const uint32_t* bp = reinterpret_cast<const uint32_t*>(
+              (const char*)Blds + (size_t)R * BK + c16 * 16 + (cbase & 1) * 8);
+          bf[ni] = dequant_fp8x8_bf16(bp[0], bp[1]);
+        } else {
+          bf[ni] = *reinterpret_cast<const vec8*>(
+              &Blds[R * BK + ((cbase ^ (R & 7)) * 8)]);
+        }
       }
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
@@ -540,21 +588,28 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
 // Same phase semantics/epilogues as k_group_gemm_bf16.
 // ---------------------------------------------------------------------------
 
-template <typename ET, int PHASE, int ACT, bool HAS_BIAS, int BN, int BM = 256>
+template <typename ET, int PHASE, int ACT, bool HAS_BIAS, int BN, int BM = 256,
+          typename WET = ET>
 __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
   using vec8 = typename ETr<ET>::vec8;
   constexpr int BK = 64;
   constexpr int NF = BN / 64;           // B fragments per wave (4 or 2)
   constexpr int MI = BM / 32;           // A fragments per wave (8 or 4)
+  // WET: weight (B) storage element - == ET normally, fp8e4m3 for the
+  // config-5 fp8-weight path (W8A16). fp8 B is staged through glds as
+  // raw bytes (halving B's HBM + LDS traffic) and dequantized to bf16
+  // at fragment-read time; MFMA stream unchanged.
+  constexpr int BEZ = (int)sizeof(WET);
+  constexpr int BCH = BK * BEZ / 16;        // 16B chunks per B row
   constexpr int AGRP = BM * BK * 2 / 1024;  // glds 1KiB groups per A tile
-  constexpr int BGRP = BN * BK * 2 / 1024;
+  constexpr int BGRP = BN * BK * BEZ / 1024;
   constexpr int GPW_A = AGRP / 8, GPW_B = BGRP / 8;  // per wave
   constexpr int GPT = GPW_A + GPW_B;    // glds per wave per K-tile
   __shared__ __attribute__((aligned(16))) char smem[
-      2 * BM * BK * 2 + 2 * BN * BK * 2 + BM * 8 + 16];
+      2 * BM * BK * 2 + 2 * BN * BK * BEZ + BM * 8 + 16];
   ET* Abase = reinterpret_cast<ET*>(smem);          // 2 x [BM][BK]
-  ET* Bbase = Abase + 2 * BM * BK;                    // 2 x [BN][BK]
-  TPS* sTps = reinterpret_cast<TPS*>(Bbase + 2 * BN * BK);
+  WET* Bbase = reinterpret_cast<WET*>(smem + 2 * BM * BK * 2);  // 2 x [BN][BK]
+  TPS* sTps = reinterpret_cast<TPS*>(smem + 2 * BM * BK * 2 + 2 * BN * BK * BEZ);
   uint32_t* sRouted = reinterpret_cast<uint32_t*>(sTps + BM);
 
   // XCD-aware block remap (perf only, placement-independent for
@@ -602,16 +657,20 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
   __syncthreads();
 
   const ET* __restrict__ Ag = reinterpret_cast<const ET*>(a.A);
-  const ET* __restrict__ Bg =
-      reinterpret_cast<const ET*>(a.B) + (size_t)we * a.strideBExpert;
+  const WET* __restrict__ Bg =
+      reinterpret_cast<const WET*>(a.B) + (size_t)we * a.strideBExpert;
 
   // per-lane glds source bases, hoisted out of the K loop
   const int grow8 = lane >> 3;
   const int schunk = (lane & 7) ^ grow8;
+  // B staging geometry in WET units: 1KiB group = 1024/(BK*BEZ) rows;
+  // fp8 swizzle is on the 16B chunk16 index: ^ ((row >> 2) & 3)
+  const int bgrow = lane / BCH;
+  const int bsc = (lane % BCH) ^ (BEZ == 1 ? ((bgrow >> 2) & 3) : bgrow);
   const int aRowStride = (PHASE == 0) ? a.H : K;
   const size_t aBase = (size_t)e * a.strideAExpert;  // 0 for the x-gather up phase
   const ET* aSrc[GPW_A];
-  const ET* bSrc[GPW_B];
+  const WET* bSrc[GPW_B];
 #pragma unroll
   for (int i = 0; i < GPW_A; ++i) {
     const int row = (wave * GPW_A + i) * 8 + grow8;
@@ -621,8 +680,8 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
   }
 #pragma unroll
   for (int i = 0; i < GPW_B; ++i) {
-    const int row = (wave * GPW_B + i) * 8 + grow8;
-    bSrc[i] = Bg + (size_t)min(n0 + row, N - 1) * K + kStart + schunk * 8;
+    const int row = (wave * GPW_B + i) * (1024 / (BK * BEZ)) + bgrow;
+    bSrc[i] = Bg + (size_t)min(n0 + row, N - 1) * K + kStart + bsc * (16 / BEZ);
   }
 
   auto stage = [&](int kt, int buf) {
@@ -635,7 +694,8 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
     for (int i = 0; i < GPW_B; ++i)
       __builtin_amdgcn_global_load_lds(
           (gas_u32*)(bSrc[i] + kt),
-          (las_u32*)(Bbase + buf * BN * BK + (wave * GPW_B + i) * 512), 16, 0, 0);
+          (las_u32*)((char*)Bbase + buf * BN * BK * BEZ +
+                     (wave * GPW_B + i) * 1024), 16, 0, 0);
   };
 
   const int wr = wave >> 2, wc = wave & 3;  // 2M x 4N wave grid
@@ -664,7 +724,7 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
     const bool late = (NF == 4) && wave >= 4;
     if (stageNow && !late) stage((t + 1) * BK, (t + 1) & 1);
     const ET* Al = Abase + (t & 1) * BM * BK;
-    const ET* Bl = Bbase + (t & 1) * BN * BK;
+    const WET* Bl = Bbase + (t & 1) * BN * BK;
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int s = 0; s < 2; ++s) {
@@ -688,8 +748,15 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
 #pragma unroll
       for (int ni = 0; ni < NF; ++ni) {
         const int R = wc * (BN / 4) + ni * 16 + rl;
-        bfr[ni] = *reinterpret_cast<const vec8*>(
-            &Bl[R * BK + ((cbase ^ (R & 7)) * 8)]);
+        if constexpr (BEZ == 1) {
+          const int c16 = (cbase >> 1) ^ ((R >> 2) & 3);
+          const uint32_t* bp = reinterpret_cast<const uint32_t*>(
+              (const char*)Bl + (size_t)R * BK + c16 * 16 + (cbase & 1) * 8);
+          bfr[ni] = dequant_fp8x8_bf16(bp[0], bp[1]);
+        } else {
+          bfr[ni] = *reinterpret_cast<const vec8*>(
+              &Bl[R * BK + ((cbase ^ (R & 7)) * 8)]);
+        }
       }
       aread(0);
       aread(1);
@@ -1330,7 +1397,7 @@ int launch_gate(hipStream_t st, const void* x, const void* gate_w,
     int rc = launch_group_gemm(st, 3, ga, (int)S, 1);
     if (rc != FM_OK) return rc;
   }
-  if (g.cfg.dtype == 2) {
+  if (g.cfg.dtype == 2 || g.cfg.dtype == 4) {
     if (!mfmaLogits) GATE_LOGITS(bf16);
     GATE_K(bf16)
   } else if (g.cfg.dtype == 3) {
@@ -1376,6 +1443,7 @@ int fm_initialize(const fm_config* cfg, int rank, int world_size) {
   switch (cfg->dtype) {
     case 0: case 1: g.esz = 4; break;
     case 2: case 3: g.esz = 2; break;
+    case 4: g.esz = 2; break;  // fp8e4m3 expert weights, bf16 activations
     default: setErr("unknown dtype");
              return FM_ERR_UNSUPPORTED;
   }
@@ -1500,28 +1568,33 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
       return FM_OK;
     }
     const int sel = phase * 4 + act * 2 + (hasBias ? 1 : 0);
-#define GG_ET(ET, PH, AC, HB)                                                 \
+#define GG_ET(ET, PH, AC, HB, WET)                                            \
     do {                                                                      \
       if (mode == 0)                                                          \
-        hipLaunchKernelGGL((k_group_gemm_bf16_big<ET, PH, AC, HB, 256, 256>), \
-                           grid, block, 0, st, aa);                           \
+        hipLaunchKernelGGL(                                                   \
+            (k_group_gemm_bf16_big<ET, PH, AC, HB, 256, 256, WET>),           \
+            grid, block, 0, st, aa);                                          \
       else if (mode == 1)                                                     \
-        hipLaunchKernelGGL((k_group_gemm_bf16_big<ET, PH, AC, HB, 128, 256>), \
-                           grid, block, 0, st, aa);                           \
+        hipLaunchKernelGGL(                                                   \
+            (k_group_gemm_bf16_big<ET, PH, AC, HB, 128, 256, WET>),           \
+            grid, block, 0, st, aa);                                          \
       else if (mode == 3)                                                     \
-        hipLaunchKernelGGL((k_group_gemm_bf16_big<ET, PH, AC, HB, 256, 128>), \
-                           grid, block, 0, st, aa);                           \
+        hipLaunchKernelGGL(                                                   \
+            (k_group_gemm_bf16_big<ET, PH, AC, HB, 256, 128, WET>),           \
+            grid, block, 0, st, aa);                                          \
       else if (mode == 4)                                                     \
-        hipLaunchKernelGGL((k_group_gemm_bf16_big<ET, PH, AC, HB, 128, 128>), \
-                           grid, block, 0, st, aa);                           \
+        hipLaunchKernelGGL(                                                   \
+            (k_group_gemm_bf16_big<ET, PH, AC, HB, 128, 128, WET>),           \
+            grid, block, 0, st, aa);                                          \
       else                                                                    \
-        hipLaunchKernelGGL((k_group_gemm_bf16<ET, PH, AC, HB>), grid, block,  \
-                           0, st, aa);                                        \
+        hipLaunchKernelGGL((k_group_gemm_bf16<ET, PH, AC, HB, WET>), grid,    \
+                           block, 0, st, aa);                                 \
     } while (0)
 #define GG_CASE(PH, AC, HB)                                                   \
     do {                                                                      \
-      if (g.cfg.dtype == 3) GG_ET(fp16, PH, AC, HB);                          \
-      else GG_ET(bf16, PH, AC, HB);                                           \
+      if (g.cfg.dtype == 3) GG_ET(fp16, PH, AC, HB, fp16);                    \
+      else if (g.cfg.dtype == 4) GG_ET(bf16, PH, AC, HB, fp8e4m3);            \
+      else GG_ET(bf16, PH, AC, HB, bf16);                                     \
     } while (0)
     switch (sel) {
       case 0: GG_CASE(0, 0, false); break;

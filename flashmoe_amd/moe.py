@@ -9,7 +9,8 @@ import ctypes
 import os
 
 from . import _ext
-from .config import DEFAULT_CONFIG_PATH, element_size_of, load_config, torch_dtype_of
+from .config import (DEFAULT_CONFIG_PATH, element_size_of, load_config,
+                     torch_dtype_of, weight_dtype_of)
 
 _state = {
     "initialized": False,
@@ -155,10 +156,12 @@ def _validate_forward_args(input, gate_weights, expert_weights):
             f"[*, 2, {expert_weights.size(2)}, {expert_weights.size(3)}]"
         )
     expect_dtype = torch_dtype_of(_state["cfg"]["torch_dtype"])
-    for name, t in (("Input", input), ("Gate weights", gate_weights),
-                    ("Expert weights", expert_weights)):
-        if t.dtype != expect_dtype:
-            raise ValueError(f"{name} dtype {t.dtype} != compiled {expect_dtype}")
+    expect_wdtype = weight_dtype_of(_state["cfg"]["torch_dtype"])
+    for name, t, want in (("Input", input, expect_dtype),
+                          ("Gate weights", gate_weights, expect_dtype),
+                          ("Expert weights", expert_weights, expect_wdtype)):
+        if t.dtype != want:
+            raise ValueError(f"{name} dtype {t.dtype} != compiled {want}")
     return S, H, E, P, nLx
 
 

@@ -21,7 +21,7 @@ def main():
 
     sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
     from flashmoe_amd import moe
-    from flashmoe_amd.config import torch_dtype_of
+    from flashmoe_amd.config import torch_dtype_of, weight_dtype_of
 
     moe.initialize(config_path, rank=rank, world_size=world_size)
     if world_size > 1:
@@ -39,6 +39,7 @@ def main():
     inter = config["intermediate_size"]
     E = config["num_experts"]
     dtype = torch_dtype_of(config["torch_dtype"])
+    wdtype = weight_dtype_of(config["torch_dtype"])
     nLx = moe.get_num_local_experts()
     print(f"Process {rank}: Creating {nLx} local experts (total {E})", flush=True)
 
@@ -47,7 +48,8 @@ def main():
     torch.manual_seed(47 + rank)
     input_tensor = torch.randn(mini_batch, seq_len, H, dtype=dtype, device="cuda")
     gate_weights = torch.randn(H, E, dtype=dtype, device="cuda")
-    expert_weights = torch.randn(nLx, 2, inter, H, dtype=dtype, device="cuda")
+    expert_weights = torch.randn(
+        nLx, 2, inter, H, dtype=dtype, device="cuda").to(wdtype)
 
     print(f"Process {rank}: Calling moe_forward...", flush=True)
     output = moe.moe_forward(input_tensor, gate_weights, expert_weights)

@@ -53,21 +53,25 @@ def fresh_moe():
 
 def run_pair(moe, cfg, cfg_path, seed=47):
     """Run HIP moe_forward and the oracle on identical inputs."""
-    from flashmoe_amd.config import torch_dtype_of
+    from flashmoe_amd.config import torch_dtype_of, weight_dtype_of
 
     moe.initialize(cfg_path, rank=0, world_size=1)
     S = cfg["sequence_len"] * cfg["mini_batch"]
     H, P, E = cfg["hidden_size"], cfg["intermediate_size"], cfg["num_experts"]
     dt = torch_dtype_of(cfg["torch_dtype"])
+    wdt = weight_dtype_of(cfg["torch_dtype"])
     g = torch.Generator(device="cpu").manual_seed(seed)
     x = torch.randn(cfg["mini_batch"], cfg["sequence_len"], H, generator=g).to(dt).cuda()
     gw = torch.randn(H, E, generator=g).to(dt).cuda()
-    ew = torch.randn(E, 2, P, H, generator=g).to(dt).cuda()
+    # dtype 4: weights quantized to fp8e4m3; .float() below dequantizes
+    # EXACTLY (every e4m3 value is representable in fp32), so the oracle
+    # computes on the same effective weights and the bf16 bar applies
+    ew = torch.randn(E, 2, P, H, generator=g).to(wdt).cuda()
     out = moe.moe_forward(x, gw, ew)
     gate_out = moe.gate_output().clone()
     torch.cuda.synchronize()
 
-    element = {0: "fp32", 1: "fp32", 2: "bf16", 3: "fp16"}[cfg["torch_dtype"]]
+    element = {0: "fp32", 1: "fp32", 2: "bf16", 3: "fp16", 4: "bf16"}[cfg["torch_dtype"]]
     ocfg = OracleConfig(
         num_experts=E, expert_top_k=cfg["expert_top_k"],
         capacity_factor=cfg["capacity_factor"], drop_tokens=cfg["drop_tokens"],
@@ -199,6 +203,25 @@ def test_multi_tile_fp16_big_kernel(fresh_moe):
     got = out.float().cpu().numpy()
     scale = max(1.0, float(np.abs(ref["moe_out"]).max()))
     assert np.allclose(got, ref["moe_out"], rtol=2e-2, atol=2e-3 * scale)
+
+
+def test_single_tile_fp8_weights(fresh_moe):
+    """torch_dtype 4 (extension): fp8e4m3 expert weights, bf16
+    activations/accumulate — the config-5 dtype regime. B is staged as
+    raw fp8 through glds and dequantized at fragment read (W8A16)."""
+    cfg, path = make_cfg(torch_dtype=4)
+    out, gate_out, ref, _ = run_pair(fresh_moe, cfg, path)
+    assert_values(gate_out, ref["gate_out"], "bf16", "gate_out")
+    assert_values(out, ref["moe_out"], "bf16", "moe_out")
+
+
+def test_multi_tile_fp8_big_kernel(fresh_moe):
+    """fp8 weights through the pipelined big kernel (S=2048, CF=2
+    overflow-free -> deterministic routing)."""
+    cfg, path = make_cfg(torch_dtype=4, sequence_len=2048, capacity_factor=2,
+                         hidden_size=256, intermediate_size=512)
+    out, gate_out, ref, _ = run_pair(fresh_moe, cfg, path)
+    assert_values(out, ref["moe_out"], "bf16", "moe_out")
 
 
 def test_single_tile_gelu(fresh_moe):
