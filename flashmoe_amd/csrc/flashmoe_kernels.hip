@@ -1250,6 +1250,13 @@ __global__ void k_export_routing(const TPS* __restrict__ tokenIds,
 
 // MFMA layout probe (test-only): D = A[16x32] x B[32x16] via one
 // mfma_f32_16x16x32_bf16, written with the assumed C/D mapping.
+__global__ void k_fp8cvt_probe(const uint32_t* in, float* out) {
+  const int t = threadIdx.x;  // 32 threads x 8 values
+  const bf16x8 v = dequant_fp8x8_bf16(in[t * 2], in[t * 2 + 1]);
+#pragma unroll
+  for (int i = 0; i < 8; ++i) out[t * 8 + i] = (float)v[i];
+}
+
 __global__ void k_mfma_probe(const bf16* A, const bf16* B, float* D) {
   const int lane = threadIdx.x & 63;
   bf16x8 af, bfr;
@@ -1281,6 +1288,7 @@ struct State {
   int rank = 0, world = 1;
   int S = 0, H = 0, P = 0, E = 0, PX = 0, nLx = 0, EC = 0, pEC = 0;
   size_t esz = 0;
+  size_t wesz = 0;  // expert-weight element size (1 for fp8 weights)
   // workspace
   TPS* tokenIds = nullptr;   // [E, pEC]
   uint32_t* eC = nullptr;    // [E]
@@ -1447,6 +1455,7 @@ int fm_initialize(const fm_config* cfg, int rank, int world_size) {
     default: setErr("unknown dtype");
              return FM_ERR_UNSUPPORTED;
   }
+  g.wesz = (cfg->dtype == 4) ? 1 : g.esz;
   if (g.E > 256) { setErr("E > 256 not supported this round"); return FM_ERR_UNSUPPORTED; }
   if (g.H % 64 || g.P % 64) { setErr("H and P must be multiples of 64"); return FM_ERR_SHAPE; }
   if (g.esz == 2 && (g.H % 128 || g.P % 128)) {
@@ -1691,7 +1700,7 @@ static int moe_forward_impl(hipStream_t st, const void* x, const void* gate_w,
 
   GemmArgs dn = up;
   dn.A = g.xM;
-  dn.B = reinterpret_cast<const char*>(expert_w) + (size_t)g.P * g.H * g.esz;
+  dn.B = reinterpret_cast<const char*>(expert_w) + (size_t)g.P * g.H * g.wesz;
   dn.bias = b_dn;
   dn.out = nullptr;
   dn.O32 = reinterpret_cast<float*>(g.cbuf);  // non-atomic combine slots
@@ -1843,7 +1852,7 @@ int fm_expert_ffn(void* stream, const void* rows, const void* expert_w,
   GemmArgs up{};
   up.A = rows;
   up.B = reinterpret_cast<const char*>(expert_w) +
-         (size_t)local_e * 2 * g.P * g.H * g.esz;
+         (size_t)local_e * 2 * g.P * g.H * g.wesz;
   up.bias = b_up;
   up.out = g.xM;  // scratch [n_rows, P]
   up.tokenIds = nullptr; up.eC = nullptr;
@@ -1857,7 +1866,7 @@ int fm_expert_ffn(void* stream, const void* rows, const void* expert_w,
   GemmArgs dn = up;
   dn.A = g.xM;
   dn.B = reinterpret_cast<const char*>(expert_w) +
-         ((size_t)local_e * 2 + 1) * g.P * g.H * g.esz;
+         ((size_t)local_e * 2 + 1) * g.P * g.H * g.wesz;
   dn.bias = b_dn;
   dn.out = out_rows;
   dn.K = g.P; dn.N = g.H;
@@ -2073,7 +2082,7 @@ int fm_expert_ffn_segments(void* stream, const void* rows,
   if (rc != FM_OK) return rc;
   GemmArgs dn = up;
   dn.A = g.xM;
-  dn.B = reinterpret_cast<const char*>(expert_w) + (size_t)g.P * g.H * g.esz;
+  dn.B = reinterpret_cast<const char*>(expert_w) + (size_t)g.P * g.H * g.wesz;
   dn.out = out_rows;
   dn.strideAExpert = (long long)g.EC * g.P;
   dn.strideOExpert = (long long)g.EC * g.H;
@@ -2175,6 +2184,15 @@ int fm_read_aux_loss(void* stream, float* gML, float* gMeC) {
 }
 
 // test-only probe (see k_mfma_probe)
+int fm_debug_fp8cvt(void* stream, const void* in256, void* out256_f32) {
+  hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+  hipLaunchKernelGGL(k_fp8cvt_probe, dim3(1), dim3(32), 0, st,
+                     reinterpret_cast<const uint32_t*>(in256),
+                     reinterpret_cast<float*>(out256_f32));
+  FM_HIP_CHECK(hipGetLastError());
+  return FM_OK;
+}
+
 int fm_debug_mfma(void* stream, const void* A, const void* B, void* D) {
   hipStream_t st = reinterpret_cast<hipStream_t>(stream);
   hipLaunchKernelGGL(k_mfma_probe, dim3(1), dim3(64), 0, st,
