@@ -21,12 +21,13 @@ sys.path.insert(0, REPO)
 def main():
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
+    dtype_code = int(os.environ.get("FM_TEST_DTYPE", "2"))
     cfg = {
         "capacity_factor": 2, "drop_tokens": 1, "expert_top_k": 2,
         "global_batch": 256, "is_training": 0, "hidden_act": 0,
         "hidden_size": 256, "intermediate_size": 512, "mini_batch": 1,
         "moe_frequency": 1, "num_experts": 8, "num_layers": 1,
-        "sequence_len": 512, "torch_dtype": 2, "vocab_size": 32000,
+        "sequence_len": 512, "torch_dtype": dtype_code, "vocab_size": 32000,
     }
     with tempfile.NamedTemporaryFile("w", suffix=".json", delete=False) as f:
         json.dump(cfg, f)
@@ -40,9 +41,12 @@ def main():
 
     S, H, P, E = 512, 256, 512, 8
     nLx = E // world
+    from flashmoe_amd.config import weight_dtype_of
+
     torch.manual_seed(1234)  # global weights identical on all ranks
     gw = torch.randn(H, E, dtype=torch.bfloat16, device="cuda")
-    ew_full = torch.randn(E, 2, P, H, dtype=torch.bfloat16, device="cuda")
+    ew_full = torch.randn(E, 2, P, H, dtype=torch.bfloat16,
+                          device="cuda").to(weight_dtype_of(dtype_code))
     ew_local = ew_full[rank * nLx:(rank + 1) * nLx].contiguous()
     torch.manual_seed(47 + rank)  # per-rank tokens
     x = torch.randn(1, S, H, dtype=torch.bfloat16, device="cuda")

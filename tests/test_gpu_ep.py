@@ -146,16 +146,18 @@ def test_p2p_transport_world1_matches_single_rank():
         moe.finalize()
 
 
-def test_ep_pipeline_under_torchrun():
-    """Full EP pipeline over torch.distributed (RCCL) at world = #GPUs."""
+@pytest.mark.parametrize("dtype_code", [2, 4])
+def test_ep_pipeline_under_torchrun(dtype_code):
+    """Full EP pipeline over torch.distributed (RCCL) at world = #GPUs.
+    dtype 4 = fp8e4m3 expert weights through fm_expert_ffn_segments."""
     n = torch.cuda.device_count()
     worker = os.path.join(REPO_ROOT, "tests", "ep_gpu_worker.py")
     cmd = [
         sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
         f"--nproc-per-node={n}", "--master-addr", "127.0.0.1",
-        "--master-port", "29519", worker,
+        "--master-port", str(29519 + dtype_code), worker,
     ]
-    env = dict(os.environ, PYTHONPATH=REPO_ROOT)
+    env = dict(os.environ, PYTHONPATH=REPO_ROOT, FM_TEST_DTYPE=str(dtype_code))
     r = subprocess.run(cmd, capture_output=True, text=True, timeout=600, env=env)
     sys.stdout.write(r.stdout[-2000:])
     sys.stderr.write(r.stderr[-2000:])

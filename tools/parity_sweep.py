@@ -16,7 +16,7 @@ import torch
 
 from oracle.moe_oracle import OracleConfig, moe_forward as oracle_forward
 from flashmoe_amd import moe
-from flashmoe_amd.config import torch_dtype_of
+from flashmoe_amd.config import torch_dtype_of, weight_dtype_of
 
 n_cases = int(sys.argv[1]) if len(sys.argv) > 1 else 20
 rng = random.Random(int(sys.argv[2]) if len(sys.argv) > 2 else 1234)
@@ -28,7 +28,7 @@ for case in range(n_cases):
     H = rng.choice([128, 256, 512, 1024])
     P = rng.choice([128, 256, 512, 1024, 2048])
     act = rng.choice([0, 1])
-    dtype = rng.choice([0, 2, 2, 3])  # fp32, bf16 (x2 weight), fp16
+    dtype = rng.choice([0, 2, 2, 3, 4])  # fp32, bf16 (x2), fp16, fp8-weights
     if dtype != 0 and (H % 128 or P % 128):
         H = max(128, H // 128 * 128)
         P = max(128, P // 128 * 128)
@@ -54,10 +54,11 @@ for case in range(n_cases):
         gten = torch.Generator().manual_seed(1000 + case)
         x = torch.randn(1, S, H, generator=gten).to(dt).cuda()
         gw = torch.randn(H, E, generator=gten).to(dt).cuda()
-        ew = torch.randn(E, 2, P, H, generator=gten).to(dt).cuda()
+        ew = torch.randn(E, 2, P, H, generator=gten).to(
+            weight_dtype_of(dtype)).cuda()
         out = moe.moe_forward(x, gw, ew)
         torch.cuda.synchronize()
-        element = {0: "fp32", 2: "bf16", 3: "fp16"}[dtype]
+        element = {0: "fp32", 2: "bf16", 3: "fp16", 4: "bf16"}[dtype]
         ocfg = OracleConfig(num_experts=E, expert_top_k=k, capacity_factor=cf,
                             drop_tokens=drop, hidden_act=act, element=element)
         ref = oracle_forward(x.view(S, H).float().cpu().numpy(),
