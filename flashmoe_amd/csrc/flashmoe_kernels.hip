@@ -47,6 +47,19 @@ using fp16 = __half;
 struct fp8e4m3 { uint8_t v; };  // storage-only; dequant at fragment read
 
 // convert 8 packed fp8-e4m3 (as 2 dwords) to 8 bf16 lanes
+template <int N> __device__ __forceinline__ void wait_vmcnt() {
+  if constexpr (N == 0) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  else if constexpr (N == 1) asm volatile("s_waitcnt vmcnt(1)" ::: "memory");
+  else if constexpr (N == 2) asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+  else if constexpr (N == 3) asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
+  else if constexpr (N == 4) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  else if constexpr (N == 5) asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
+  else if constexpr (N == 6) asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+  else if constexpr (N == 8) asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+  else if constexpr (N == 12) asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
+  else static_assert(N == 0, "unsupported vmcnt literal");
+}
+
 typedef float f32x2n __attribute__((ext_vector_type(2)));
 __device__ __forceinline__ bf16x8 dequant_fp8x8_bf16(uint32_t lo, uint32_t hi) {
   bf16x8 r;
@@ -348,6 +361,7 @@ struct GemmArgs {
                         // atomicAdd combine makes split-K partials free)
   const int32_t* segExpert;  // optional [gridZ] device map: z -> weight
                              // expert (padded-EP segments; null = identity)
+  int noRemap;               // debug: 1 disables the XCD block remap
 };
 
 // address-space helpers for global_load_lds (direct HBM->LDS DMA)
@@ -589,7 +603,7 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
 // ---------------------------------------------------------------------------
 
 template <typename ET, int PHASE, int ACT, bool HAS_BIAS, int BN, int BM = 256,
-          typename WET = ET>
+          typename WET = ET, int STAGES = 2>
 __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
   using vec8 = typename ETr<ET>::vec8;
   constexpr int BK = 64;
@@ -605,11 +619,20 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
   constexpr int BGRP = BN * BK * BEZ / 1024;
   constexpr int GPW_A = AGRP / 8, GPW_B = BGRP / 8;  // per wave
   constexpr int GPT = GPW_A + GPW_B;    // glds per wave per K-tile
+  // STAGES: staging pipeline depth. 2 = classic double buffer (stage t+1
+  // during compute t, then a vmcnt(0) drain per tile). 3 = triple buffer
+  // (stage t+2 during compute t, end-of-tile wait is a COUNTED
+  // vmcnt(GPT) that only requires t+1 landed) - each stage gets TWO
+  // compute phases of latency budget, absorbing L2/HBM staging jitter
+  // that the mfma_probe showed to be the real wall (the LDS+MFMA inner
+  // loop alone reaches 84% of peak; profiles/r01). Fits LDS only for
+  // the BNxBM = 128x256 / 256x128 tiles (144 KB).
   __shared__ __attribute__((aligned(16))) char smem[
-      2 * BM * BK * 2 + 2 * BN * BK * BEZ + BM * 8 + 16];
-  ET* Abase = reinterpret_cast<ET*>(smem);          // 2 x [BM][BK]
-  WET* Bbase = reinterpret_cast<WET*>(smem + 2 * BM * BK * 2);  // 2 x [BN][BK]
-  TPS* sTps = reinterpret_cast<TPS*>(smem + 2 * BM * BK * 2 + 2 * BN * BK * BEZ);
+      STAGES * BM * BK * 2 + STAGES * BN * BK * BEZ + BM * 8 + 16];
+  ET* Abase = reinterpret_cast<ET*>(smem);          // STAGES x [BM][BK]
+  WET* Bbase = reinterpret_cast<WET*>(smem + STAGES * BM * BK * 2);
+  TPS* sTps = reinterpret_cast<TPS*>(
+      smem + STAGES * BM * BK * 2 + STAGES * BN * BK * BEZ);
   uint32_t* sRouted = reinterpret_cast<uint32_t*>(sTps + BM);
 
   // XCD-aware block remap (perf only, placement-independent for
@@ -624,8 +647,9 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
   const int lin = blockIdx.x + mT * (blockIdx.y + nT * blockIdx.z);
   const int qx = nBlocks / 8, rx = nBlocks % 8;
   const int xcd = lin % 8, pos = lin / 8;
-  const int swz =
-      (xcd < rx ? xcd * (qx + 1) : rx * (qx + 1) + (xcd - rx) * qx) + pos;
+  const int swz = a.noRemap
+      ? lin
+      : (xcd < rx ? xcd * (qx + 1) : rx * (qx + 1) + (xcd - rx) * qx) + pos;
   const int sk = a.splitK > 0 ? a.splitK : 1;
   const int eEff = swz / (mT * nT);  // in [0, E*splitK)
   const int e = eEff % (gridDim.z / sk);
@@ -711,26 +735,38 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
   // whole compute phase to land).
   const int nK = kLen / BK;
   stage(0, 0);
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  if constexpr (STAGES == 3) {
+    if (nK > 1) {
+      stage(BK, 1);
+      wait_vmcnt<GPT>();  // tile 0 landed; tile 1 still in flight
+    } else {
+      wait_vmcnt<0>();
+    }
+  } else {
+    wait_vmcnt<0>();
+  }
   __builtin_amdgcn_s_barrier();
   for (int t = 0; t < nK; ++t) {
-    // staggered staging: waves 0-3 stage tile t+1 before their first
-    // MFMA half, waves 4-7 between the halves - so on each SIMD the
-    // paired waves run complementary {stage issue | MFMA} segments
-    // instead of lockstep (microarch guide, two-waves-per-SIMD)
-    const bool stageNow = t + 1 < nK;
+    // STAGES==2: staggered staging - waves 0-3 stage tile t+1 before
+    // their first MFMA half, waves 4-7 between the halves, so on each
+    // SIMD the paired waves run complementary {stage issue | MFMA}
+    // segments instead of lockstep (microarch guide, two-waves-per-SIMD).
+    // STAGES==3: stage t+2 up front; the latency budget is two compute
+    // phases, so no stagger is needed.
+    const int pre = t + STAGES - 1;  // tile staged this iteration
+    const bool stageNow = pre < nK;
     // stagger only pays on the wide (NF==4) stream; the BN=128 kernel's
     // shorter MFMA halves lose more to the mid-stream insertion (measured)
-    const bool late = (NF == 4) && wave >= 4;
-    if (stageNow && !late) stage((t + 1) * BK, (t + 1) & 1);
-    const ET* Al = Abase + (t & 1) * BM * BK;
-    const WET* Bl = Bbase + (t & 1) * BN * BK;
+    const bool late = (STAGES == 2) && (NF == 4) && wave >= 4;
+    if (stageNow && !late) stage(pre * BK, pre % STAGES);
+    const ET* Al = Abase + (t % STAGES) * BM * BK;
+    const WET* Bl = Bbase + (t % STAGES) * BN * BK;
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int s = 0; s < 2; ++s) {
       if (s == 1 && late && stageNow) {
         __builtin_amdgcn_s_setprio(0);
-        stage((t + 1) * BK, (t + 1) & 1);
+        stage(pre * BK, pre % STAGES);
         __builtin_amdgcn_s_setprio(1);
       }
       vec8 af[MI], bfr[NF];
@@ -776,7 +812,13 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
       }
     }
     __builtin_amdgcn_s_setprio(0);
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // t+1 landed
+    if constexpr (STAGES == 3) {
+      // only t+1 must have landed; t+2 (the GPT glds issued this
+      // iteration) may still be in flight
+      if (stageNow) wait_vmcnt<GPT>(); else wait_vmcnt<0>();
+    } else {
+      wait_vmcnt<0>();  // t+1 landed
+    }
     __builtin_amdgcn_s_barrier();  // readers done AND next tile ready
   }
 
@@ -1539,18 +1581,31 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
     // saving bought (profiles/r01). Revisit for huge-K shapes.
     int mode;
     int skf = 1;
+    static const int forceMode = [] {
+      const char* e = getenv("FM_FORCE_MODE");
+      return e ? atoi(e) : -1;
+    }();
+    static const int noRemap = [] {
+      const char* e = getenv("FM_NO_REMAP");
+      return (e && e[0] == '1') ? 1 : 0;
+    }();
     if (M >= 256 && b256 >= 256) mode = 0;
-    else if (M >= 256 && b128n >= 256) mode = 1;
     // 128x128 runs 2 blocks/CU (65 KB LDS, 71 VGPR) so block-level
-    // overlap hides the per-tile sync; prefer it when the grid can
-    // fill 2/CU AND K is not huge (at K=14336 the 1.33x staging of the
-    // smaller tile outweighs the overlap - measured on the cfg4 shape)
+    // overlap hides the per-tile sync; prefer it over the BN=128 /
+    // BM=128 single-resident tiles whenever the grid can fill 2/CU AND
+    // K is not huge (at K=14336 the 1.33x staging of the smaller tile
+    // outweighs the overlap - measured on the cfg4 shape). Checked
+    // BEFORE mode 1: the cfg2 down GEMM (M=N=1024) runs 69 us in mode 4
+    // vs 73 us in mode 1 (measured).
     else if (M >= 128 && h128 >= 512 && a.K <= 8192) mode = 4;
+    else if (M >= 256 && b128n >= 256) mode = 1;
     else if (M >= 128 && h256 >= 256) mode = 3;
     else if (M >= 128 && h128 >= 256) mode = 4;
     else mode = 2;
+    if (forceMode >= 0) mode = forceMode;
     GemmArgs aa = a;
     aa.splitK = skf;
+    aa.noRemap = noRemap;
     const int bmSel = (mode == 0 || mode == 1) ? 256 : 128;
     const int bnSel = (mode == 0 || mode == 3) ? 256 : 128;
     dim3 block(mode == 2 ? 256 : 512);
@@ -1577,15 +1632,31 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
       return FM_OK;
     }
     const int sel = phase * 4 + act * 2 + (hasBias ? 1 : 0);
+    // modes 1/3 (the 128-wide tiles) run the triple-buffered staging
+    // pipeline by default (LDS 144 KB; counted vmcnt waits, two compute
+    // phases of staging latency budget); FM_STAGES=2 reverts for A/B
+    // comparison. Modes 0/4 keep double buffering (LDS / occupancy).
+    static const bool s3 = [] {
+      const char* e = getenv("FM_STAGES");
+      return !(e && e[0] == '2');
+    }();
 #define GG_ET(ET, PH, AC, HB, WET)                                            \
     do {                                                                      \
       if (mode == 0)                                                          \
         hipLaunchKernelGGL(                                                   \
             (k_group_gemm_bf16_big<ET, PH, AC, HB, 256, 256, WET>),           \
             grid, block, 0, st, aa);                                          \
+      else if (mode == 1 && s3)                                               \
+        hipLaunchKernelGGL(                                                   \
+            (k_group_gemm_bf16_big<ET, PH, AC, HB, 128, 256, WET, 3>),        \
+            grid, block, 0, st, aa);                                          \
       else if (mode == 1)                                                     \
         hipLaunchKernelGGL(                                                   \
             (k_group_gemm_bf16_big<ET, PH, AC, HB, 128, 256, WET>),           \
+            grid, block, 0, st, aa);                                          \
+      else if (mode == 3 && s3)                                               \
+        hipLaunchKernelGGL(                                                   \
+            (k_group_gemm_bf16_big<ET, PH, AC, HB, 256, 128, WET, 3>),        \
             grid, block, 0, st, aa);                                          \
       else if (mode == 3)                                                     \
         hipLaunchKernelGGL(                                                   \
