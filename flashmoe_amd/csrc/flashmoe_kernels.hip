@@ -362,6 +362,9 @@ struct GemmArgs {
   const int32_t* segExpert;  // optional [gridZ] device map: z -> weight
                              // expert (padded-EP segments; null = identity)
   int noRemap;               // debug: 1 disables the XCD block remap
+  int totalJobs;             // persistent grid: total (m,n,e) tiles to
+                             // cover (0 = one tile per block, classic)
+  int jobsMT, jobsNT;        // tile grid dims when persistent
 };
 
 // address-space helpers for global_load_lds (direct HBM->LDS DMA)
@@ -642,34 +645,47 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
   // (and, for small E, a whole expert) run on one XCD and hit its L2
   // instead of re-reading HBM/L3 (guide T1; staging-traffic-bound at
   // 6.2 TB/s before this, profiles/r01 pmc2).
-  const int mT = gridDim.x, nT = gridDim.y;
-  const int nBlocks = mT * nT * gridDim.z;
-  const int lin = blockIdx.x + mT * (blockIdx.y + nT * blockIdx.z);
+  const int mT = a.totalJobs > 0 ? a.jobsMT : gridDim.x;
+  const int nT = a.totalJobs > 0 ? a.jobsNT : gridDim.y;
+  const int gstride = gridDim.x * gridDim.y * gridDim.z;
+  const int nBlocks = a.totalJobs > 0 ? a.totalJobs : gstride;
+  const int lin0 =
+      blockIdx.x + gridDim.x * (blockIdx.y + gridDim.y * blockIdx.z);
   const int qx = nBlocks / 8, rx = nBlocks % 8;
-  const int xcd = lin % 8, pos = lin / 8;
-  const int swz = a.noRemap
-      ? lin
-      : (xcd < rx ? xcd * (qx + 1) : rx * (qx + 1) + (xcd - rx) * qx) + pos;
   const int sk = a.splitK > 0 ? a.splitK : 1;
-  const int eEff = swz / (mT * nT);  // in [0, E*splitK)
-  const int e = eEff % (gridDim.z / sk);
-  const int ksplit = eEff / (gridDim.z / sk);
-  const int rem = swz % (mT * nT);
+  const int zE = (a.totalJobs > 0 ? a.totalJobs / (mT * nT)
+                                  : (int)gridDim.z) / sk;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
+  const int K = a.K, N = a.N;
+  // persistent grid (totalJobs > 0): each block walks jobs lin0,
+  // lin0+gstride, ... - the per-launch block setup (sTps load, source
+  // pointer setup, I$ warm, prologue staging bubble) amortizes over
+  // several tiles instead of being paid once per 16-tile block
+  // (mfma_probe: the identical tile loop runs 52.5 us/forward-equiv
+  // persistent vs 71.4 us launched per-tile at the cfg2 up shape)
+  for (int jl = lin0; jl < nBlocks; jl += gstride) {
+  const int xcd = jl % 8, pos = jl / 8;
+  const int swz = a.noRemap
+      ? jl
+      : (xcd < rx ? xcd * (qx + 1) : rx * (qx + 1) + (xcd - rx) * qx) + pos;
+  const int eEff = swz / (mT * nT);  // in [0, E*splitK)
+  const int e = eEff % zE;
+  const int ksplit = eEff / zE;
+  const int rem = swz % (mT * nT);
   const int m0 = (rem % mT) * BM;
   const int n0 = (rem / mT) * BN;
-  const int K = a.K, N = a.N;
   const int kLen = K / sk;            // this split's K range
   const int kStart = ksplit * kLen;
 
   const TPS* tpsE = a.tokenIds ? a.tokenIds + (size_t)e * a.pEC : nullptr;
+  __syncthreads();  // job t-1's epilogue sTps/sScale readers done
   if (tid == 0)
     *sRouted = tpsE ? min(a.eC[e], (uint32_t)a.EC) : (uint32_t)a.nRows;
   __syncthreads();
   const uint32_t routed = *sRouted;
-  if ((uint32_t)m0 >= routed) return;
+  if ((uint32_t)m0 >= routed) continue;
   const int mCap = a.tokenIds ? a.pEC : a.nRows;  // A-row clamp bound
   const int we = a.segExpert ? a.segExpert[e] : e;  // weight/bias expert
   if (tid < BM) {
@@ -894,6 +910,7 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
       }
     }
   }
+  }  // job loop (persistent grid)
 }
 
 // ---------------------------------------------------------------------------
@@ -1606,11 +1623,31 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
     GemmArgs aa = a;
     aa.splitK = skf;
     aa.noRemap = noRemap;
+    aa.totalJobs = 0;
     const int bmSel = (mode == 0 || mode == 1) ? 256 : 128;
     const int bnSel = (mode == 0 || mode == 3) ? 256 : 128;
     dim3 block(mode == 2 ? 256 : 512);
     dim3 grid(DIVUP(M, mode == 2 ? 128 : bmSel),
               DIVUP(N, mode == 2 ? 128 : bnSel), nE * skf);
+    // persistent grid for the big-tile modes: launch exactly the number
+    // of co-resident blocks (1/CU for the 1-block modes, 2/CU for the
+    // 128^2 tile) and let each walk several (m,n,e) tiles - amortizes
+    // the per-block setup + prologue staging bubble (measured ~19 us of
+    // the 71 us cfg2 up kernel). FM_PERSIST=0 reverts to 1 tile/block.
+    static const bool persist = [] {
+      const char* e = getenv("FM_PERSIST");
+      return !(e && e[0] == '0');
+    }();
+    if (persist && mode != 2) {
+      const int jm = grid.x, jn = grid.y, J = jm * jn * nE * skf;
+      const int resident = (mode == 4) ? 512 : 256;
+      if (J > resident) {
+        aa.totalJobs = J;
+        aa.jobsMT = jm;
+        aa.jobsNT = jn;
+        grid = dim3(resident, 1, 1);
+      }
+    }
     if (phase == 3) {  // gate logits: small kernel, no bias/act; K-split
       // to fill the chip (the logits grid is only S/128 x E/128 blocks)
       const int b3 = DIVUP(M, 128) * DIVUP(N, 128);

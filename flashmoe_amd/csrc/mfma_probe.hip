@@ -359,6 +359,161 @@ __global__ __launch_bounds__(512) void k_stage_gemm(const bf16* __restrict__ A,
   if (v == 1234.5678f) sink[tid] = v;
 }
 
+// Variant 5: the FULL real up-kernel structure on synthetic data - 3D
+// grid (mT,nT,E), XCD remap, sTps token gather from LDS, routed guard,
+// per-expert B, strided xM epilogue with activation. iters repeats the
+// K loop + epilogue. If this matches the real kernel's ~950 TF, the gap
+// vs variant 4 (1689 TF) lives in code we can bisect; if it stays fast,
+// the gap is host-side context.
+struct MiniTPS { uint32_t tokenIdx; float probSum; };
+
+__global__ __launch_bounds__(512, 2) void k_stage_real(
+    const bf16* __restrict__ x, const bf16* __restrict__ W,
+    bf16* __restrict__ xM, const MiniTPS* __restrict__ tokenIds,
+    const uint32_t* __restrict__ eC, int K, int N, int pEC, int iters) {
+  constexpr int BM = 256, BN = 256, BK = 64;
+  constexpr int GPW_A = 2, GPW_B = 2;
+  __shared__ __attribute__((aligned(16))) char smem[
+      2 * BM * BK * 2 + 2 * BN * BK * 2 + BM * 8 + 16];
+  bf16* Abase = reinterpret_cast<bf16*>(smem);
+  bf16* Bbase = Abase + 2 * BM * BK;
+  MiniTPS* sTps = reinterpret_cast<MiniTPS*>(Bbase + 2 * BN * BK);
+  uint32_t* sRouted = reinterpret_cast<uint32_t*>(sTps + BM);
+
+  const int mT = gridDim.x, nT = gridDim.y;
+  const int nBlocks = mT * nT * gridDim.z;
+  const int lin = blockIdx.x + mT * (blockIdx.y + nT * blockIdx.z);
+  const int qx = nBlocks / 8, rx = nBlocks % 8;
+  const int xcd = lin % 8, pos = lin / 8;
+  const int swz =
+      (xcd < rx ? xcd * (qx + 1) : rx * (qx + 1) + (xcd - rx) * qx) + pos;
+  const int e = swz / (mT * nT);
+  const int rem = swz % (mT * nT);
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int m0 = (rem % mT) * BM;
+  const int n0 = (rem / mT) * BN;
+
+  const MiniTPS* tpsE = tokenIds + (size_t)e * pEC;
+  if (tid == 0) *sRouted = eC[e];
+  __syncthreads();
+  const uint32_t routed = *sRouted;
+  if ((uint32_t)m0 >= routed) return;
+  if (tid < BM) {
+    MiniTPS t{0u, 1.0f};
+    if ((uint32_t)(m0 + tid) < routed) t = tpsE[m0 + tid];
+    sTps[tid] = t;
+  }
+  __syncthreads();
+
+  const bf16* __restrict__ Bg = W + (size_t)e * 2 * (size_t)N * K;
+  const int grow8 = lane >> 3;
+  const int schunk = (lane & 7) ^ grow8;
+  const bf16* aSrc[GPW_A];
+  const bf16* bSrc[GPW_B];
+#pragma unroll
+  for (int i = 0; i < GPW_A; ++i) {
+    const int row = (wave * GPW_A + i) * 8 + grow8;
+    const size_t arow = (size_t)(sTps[row].tokenIdx & 0x0FFFFFFF);
+    aSrc[i] = x + arow * K + schunk * 8;
+  }
+#pragma unroll
+  for (int i = 0; i < GPW_B; ++i) {
+    const int row = (wave * GPW_B + i) * 8 + grow8;
+    bSrc[i] = Bg + (size_t)min(n0 + row, N - 1) * K + schunk * 8;
+  }
+  auto stage = [&](int kt, int buf) {
+#pragma unroll
+    for (int i = 0; i < GPW_A; ++i)
+      __builtin_amdgcn_global_load_lds(
+          (gas_u32p*)(aSrc[i] + kt),
+          (las_u32p*)(Abase + buf * BM * BK + (wave * GPW_A + i) * 512), 16, 0, 0);
+#pragma unroll
+    for (int i = 0; i < GPW_B; ++i)
+      __builtin_amdgcn_global_load_lds(
+          (gas_u32p*)(bSrc[i] + kt),
+          (las_u32p*)(Bbase + buf * BN * BK + (wave * GPW_B + i) * 512), 16, 0, 0);
+  };
+  const int wr = wave >> 2, wc = wave & 3;
+  f32x4 acc[8][4];
+  const int rl = lane & 15;
+  const int nK = K / BK;
+  for (int it = 0; it < iters; ++it) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+    stage(0, 0);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    for (int t = 0; t < nK; ++t) {
+      const bool stageNow = t + 1 < nK;
+      const bool late = wave >= 4;
+      if (stageNow && !late) stage((t + 1) * BK, (t + 1) & 1);
+      const bf16* Al = Abase + (t & 1) * BM * BK;
+      const bf16* Bl = Bbase + (t & 1) * BN * BK;
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int ss = 0; ss < 2; ++ss) {
+        if (ss == 1 && late && stageNow) {
+          __builtin_amdgcn_s_setprio(0);
+          stage((t + 1) * BK, (t + 1) & 1);
+          __builtin_amdgcn_s_setprio(1);
+        }
+        bf16x8 af[8], bfr[4];
+        const int cbase = 4 * ss + (lane >> 4);
+        auto aread = [&](int mi) {
+          const int R = wr * 128 + mi * 16 + rl;
+          af[mi] = *reinterpret_cast<const bf16x8*>(
+              &Al[R * BK + ((cbase ^ (R & 7)) * 8)]);
+        };
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni) {
+          const int R = wc * 64 + ni * 16 + rl;
+          bfr[ni] = *reinterpret_cast<const bf16x8*>(
+              &Bl[R * BK + ((cbase ^ (R & 7)) * 8)]);
+        }
+        aread(0); aread(1);
+        __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+        for (int g2 = 0; g2 < 4; ++g2) {
+          if (g2 < 3) { aread(2 * g2 + 2); aread(2 * g2 + 3); }
+#pragma unroll
+          for (int mi = 2 * g2; mi < 2 * g2 + 2; ++mi)
+#pragma unroll
+            for (int ni = 0; ni < 4; ++ni)
+              acc[mi][ni] = MFMA(af[mi], bfr[ni], acc[mi][ni]);
+          __builtin_amdgcn_sched_barrier(0);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+    }
+    // real PHASE-0 epilogue: relu + strided xM store
+    const int cl = lane & 15;
+    const int r0 = (lane >> 4) * 4;
+#pragma unroll
+    for (int mi = 0; mi < 8; ++mi) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = wr * 128 + mi * 16 + r0 + r;
+        const int m = m0 + row;
+        if ((uint32_t)m >= routed) continue;
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni) {
+          const int col = n0 + wc * 64 + ni * 16 + cl;
+          if (col >= N) continue;
+          float v = acc[mi][ni][r];
+          v = fmaxf(v, 0.0f);
+          xM[(size_t)e * pEC * N + (size_t)m * N + col] = (bf16)v;
+        }
+      }
+    }
+  }
+}
+
 static double run(void (*kern)(const bf16*, float*, int), const bf16* seed,
                   float* sink, int blocks, int iters) {
   hipEvent_t e0, e1;
@@ -447,6 +602,48 @@ int main() {
                            sink, blocks, nK, sIters);
     printf("blocks=%d nK=%d  COLD-B: staged2-BN256 %.0f TF  "
            "staged3c-BN128 %.0f TF\n", blocks, nK, cb * 2, cb3);
+  }
+  // variant 5: full real structure. cfg2 up shapes: K=1024, N=4096,
+  // pEC=1024, E=8; grid (4,16,8); tokenIds scattered in a 4096-row x.
+  {
+    const int K5 = 1024, N5 = 4096, pEC5 = 1024, E5 = 8;
+    bf16 *x5, *W5, *xM5;
+    MiniTPS* tps5;
+    uint32_t* eC5;
+    (void)hipMalloc(&x5, (size_t)4096 * K5 * 2);
+    (void)hipMalloc(&W5, (size_t)E5 * 2 * N5 * K5 * 2);  // 128 MB
+    (void)hipMalloc(&xM5, (size_t)E5 * pEC5 * N5 * 2);
+    (void)hipMalloc(&tps5, (size_t)E5 * pEC5 * sizeof(MiniTPS));
+    (void)hipMalloc(&eC5, E5 * sizeof(uint32_t));
+    (void)hipMemset(x5, 0x3c, (size_t)4096 * K5 * 2);
+    (void)hipMemset(W5, 0x3c, (size_t)E5 * 2 * N5 * K5 * 2);
+    MiniTPS* htps = new MiniTPS[E5 * pEC5];
+    for (int e = 0; e < E5; ++e)
+      for (int i = 0; i < pEC5; ++i)
+        htps[e * pEC5 + i] = MiniTPS{(uint32_t)((e * pEC5 + i * 997) & 4095), 1.0f};
+    uint32_t hec[E5];
+    for (int e = 0; e < E5; ++e) hec[e] = pEC5;
+    (void)hipMemcpy(tps5, htps, (size_t)E5 * pEC5 * sizeof(MiniTPS),
+                    hipMemcpyHostToDevice);
+    (void)hipMemcpy(eC5, hec, sizeof(hec), hipMemcpyHostToDevice);
+    hipEvent_t e0, e1;
+    (void)hipEventCreate(&e0);
+    (void)hipEventCreate(&e1);
+    const int it5 = 200;
+    dim3 g5(pEC5 / 256, N5 / 256, E5);
+    hipLaunchKernelGGL(k_stage_real, g5, dim3(512), 0, 0, x5, W5, xM5, tps5,
+                       eC5, K5, N5, pEC5, 4);
+    (void)hipDeviceSynchronize();
+    (void)hipEventRecord(e0);
+    hipLaunchKernelGGL(k_stage_real, g5, dim3(512), 0, 0, x5, W5, xM5, tps5,
+                       eC5, K5, N5, pEC5, it5);
+    (void)hipEventRecord(e1);
+    (void)hipEventSynchronize(e1);
+    float ms = 0;
+    (void)hipEventElapsedTime(&ms, e0, e1);
+    const double fl = 2.0 * E5 * pEC5 * (double)N5 * K5 * it5;
+    printf("REAL-STRUCTURE cfg2-up: %.0f TF (%.1f us per forward-equiv)\n",
+           fl / (ms * 1e-3) / 1e12, ms * 1e3 / it5);
   }
   return 0;
 }
