@@ -254,10 +254,8 @@ __global__ __launch_bounds__(256) void k_gate_route(
     float d = 0.0f;
     for (int e = 0; e < E; ++e) d += __expf(lrow[e] - m);
     const float inv_d = 1.0f / d;
-    if (gML) { sInv[tid] = inv_d; sMax[tid] = m; }
-    T* grow = gate_out + (size_t)(m0 + tid) * PX;
-    for (int e = 0; e < E; ++e) fromF(__expf(lrow[e] - m) * inv_d, grow[e]);
-    for (int e = E; e < PX; ++e) fromF(0.0f, grow[e]);
+    sInv[tid] = inv_d;  // always: the coalesced gate_out pass below
+    sMax[tid] = m;      // recomputes probs from these
     // iterative argmax, strict >, first index wins; "taken" by direct
     // comparison with earlier selections (E up to 256)
 #pragma unroll
@@ -275,6 +273,25 @@ __global__ __launch_bounds__(256) void k_gate_route(
     }
 #pragma unroll
     for (int i = 0; i < K; ++i) sel[tid * K + i] = mySel[i];
+  }
+  __syncthreads();
+  // cooperative COALESCED gate_out store: the old per-token loop wrote
+  // each prob as its own 2-byte store at 2*PX-byte thread stride (one
+  // cache line per element - measured the dominant cost of this
+  // kernel); here 256 threads sweep [BM][PX] row-major in 8-element
+  // vectors recomputing exp(l-max)*inv from LDS
+  for (int i = tid * 8; i < BM * PX; i += 256 * 8) {
+    const int row = i / PX, col0 = i % PX;
+    const float mR = sMax[row], ivR = sInv[row];
+    const float* lrow = logits + row * (E + 1);
+    struct __attribute__((aligned(16))) V8 { T x[8]; } v;
+#pragma unroll
+    for (int q = 0; q < 8; ++q) {
+      const int col = col0 + q;
+      const float pv = (col < E) ? __expf(lrow[col] - mR) * ivR : 0.0f;
+      fromF(pv, v.x[q]);
+    }
+    *reinterpret_cast<V8*>(gate_out + (size_t)(m0 + row) * PX + col0) = v;
   }
   __syncthreads();
   if (tid < E) {
@@ -902,6 +919,10 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
                 a.moe_out)[(size_t)tpsTok(tp.tokenIdx) * a.H + col] =
                 ETr<ET>::fromf(v);
           }
+        } else if constexpr (PHASE == 3) {
+          // gate logits, fp32 (the small kernel handles the K-split
+          // variant; the big kernel is only selected with splitK == 1)
+          reinterpret_cast<float*>(a.out)[(size_t)m * N + col] = v;
         } else {
           reinterpret_cast<ET*>(a.out)[(size_t)e * a.strideOExpert +
                                        (size_t)m * N + col] =
@@ -1648,8 +1669,43 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
         grid = dim3(resident, 1, 1);
       }
     }
-    if (phase == 3) {  // gate logits: small kernel, no bias/act; K-split
-      // to fill the chip (the logits grid is only S/128 x E/128 blocks)
+    if (phase == 3) {  // gate logits (fp32 out, B = gate_w)
+      // big-E shapes (e.g. cfg5: S=16384, E=256) fill the chip with the
+      // pipelined 128x128 tile (zeroing of eC still needs the small
+      // kernel's block-0 path, so hand that to a tiny memset instead)
+      if (mode != 2 && N >= 128) {
+        GemmArgs a3 = a;
+        a3.splitK = 1;
+        a3.noRemap = noRemap;
+        a3.totalJobs = 0;
+        if (a3.eC)  // PHASE 3 big kernel does not zero eC; do it here
+          FM_HIP_CHECK(hipMemsetAsync(const_cast<uint32_t*>(a3.eC), 0,
+                                      (size_t)a3.EC * sizeof(uint32_t), st));
+        dim3 g3(DIVUP(M, bmSel), DIVUP(N, bnSel), 1);
+        if (g.cfg.dtype == 3) {
+          if (mode == 0)
+            hipLaunchKernelGGL((k_group_gemm_bf16_big<fp16, 3, 0, false, 256, 256>), g3, dim3(512), 0, st, a3);
+          else if (mode == 1)
+            hipLaunchKernelGGL((k_group_gemm_bf16_big<fp16, 3, 0, false, 128, 256>), g3, dim3(512), 0, st, a3);
+          else if (mode == 3)
+            hipLaunchKernelGGL((k_group_gemm_bf16_big<fp16, 3, 0, false, 256, 128>), g3, dim3(512), 0, st, a3);
+          else
+            hipLaunchKernelGGL((k_group_gemm_bf16_big<fp16, 3, 0, false, 128, 128>), g3, dim3(512), 0, st, a3);
+        } else {
+          if (mode == 0)
+            hipLaunchKernelGGL((k_group_gemm_bf16_big<bf16, 3, 0, false, 256, 256>), g3, dim3(512), 0, st, a3);
+          else if (mode == 1)
+            hipLaunchKernelGGL((k_group_gemm_bf16_big<bf16, 3, 0, false, 128, 256>), g3, dim3(512), 0, st, a3);
+          else if (mode == 3)
+            hipLaunchKernelGGL((k_group_gemm_bf16_big<bf16, 3, 0, false, 256, 128>), g3, dim3(512), 0, st, a3);
+          else
+            hipLaunchKernelGGL((k_group_gemm_bf16_big<bf16, 3, 0, false, 128, 128>), g3, dim3(512), 0, st, a3);
+        }
+        FM_HIP_CHECK(hipGetLastError());
+        return FM_OK;
+      }
+      // small-E: 128^2 kernel with K-split to fill the chip (the logits
+      // grid is only S/128 x E/128 blocks)
       const int b3 = DIVUP(M, 128) * DIVUP(N, 128);
       int sk3 = 1;
       while (sk3 < 8 && b3 * sk3 * 2 <= 512 && (a.K / 64) % (sk3 * 2) == 0)
