@@ -229,7 +229,10 @@ __global__ __launch_bounds__(256) void k_gate_route(
   float* logits = reinterpret_cast<float*>(smem);                  // [BM][E+1]
   uint16_t* sel = reinterpret_cast<uint16_t*>(logits + BM * (E + 1));
   uint16_t* localIdx = sel + BM * K;
-  uint32_t* base = reinterpret_cast<uint32_t*>(localIdx + BM * K);  // [E]
+  // +2: one dump slot so the counting scan's store is UNCONDITIONAL
+  // (an exec-masked conditional store compiled into ~520 serial
+  // s_and_saveexec blocks = ~80% of this kernel's runtime, see ISA)
+  uint32_t* base = reinterpret_cast<uint32_t*>(localIdx + BM * K + 2);  // [E]
   float* sInv = reinterpret_cast<float*>(base + E);                // [BM] 1/d
   float* sMax = sInv + BM;                                         // [BM]
   const int tid = threadIdx.x;
@@ -304,7 +307,10 @@ __global__ __launch_bounds__(256) void k_gate_route(
       for (int q = 0; q < 8; ++q) {
         const uint32_t w = v[q >> 1];
         const uint16_t sv = (q & 1) ? (uint16_t)(w >> 16) : (uint16_t)(w & 0xffff);
-        if (sv == tid) localIdx[mj + q] = (uint16_t)cnt++;
+        // branchless: non-matching entries write the dump slot
+        const bool m = (sv == (uint16_t)tid);
+        localIdx[m ? (mj + q) : BM * K] = (uint16_t)cnt;
+        cnt += m;
       }
     }
     base[tid] = atomicAdd(eC + tid, cnt);
