@@ -642,8 +642,20 @@ int main() {
     float ms = 0;
     (void)hipEventElapsedTime(&ms, e0, e1);
     const double fl = 2.0 * E5 * pEC5 * (double)N5 * K5 * it5;
-    printf("REAL-STRUCTURE cfg2-up: %.0f TF (%.1f us per forward-equiv)\n",
+    printf("REAL-STRUCTURE cfg2-up amortized(it=200): %.0f TF (%.1f us per fwd-equiv)\n",
            fl / (ms * 1e-3) / 1e12, ms * 1e3 / it5);
+    // per-launch variant: iters=1, 50 back-to-back launches - isolates
+    // the per-launch cost the persistent grid was meant to amortize
+    (void)hipEventRecord(e0);
+    for (int r = 0; r < 50; ++r)
+      hipLaunchKernelGGL(k_stage_real, g5, dim3(512), 0, 0, x5, W5, xM5, tps5,
+                         eC5, K5, N5, pEC5, 1);
+    (void)hipEventRecord(e1);
+    (void)hipEventSynchronize(e1);
+    (void)hipEventElapsedTime(&ms, e0, e1);
+    const double fl1 = 2.0 * E5 * pEC5 * (double)N5 * K5 * 50;
+    printf("REAL-STRUCTURE cfg2-up per-launch(it=1 x50): %.0f TF (%.1f us per launch)\n",
+           fl1 / (ms * 1e-3) / 1e12, ms * 1e3 / 50);
   }
   return 0;
 }
