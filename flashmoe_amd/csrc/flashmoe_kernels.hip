@@ -653,12 +653,19 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
   // that the mfma_probe showed to be the real wall (the LDS+MFMA inner
   // loop alone reaches 84% of peak; profiles/r01). Fits LDS only for
   // the BNxBM = 128x256 / 256x128 tiles (144 KB).
+  // effective depth: degrade to double buffering when triple would not
+  // fit the 160 KB LDS (e.g. 256x256 bf16); fp8 B (BEZ==1) halves the B
+  // footprint, so even the 256-wide tiles triple-buffer there
+  constexpr int SE =
+      (STAGES == 3 &&
+       3 * BM * BK * 2 + 3 * BN * BK * BEZ + BM * 8 + 16 > 160 * 1024)
+          ? 2 : STAGES;
   __shared__ __attribute__((aligned(16))) char smem[
-      STAGES * BM * BK * 2 + STAGES * BN * BK * BEZ + BM * 8 + 16];
-  ET* Abase = reinterpret_cast<ET*>(smem);          // STAGES x [BM][BK]
-  WET* Bbase = reinterpret_cast<WET*>(smem + STAGES * BM * BK * 2);
+      SE * BM * BK * 2 + SE * BN * BK * BEZ + BM * 8 + 16];
+  ET* Abase = reinterpret_cast<ET*>(smem);          // SE x [BM][BK]
+  WET* Bbase = reinterpret_cast<WET*>(smem + SE * BM * BK * 2);
   TPS* sTps = reinterpret_cast<TPS*>(
-      smem + STAGES * BM * BK * 2 + STAGES * BN * BK * BEZ);
+      smem + SE * BM * BK * 2 + SE * BN * BK * BEZ);
   uint32_t* sRouted = reinterpret_cast<uint32_t*>(sTps + BM);
 
   // XCD-aware block remap (perf only, placement-independent for
@@ -774,7 +781,7 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
   // whole compute phase to land).
   const int nK = kLen / BK;
   stage(0, 0);
-  if constexpr (STAGES == 3) {
+  if constexpr (SE == 3) {
     if (nK > 1) {
       stage(BK, 1);
       wait_vmcnt<GPT>();  // tile 0 landed; tile 1 still in flight
@@ -792,20 +799,20 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
     // segments instead of lockstep (microarch guide, two-waves-per-SIMD).
     // STAGES==3: stage t+2 up front; the latency budget is two compute
     // phases, so no stagger is needed.
-    const int pre = t + STAGES - 1;  // tile staged this iteration
+    const int pre = t + SE - 1;  // tile staged this iteration
     const bool stageNow = pre < nK;
     // stagger only pays on the wide (NF==4) stream; the BN=128 kernel's
     // shorter MFMA halves lose more to the mid-stream insertion (measured)
-    const bool late = (STAGES == 2) && (NF == 4) && wave >= 4;
-    if (stageNow && !late) stage(pre * BK, pre % STAGES);
-    const ET* Al = Abase + (t % STAGES) * BM * BK;
-    const WET* Bl = Bbase + (t % STAGES) * BN * BK;
+    const bool late = (SE == 2) && (NF == 4) && wave >= 4;
+    if (stageNow && !late) stage(pre * BK, pre % SE);
+    const ET* Al = Abase + (t % SE) * BM * BK;
+    const WET* Bl = Bbase + (t % SE) * BN * BK;
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int s = 0; s < 2; ++s) {
       if (s == 1 && late && stageNow) {
         __builtin_amdgcn_s_setprio(0);
-        stage(pre * BK, pre % STAGES);
+        stage(pre * BK, pre % SE);
         __builtin_amdgcn_s_setprio(1);
       }
       vec8 af[MI], bfr[NF];
@@ -851,7 +858,7 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
       }
     }
     __builtin_amdgcn_s_setprio(0);
-    if constexpr (STAGES == 3) {
+    if constexpr (SE == 3) {
       // only t+1 must have landed; t+2 (the GPT glds issued this
       // iteration) may still be in flight
       if (stageNow) wait_vmcnt<GPT>(); else wait_vmcnt<0>();
@@ -1741,7 +1748,7 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
     }();
 #define GG_ET(ET, PH, AC, HB, WET)                                            \
     do {                                                                      \
-      if (mode == 0)                                                          \
+      if (mode == 0)  /* triple measured -5% here even for fp8 B */          \
         hipLaunchKernelGGL(                                                   \
             (k_group_gemm_bf16_big<ET, PH, AC, HB, 256, 256, WET>),           \
             grid, block, 0, st, aa);                                          \
