@@ -9,8 +9,8 @@ import ctypes
 import os
 
 from . import _ext
-from .config import (DEFAULT_CONFIG_PATH, element_size_of, load_config,
-                     torch_dtype_of, weight_dtype_of)
+from .config import (element_size_of, load_config, torch_dtype_of,
+                     weight_dtype_of)
 
 _state = {
     "initialized": False,

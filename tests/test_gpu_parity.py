@@ -12,7 +12,6 @@ invariants and a capacity-safe configuration.
 """
 import ctypes
 import json
-import os
 import tempfile
 
 import numpy as np
