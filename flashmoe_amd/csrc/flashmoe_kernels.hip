@@ -564,10 +564,15 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
   }
   float bv[4] = {0.f, 0.f, 0.f, 0.f};
   if constexpr (HAS_BIAS) {
+    // multi-expert launches (strideBExpert > 0) carry per-expert bias
+    // slabs [nLx, N]; the packed-rows single-expert API passes the
+    // expert's own slab (stride 0)
+    const ET* bptr = reinterpret_cast<const ET*>(a.bias) +
+                     (a.strideBExpert ? (size_t)we * N : 0);
 #pragma unroll
     for (int ni = 0; ni < 4; ++ni) {
       const int col = n0 + wc * 64 + ni * 16 + cl;
-      if (col < N) bv[ni] = toF(reinterpret_cast<const ET*>(a.bias)[col]);
+      if (col < N) bv[ni] = toF(bptr[col]);
     }
   }
 #pragma unroll
@@ -893,10 +898,13 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
   for (int ni = 0; ni < NF; ++ni) bv[ni] = 0.f;
   if constexpr (HAS_BIAS) {
     if (ksplit == 0) {  // split partials must add the bias exactly once
+      // per-expert bias slabs when multi-expert (see the 128^2 kernel)
+      const ET* bptr = reinterpret_cast<const ET*>(a.bias) +
+                       (a.strideBExpert ? (size_t)we * N : 0);
 #pragma unroll
       for (int ni = 0; ni < NF; ++ni) {
         const int col = n0 + wc * (BN / 4) + ni * 16 + cl;
-        if (col < N) bv[ni] = toF(reinterpret_cast<const ET*>(a.bias)[col]);
+        if (col < N) bv[ni] = toF(bptr[col]);
       }
     }
   }
@@ -1021,10 +1029,13 @@ __global__ __launch_bounds__(256) void k_group_gemm_f32(GemmArgs a) {
   const bool multi = (PHASE == 1) && a.topk > 1;
   float bv[4] = {0.f, 0.f, 0.f, 0.f};
   if constexpr (HAS_BIAS) {
+    // per-expert bias slabs when multi-expert (see the 128^2 kernel)
+    const float* bptr = reinterpret_cast<const float*>(a.bias) +
+                        (a.strideBExpert ? (size_t)we * N : 0);
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
       const int col = n0 + tc + j;
-      if (col < N) bv[j] = reinterpret_cast<const float*>(a.bias)[col];
+      if (col < N) bv[j] = bptr[col];
     }
   }
 #pragma unroll

@@ -223,6 +223,75 @@ def test_multi_tile_fp8_big_kernel(fresh_moe):
     assert_values(out, ref["moe_out"], "bf16", "moe_out")
 
 
+def test_moe_forward_with_biases(fresh_moe):
+    """Per-expert bias slabs b_up [E,P] / b_dn [E,H] through the C-ABI
+    (the reference's Python API zero-fills biases, python_bindings.cu:80-82,
+    but the slabs exist per expert; the C-ABI exposes them)."""
+    import flashmoe_amd._ext as _ext
+    from flashmoe_amd.config import torch_dtype_of
+
+    cfg, path = make_cfg()
+    fresh_moe.initialize(path, rank=0, world_size=1)
+    S, H, P, E = 128, 128, 256, 8
+    dt = torch_dtype_of(2)
+    g = torch.Generator().manual_seed(321)
+    x = torch.randn(S, H, generator=g).to(dt).cuda()
+    gw = torch.randn(H, E, generator=g).to(dt).cuda()
+    ew = torch.randn(E, 2, P, H, generator=g).to(dt).cuda()
+    b_up = torch.randn(E, P, generator=g).to(dt).cuda()
+    b_dn = torch.randn(E, H, generator=g).to(dt).cuda()
+    out = torch.empty(S, H, dtype=dt, device="cuda")
+    gate_out = fresh_moe.gate_output()
+    lib = _ext.load()
+    st = torch.cuda.current_stream().cuda_stream
+    _ext.check(lib.fm_moe_forward(
+        ctypes.c_void_p(st), ctypes.c_void_p(x.data_ptr()),
+        ctypes.c_void_p(gw.data_ptr()), ctypes.c_void_p(ew.data_ptr()),
+        ctypes.c_void_p(b_up.data_ptr()), ctypes.c_void_p(b_dn.data_ptr()),
+        ctypes.c_void_p(gate_out.data_ptr()), ctypes.c_void_p(out.data_ptr()),
+        S), "fm_moe_forward")
+    torch.cuda.synchronize()
+    ocfg = OracleConfig(num_experts=E, expert_top_k=2, element="bf16")
+    ref = oracle_forward(x.float().cpu().numpy(),
+                         gw.float().cpu().numpy().reshape(-1),
+                         ew.float().cpu().numpy(), ocfg,
+                         b_up=b_up.float().cpu().numpy(),
+                         b_dn=b_dn.float().cpu().numpy())
+    assert_values(out, ref["moe_out"], "bf16", "moe_out(bias)")
+
+
+def test_expert_ffn_with_bias(fresh_moe):
+    """Packed-rows FFN (fm_expert_ffn) with the expert's own bias slab
+    (single-expert path: caller passes the slab directly)."""
+    import flashmoe_amd._ext as _ext
+    from oracle.moe_oracle import expert_ffn as oracle_ffn
+
+    cfg, path = make_cfg()
+    fresh_moe.initialize(path, rank=0, world_size=1)
+    H, P = 128, 256
+    g = torch.Generator().manual_seed(99)
+    rows = torch.randn(64, H, generator=g).to(torch.bfloat16).cuda()
+    ew = torch.randn(8, 2, P, H, generator=g).to(torch.bfloat16).cuda()
+    b_up = torch.randn(P, generator=g).to(torch.bfloat16).cuda()
+    b_dn = torch.randn(H, generator=g).to(torch.bfloat16).cuda()
+    out = torch.empty_like(rows)
+    lib = _ext.load()
+    st = torch.cuda.current_stream().cuda_stream
+    _ext.check(lib.fm_expert_ffn(
+        ctypes.c_void_p(st), ctypes.c_void_p(rows.data_ptr()),
+        ctypes.c_void_p(ew.data_ptr()), ctypes.c_void_p(b_up.data_ptr()),
+        ctypes.c_void_p(b_dn.data_ptr()), ctypes.c_void_p(out.data_ptr()),
+        64, 3), "ffn")
+    torch.cuda.synchronize()
+    ocfg = OracleConfig(num_experts=8, expert_top_k=2, element="bf16")
+    want = oracle_ffn(rows.float().cpu().numpy(),
+                      ew[3, 0].float().cpu().numpy(),
+                      ew[3, 1].float().cpu().numpy().reshape(-1),
+                      b_up.float().cpu().numpy(), b_dn.float().cpu().numpy(),
+                      ocfg)
+    assert_values(out, want, "bf16", "expert_ffn(bias)")
+
+
 def test_single_tile_gelu(fresh_moe):
     cfg, path = make_cfg(hidden_act=1)
     out, gate_out, ref, _ = run_pair(fresh_moe, cfg, path)
