@@ -223,6 +223,42 @@ def test_multi_tile_fp8_big_kernel(fresh_moe):
     assert_values(out, ref["moe_out"], "bf16", "moe_out")
 
 
+def test_graph_replay_across_input_changes(fresh_moe):
+    """fm_moe_forward captures/replays a hipGraph keyed on the pointer
+    tuple; successive calls with DIFFERENT tensors (new allocations) and
+    then the ORIGINAL tensors again must each compute on the right data
+    (guards against stale-capture bugs)."""
+    cfg, path = make_cfg()
+    fresh_moe.initialize(path, rank=0, world_size=1)
+    from flashmoe_amd.config import torch_dtype_of
+
+    S, H, P, E = 128, 128, 256, 8
+    dt = torch_dtype_of(2)
+    outs = {}
+    tensors = {}
+    for tag, seed in (("a", 7), ("b", 1234)):
+        g = torch.Generator().manual_seed(seed)
+        x = torch.randn(1, S, H, generator=g).to(dt).cuda()
+        gw = torch.randn(H, E, generator=g).to(dt).cuda()
+        ew = torch.randn(E, 2, P, H, generator=g).to(dt).cuda()
+        tensors[tag] = (x, gw, ew)
+    # a, b, a again (replay of the first graph), repeat b
+    for tag in ("a", "b", "a", "b"):
+        x, gw, ew = tensors[tag]
+        out = fresh_moe.moe_forward(x, gw, ew).clone()
+        torch.cuda.synchronize()
+        if tag in outs:
+            assert torch.equal(out, outs[tag]), f"replay of {tag} diverged"
+        else:
+            outs[tag] = out
+            ocfg = OracleConfig(num_experts=E, expert_top_k=2, element="bf16")
+            ref = oracle_forward(x.view(S, H).float().cpu().numpy(),
+                                 gw.float().cpu().numpy().reshape(-1),
+                                 ew.float().cpu().numpy(), ocfg)
+            assert_values(out.view(S, H), ref["moe_out"], "bf16",
+                          f"moe_out[{tag}]")
+
+
 def test_moe_forward_with_biases(fresh_moe):
     """Per-expert bias slabs b_up [E,P] / b_dn [E,H] through the C-ABI
     (the reference's Python API zero-fills biases, python_bindings.cu:80-82,
