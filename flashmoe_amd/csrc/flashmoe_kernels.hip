@@ -1489,9 +1489,13 @@ int launch_gate(hipStream_t st, const void* x, const void* gate_w,
   switch (g.cfg.expert_top_k) {                                               \
     case 1: GATE_ROUTE(T, 1); break;                                          \
     case 2: GATE_ROUTE(T, 2); break;                                          \
+    case 3: GATE_ROUTE(T, 3); break;                                          \
     case 4: GATE_ROUTE(T, 4); break;                                          \
+    case 5: GATE_ROUTE(T, 5); break;                                          \
+    case 6: GATE_ROUTE(T, 6); break;                                          \
+    case 7: GATE_ROUTE(T, 7); break;                                          \
     case 8: GATE_ROUTE(T, 8); break;                                          \
-    default: setErr("unsupported expert_top_k (1,2,4,8)"); return FM_ERR_UNSUPPORTED; \
+    default: setErr("unsupported expert_top_k (1..8)"); return FM_ERR_UNSUPPORTED; \
   }
   if (mfmaLogits) {
     GemmArgs ga{};
@@ -1561,6 +1565,11 @@ int fm_initialize(const fm_config* cfg, int rank, int world_size) {
   }
   g.wesz = (cfg->dtype == 4) ? 1 : g.esz;
   if (g.E > 256) { setErr("E > 256 not supported this round"); return FM_ERR_UNSUPPORTED; }
+  if (cfg->expert_top_k < 1 || cfg->expert_top_k > 8 ||
+      cfg->expert_top_k > g.E) {
+    setErr("expert_top_k must be in [1, min(E, 8)]");
+    return FM_ERR_UNSUPPORTED;
+  }
   if (g.H % 64 || g.P % 64) { setErr("H and P must be multiples of 64"); return FM_ERR_SHAPE; }
   if (g.esz == 2 && (g.H % 128 || g.P % 128)) {
     setErr("bf16 path requires H, P multiples of 128 this round");
@@ -1836,7 +1845,11 @@ static int launch_cast_combine(hipStream_t st, void* moe_out) {
 #define CC_K(T)                                                               \
   switch (g.cfg.expert_top_k) {                                               \
     case 2: CC(T, 2); break;                                                  \
+    case 3: CC(T, 3); break;                                                  \
     case 4: CC(T, 4); break;                                                  \
+    case 5: CC(T, 5); break;                                                  \
+    case 6: CC(T, 6); break;                                                  \
+    case 7: CC(T, 7); break;                                                  \
     case 8: CC(T, 8); break;                                                  \
     default: setErr("bad topk"); return FM_ERR_UNSUPPORTED;                   \
   }

@@ -340,6 +340,14 @@ def test_single_tile_top4(fresh_moe):
     assert_values(out, ref["moe_out"], "bf16", "moe_out")
 
 
+@pytest.mark.parametrize("k", [3, 5, 7])
+def test_single_tile_odd_topk(fresh_moe, k):
+    """Non-power-of-two top-k (the reference schema allows any k >= 1)."""
+    cfg, path = make_cfg(expert_top_k=k, num_experts=16, capacity_factor=2)
+    out, gate_out, ref, _ = run_pair(fresh_moe, cfg, path)
+    assert_values(out, ref["moe_out"], "bf16", "moe_out")
+
+
 @pytest.mark.parametrize("E,k", [(192, 2), (256, 4)])
 def test_many_experts_gate(fresh_moe, E, k):
     """E > 128 (config-5 shape): the expert-chunked gate logits kernel
