@@ -340,6 +340,16 @@ def test_single_tile_top4(fresh_moe):
     assert_values(out, ref["moe_out"], "bf16", "moe_out")
 
 
+def test_mini_batch_2(fresh_moe):
+    """mini_batch > 1: input [b, s, H] with S = b*s (python_bindings.cu
+    validates batch*seq == compiled S; tokens are row-major across the
+    batch)."""
+    cfg, path = make_cfg(mini_batch=2, sequence_len=128, capacity_factor=2)
+    out, gate_out, ref, _ = run_pair(fresh_moe, cfg, path)
+    assert out.shape[0] == 256
+    assert_values(out, ref["moe_out"], "bf16", "moe_out")
+
+
 @pytest.mark.parametrize("k", [3, 5, 7])
 def test_single_tile_odd_topk(fresh_moe, k):
     """Non-power-of-two top-k (the reference schema allows any k >= 1)."""
