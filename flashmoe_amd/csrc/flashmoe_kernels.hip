@@ -46,7 +46,8 @@ typedef __attribute__((ext_vector_type(4))) uint32_t u32x4;
 using fp16 = __half;
 struct fp8e4m3 { uint8_t v; };  // storage-only; dequant at fragment read
 
-// convert 8 packed fp8-e4m3 (as 2 dwords) to 8 bf16 lanes
+// counted vmcnt wait with a compile-time literal (the triple-buffered
+// staging pipeline waits for tile t+1 while t+2's glds stay in flight)
 template <int N> __device__ __forceinline__ void wait_vmcnt() {
   if constexpr (N == 0) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   else if constexpr (N == 1) asm volatile("s_waitcnt vmcnt(1)" ::: "memory");
@@ -61,6 +62,8 @@ template <int N> __device__ __forceinline__ void wait_vmcnt() {
 }
 
 typedef float f32x2n __attribute__((ext_vector_type(2)));
+// convert 8 packed fp8-e4m3 (as 2 dwords) to 8 bf16 lanes - exact: every
+// e4m3 value is bf16-representable (used at B fragment read, dtype 4)
 __device__ __forceinline__ bf16x8 dequant_fp8x8_bf16(uint32_t lo, uint32_t hi) {
   bf16x8 r;
 #pragma unroll
@@ -101,14 +104,9 @@ template <> struct ETr<fp16> {
   static __device__ __forceinline__ fp16 fromf(float v) { return __float2half(v); }
 };
 
-__device__ __forceinline__ float applyAct(float v, int act) {
-  // hidden_act 0: ReLU, 1: exact-erf GELU (cutlass epilogue::thread::{ReLU,GELU},
-  // schema csrc/flashmoe_config.schema.json:33-37, types.cuh:151-159)
-  if (act == 0) return fmaxf(v, 0.0f);
-  return 0.5f * v * (1.0f + erff(v * 0.70710678118654752f));
-}
-
-template <typename T> constexpr bool cuda_false() { return false; }
+// hidden_act 0: ReLU, 1: exact-erf GELU (cutlass epilogue::thread::{ReLU,
+// GELU}, schema csrc/flashmoe_config.schema.json:33-37, types.cuh:151-159)
+// - applied inline in the GEMM epilogues via the compile-time ACT param.
 
 #define DIVUP(a, b) (((a) + (b) - 1) / (b))
 
