@@ -413,6 +413,20 @@ def test_training_aux_loss(fresh_moe):
     np.testing.assert_allclose(gMeC, want_gMeC, rtol=1e-5, atol=1e-6)
 
 
+def test_full_bench_size_exact_no_overflow(fresh_moe):
+    """Exact value parity at the FULL bench size (S=4096, H=1024, P=4096,
+    E=8, k=2) with capacity_factor 2, where no expert can overflow -> the
+    kept set is deterministic and the whole [4096, 1024] output must meet
+    the tolerance bar vs the oracle (the strongest full-size guarantee the
+    routing nondeterminism permits)."""
+    cfg, path = make_cfg(sequence_len=4096, hidden_size=1024,
+                         intermediate_size=4096, capacity_factor=2)
+    out, gate_out, ref, _ = run_pair(fresh_moe, cfg, path)
+    assert int(ref["eC"].sum()) == 4096 * 2  # nothing dropped
+    assert_values(gate_out, ref["gate_out"], "bf16", "gate_out")
+    assert_values(out, ref["moe_out"], "bf16", "moe_out")
+
+
 def test_bench_config_invariants(fresh_moe):
     """BASELINE config 2 at full size (S=4096, H=1024, P=4096, E=8, k=2,
     CF=1): size-independent properties + routing counts vs oracle
