@@ -2382,6 +2382,77 @@ int fm_read_aux_loss(void* stream, float* gML, float* gMeC) {
   return FM_OK;
 }
 
+// ---------------------------------------------------------------------------
+// Device task-queue primitive (round-2 groundwork; reference semantics:
+// subscriber::start polling + scheduler::start matching,
+// os/subscriber.cuh:333-451 / os/scheduler.cuh:296-441). A seqlock MPMC
+// ring of 128-B task descriptors with generation tags:
+//   producer of slot s (gen g = s/ring):  wait seq[pos]==2g  -> write desc
+//     -> release-store seq[pos]=2g+1
+//   consumer of slot s:                   wait seq[pos]==2g+1 -> read desc
+//     -> release-store seq[pos]=2g+2
+// Device-scope atomics (single GPU); bounded spins so a logic bug aborts
+// instead of hanging the box. The fused persistent kernel
+// will drive expert tiles through this queue; here it is validated by a
+// standalone probe (fm_debug_taskq): P producer blocks enqueue nTasks
+// descriptors, C consumer blocks drain them, each task consumed exactly
+// once (checksum + count verified on the host).
+// ---------------------------------------------------------------------------
+
+struct __align__(128) FmTask {
+  uint32_t kind, payload, slot, gen;
+  uint32_t pad[28];
+};
+
+__global__ void k_taskq_probe(FmTask* ring, uint32_t* seq, uint32_t ringSz,
+                              uint32_t nTasks, uint64_t* tail,
+                              uint64_t* claim, uint32_t* sum,
+                              uint32_t* consumed, uint32_t* errors) {
+  const bool producer = (blockIdx.x & 1) == 0;
+  if (threadIdx.x != 0) return;  // block-level protocol; lane 0 drives
+  if (producer) {
+    while (true) {
+      const uint64_t s = atomicAdd(reinterpret_cast<unsigned long long*>(tail),
+                                   1ull);
+      if (s >= nTasks) break;
+      const uint32_t pos = (uint32_t)(s % ringSz);
+      const uint32_t gen = (uint32_t)(s / ringSz);
+      uint32_t spins = 0;
+      while (__hip_atomic_load(seq + pos, __ATOMIC_ACQUIRE,
+                               __HIP_MEMORY_SCOPE_AGENT) != 2u * gen) {
+        if (++spins > (1u << 26)) { atomicAdd(errors, 1u); return; }
+        __builtin_amdgcn_s_sleep(2);
+      }
+      ring[pos].kind = 1;
+      ring[pos].payload = (uint32_t)(s * 2654435761u);  // checksum source
+      ring[pos].slot = (uint32_t)s;
+      ring[pos].gen = gen;
+      __hip_atomic_store(seq + pos, 2u * gen + 1u, __ATOMIC_RELEASE,
+                         __HIP_MEMORY_SCOPE_AGENT);
+    }
+  } else {
+    while (true) {
+      const uint64_t s = atomicAdd(reinterpret_cast<unsigned long long*>(claim),
+                                   1ull);
+      if (s >= nTasks) break;
+      const uint32_t pos = (uint32_t)(s % ringSz);
+      const uint32_t gen = (uint32_t)(s / ringSz);
+      uint32_t spins = 0;
+      while (__hip_atomic_load(seq + pos, __ATOMIC_ACQUIRE,
+                               __HIP_MEMORY_SCOPE_AGENT) != 2u * gen + 1u) {
+        if (++spins > (1u << 26)) { atomicAdd(errors, 1u); return; }
+        __builtin_amdgcn_s_sleep(2);
+      }
+      const FmTask t = ring[pos];
+      if (t.slot != (uint32_t)s || t.gen != gen) atomicAdd(errors, 1u);
+      atomicAdd(sum, t.payload);
+      atomicAdd(consumed, 1u);
+      __hip_atomic_store(seq + pos, 2u * gen + 2u, __ATOMIC_RELEASE,
+                         __HIP_MEMORY_SCOPE_AGENT);
+    }
+  }
+}
+
 // test-only probe (see k_mfma_probe)
 int fm_debug_fp8cvt(void* stream, const void* in256, void* out256_f32) {
   hipStream_t st = reinterpret_cast<hipStream_t>(stream);
@@ -2389,6 +2460,35 @@ int fm_debug_fp8cvt(void* stream, const void* in256, void* out256_f32) {
                      reinterpret_cast<const uint32_t*>(in256),
                      reinterpret_cast<float*>(out256_f32));
   FM_HIP_CHECK(hipGetLastError());
+  return FM_OK;
+}
+
+int fm_debug_taskq(void* stream, int ring_sz, long long n_tasks,
+                   int blocks, uint32_t* out_sum, uint32_t* out_consumed,
+                   uint32_t* out_errors) {
+  hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+  FmTask* ring = nullptr;
+  uint32_t* seq = nullptr;
+  uint64_t* ctrs = nullptr;  // tail, claim
+  uint32_t* acc = nullptr;   // sum, consumed, errors
+  FM_HIP_CHECK(hipMalloc(&ring, (size_t)ring_sz * sizeof(FmTask)));
+  FM_HIP_CHECK(hipMalloc(&seq, (size_t)ring_sz * sizeof(uint32_t)));
+  FM_HIP_CHECK(hipMalloc(&ctrs, 2 * sizeof(uint64_t)));
+  FM_HIP_CHECK(hipMalloc(&acc, 3 * sizeof(uint32_t)));
+  FM_HIP_CHECK(hipMemsetAsync(seq, 0, (size_t)ring_sz * sizeof(uint32_t), st));
+  FM_HIP_CHECK(hipMemsetAsync(ctrs, 0, 2 * sizeof(uint64_t), st));
+  FM_HIP_CHECK(hipMemsetAsync(acc, 0, 3 * sizeof(uint32_t), st));
+  hipLaunchKernelGGL(k_taskq_probe, dim3(blocks), dim3(64), 0, st, ring, seq,
+                     (uint32_t)ring_sz, (uint32_t)n_tasks, ctrs, ctrs + 1,
+                     acc, acc + 1, acc + 2);
+  FM_HIP_CHECK(hipGetLastError());
+  uint32_t h[3];
+  FM_HIP_CHECK(hipMemcpyAsync(h, acc, sizeof(h), hipMemcpyDeviceToHost, st));
+  FM_HIP_CHECK(hipStreamSynchronize(st));
+  if (out_sum) *out_sum = h[0];
+  if (out_consumed) *out_consumed = h[1];
+  if (out_errors) *out_errors = h[2];
+  (void)hipFree(ring); (void)hipFree(seq); (void)hipFree(ctrs); (void)hipFree(acc);
   return FM_OK;
 }
 

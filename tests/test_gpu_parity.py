@@ -451,6 +451,28 @@ def test_bench_config_invariants(fresh_moe):
         np.testing.assert_allclose(ps[e, sl], mcw[tok[e, sl]], rtol=2e-2)
 
 
+def test_device_task_queue_probe():
+    """Round-2 groundwork: the seqlock MPMC device task ring (subscriber/
+    scheduler semantics, os/subscriber.cuh:333-451 / scheduler.cuh:296-441)
+    delivers every task exactly once across producer/consumer blocks,
+    including ring wrap-around (nTasks >> ring size)."""
+    import __graft_entry__  # noqa: F401
+    import flashmoe_amd._ext as _ext
+
+    lib = _ext.load()
+    n_tasks, ring, blocks = 200_000, 1024, 64
+    out = [ctypes.c_uint32() for _ in range(3)]
+    _ext.check(lib.fm_debug_taskq(
+        None, ring, ctypes.c_longlong(n_tasks), blocks,
+        *[ctypes.byref(o) for o in out]), "fm_debug_taskq")
+    got_sum, consumed, errors = [int(o.value) for o in out]
+    want_sum = sum((s * 2654435761) & 0xFFFFFFFF for s in range(n_tasks)) \
+        & 0xFFFFFFFF
+    assert errors == 0
+    assert consumed == n_tasks
+    assert got_sum == want_sum
+
+
 def test_error_behaviour(fresh_moe):
     """Shape/device validation mirrors the reference's TORCH_CHECKs
     (python_bindings.cu:22-70)."""
