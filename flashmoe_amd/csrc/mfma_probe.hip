@@ -664,6 +664,178 @@ __global__ __launch_bounds__(512, 2) void k_stage_queued(
   }
 }
 
+// Variant 7: dynamic producer->consumer pipeline. 32 producer blocks
+// simulate route tiles (a fixed busy-delay, then emit their share of the
+// 512 tile tasks with release-tagged seq); 480 consumer blocks claim and
+// compute tiles as they appear. Measures whether the producer phase
+// hides behind GEMM fill (the fused kernel's core overlap claim) and
+// whether acquire-spin consumers cost anything once tasks flow.
+__global__ __launch_bounds__(512, 2) void k_stage_pipe(
+    const bf16* __restrict__ x, const bf16* __restrict__ W,
+    bf16* __restrict__ xM, const MiniTPS* __restrict__ tokenIds,
+    const uint32_t* __restrict__ eC, const TileTask* __restrict__ taskSrc,
+    TileTask* ring, uint32_t* seq, unsigned long long* claim,
+    int nProducers, int J, int delayLoops, int K, int N, int pEC) {
+  constexpr int BM = 256, BN = 256, BK = 64;
+  constexpr int GPW_A = 2, GPW_B = 2;
+  __shared__ __attribute__((aligned(16))) char smem[
+      2 * BM * BK * 2 + 2 * BN * BK * 2 + BM * 8 + 64];
+  bf16* Abase = reinterpret_cast<bf16*>(smem);
+  bf16* Bbase = Abase + 2 * BM * BK;
+  MiniTPS* sTps = reinterpret_cast<MiniTPS*>(Bbase + 2 * BN * BK);
+  uint32_t* sRouted = reinterpret_cast<uint32_t*>(sTps + BM);
+  uint32_t* sTask = sRouted + 1;
+  const int tid = threadIdx.x;
+  if ((int)blockIdx.x < nProducers) {
+    // producer: busy-wait (simulated route tile), then emit my tasks
+    if (tid != 0) return;
+    for (volatile int i = 0; i < delayLoops; ++i) { }
+    const int per = J / nProducers;
+    for (int i = 0; i < per; ++i) {
+      const int t = blockIdx.x * per + i;
+      ring[t] = taskSrc[t];
+      __hip_atomic_store(seq + t, 1u, __ATOMIC_RELEASE,
+                         __HIP_MEMORY_SCOPE_AGENT);
+    }
+    return;
+  }
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int rl = lane & 15;
+  const int wr = wave >> 2, wc = wave & 3;
+  const int grow8 = lane >> 3;
+  const int schunk = (lane & 7) ^ grow8;
+  const int nK = K / BK;
+  while (true) {
+    __syncthreads();
+    if (tid == 0) {
+      const uint32_t tix = (uint32_t)atomicAdd(claim, 1ull);
+      if (tix < (uint32_t)J) {
+        uint32_t spins = 0;
+        while (__hip_atomic_load(seq + tix, __ATOMIC_ACQUIRE,
+                                 __HIP_MEMORY_SCOPE_AGENT) == 0u) {
+          if (++spins > (1u << 26)) break;
+          __builtin_amdgcn_s_sleep(1);
+        }
+      }
+      *sTask = tix;
+    }
+    __syncthreads();
+    const uint32_t tix = *sTask;
+    if (tix >= (uint32_t)J) return;
+    const TileTask tk = ring[tix];
+    const int e = tk.e, m0 = tk.m0, n0 = tk.n0;
+    const MiniTPS* tpsE = tokenIds + (size_t)e * pEC;
+    if (tid == 0) *sRouted = eC[e];
+    __syncthreads();
+    const uint32_t routed = *sRouted;
+    if ((uint32_t)m0 >= routed) continue;
+    if (tid < BM) {
+      MiniTPS t{0u, 1.0f};
+      if ((uint32_t)(m0 + tid) < routed) t = tpsE[m0 + tid];
+      sTps[tid] = t;
+    }
+    __syncthreads();
+    const bf16* __restrict__ Bg = W + (size_t)e * 2 * (size_t)N * K;
+    const bf16* aSrc[GPW_A];
+    const bf16* bSrc[GPW_B];
+#pragma unroll
+    for (int i = 0; i < GPW_A; ++i) {
+      const int row = (wave * GPW_A + i) * 8 + grow8;
+      const size_t arow = (size_t)(sTps[row].tokenIdx & 0x0FFFFFFF);
+      aSrc[i] = x + arow * K + schunk * 8;
+    }
+#pragma unroll
+    for (int i = 0; i < GPW_B; ++i) {
+      const int row = (wave * GPW_B + i) * 8 + grow8;
+      bSrc[i] = Bg + (size_t)min(n0 + row, N - 1) * K + schunk * 8;
+    }
+    auto stage = [&](int kt, int buf) {
+#pragma unroll
+      for (int i = 0; i < GPW_A; ++i)
+        __builtin_amdgcn_global_load_lds(
+            (gas_u32p*)(aSrc[i] + kt),
+            (las_u32p*)(Abase + buf * BM * BK + (wave * GPW_A + i) * 512), 16, 0, 0);
+#pragma unroll
+      for (int i = 0; i < GPW_B; ++i)
+        __builtin_amdgcn_global_load_lds(
+            (gas_u32p*)(bSrc[i] + kt),
+            (las_u32p*)(Bbase + buf * BN * BK + (wave * GPW_B + i) * 512), 16, 0, 0);
+    };
+    f32x4 acc[8][4];
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+    stage(0, 0);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    for (int t = 0; t < nK; ++t) {
+      const bool stageNow = t + 1 < nK;
+      const bool late = wave >= 4;
+      if (stageNow && !late) stage((t + 1) * BK, (t + 1) & 1);
+      const bf16* Al = Abase + (t & 1) * BM * BK;
+      const bf16* Bl = Bbase + (t & 1) * BN * BK;
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int ss = 0; ss < 2; ++ss) {
+        if (ss == 1 && late && stageNow) {
+          __builtin_amdgcn_s_setprio(0);
+          stage((t + 1) * BK, (t + 1) & 1);
+          __builtin_amdgcn_s_setprio(1);
+        }
+        bf16x8 af[8], bfr[4];
+        const int cbase = 4 * ss + (lane >> 4);
+        auto aread = [&](int mi) {
+          const int R = wr * 128 + mi * 16 + rl;
+          af[mi] = *reinterpret_cast<const bf16x8*>(
+              &Al[R * BK + ((cbase ^ (R & 7)) * 8)]);
+        };
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni) {
+          const int R = wc * 64 + ni * 16 + rl;
+          bfr[ni] = *reinterpret_cast<const bf16x8*>(
+              &Bl[R * BK + ((cbase ^ (R & 7)) * 8)]);
+        }
+        aread(0); aread(1);
+        __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+        for (int g2 = 0; g2 < 4; ++g2) {
+          if (g2 < 3) { aread(2 * g2 + 2); aread(2 * g2 + 3); }
+#pragma unroll
+          for (int mi = 2 * g2; mi < 2 * g2 + 2; ++mi)
+#pragma unroll
+            for (int ni = 0; ni < 4; ++ni)
+              acc[mi][ni] = MFMA(af[mi], bfr[ni], acc[mi][ni]);
+          __builtin_amdgcn_sched_barrier(0);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+    }
+    const int cl = lane & 15;
+    const int r0 = (lane >> 4) * 4;
+#pragma unroll
+    for (int mi = 0; mi < 8; ++mi) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = wr * 128 + mi * 16 + r0 + r;
+        const int m = m0 + row;
+        if ((uint32_t)m >= routed) continue;
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni) {
+          const int col = n0 + wc * 64 + ni * 16 + cl;
+          if (col >= N) continue;
+          float v = acc[mi][ni][r];
+          v = fmaxf(v, 0.0f);
+          xM[(size_t)e * pEC * N + (size_t)m * N + col] = (bf16)v;
+        }
+      }
+    }
+  }
+}
+
 static double run(void (*kern)(const bf16*, float*, int), const bf16* seed,
                   float* sink, int blocks, int iters) {
   hipEvent_t e0, e1;
@@ -846,6 +1018,33 @@ int main() {
     printf("QUEUE-DRIVEN cfg2-up: %.0f TF (%.1f us per forward-equiv) - "
            "task-dispatch overhead vs REAL-STRUCTURE\n",
            flq / (ms * 1e-3) / 1e12, ms * 1e3 / itq);
+
+    // dynamic pipeline: 32 producers (simulated ~15 us route delay) + 480
+    // consumers, single launch per forward; compare wall vs 15 us + GEMM
+    uint32_t* seqp;
+    TileTask* ringp;
+    (void)hipMalloc(&seqp, (size_t)J * sizeof(uint32_t));
+    (void)hipMalloc(&ringp, (size_t)J * sizeof(TileTask));
+    // delay loop is scratch-bound (~100 cyc/iter): 0 / ~20 us / ~60 us
+    for (int delay : {0, 500, 5000, 15000}) {
+      float tot = 0;
+      const int reps = 30;
+      for (int r = 0; r < reps; ++r) {
+        (void)hipMemset(seqp, 0, (size_t)J * sizeof(uint32_t));
+        (void)hipMemset(claim, 0, sizeof(unsigned long long));
+        (void)hipEventRecord(e0);
+        hipLaunchKernelGGL(k_stage_pipe, dim3(512), dim3(512), 0, 0, x5, W5,
+                           xM5, tps5, eC5, tasks, ringp, seqp, claim, 32, J,
+                           delay, K5, N5, pEC5);
+        (void)hipEventRecord(e1);
+        (void)hipEventSynchronize(e1);
+        (void)hipEventElapsedTime(&ms, e0, e1);
+        tot += ms;
+      }
+      printf("PIPELINE delayLoops=%d: %.1f us per forward "
+             "(480 consumers; producer delay overlapped)\n",
+             delay, tot / reps * 1e3);
+    }
   }
   return 0;
 }
