@@ -66,6 +66,8 @@ def _bind(lib: ctypes.CDLL) -> ctypes.CDLL:
     lib.fm_dispatch_p2p.restype = ctypes.c_int
     lib.fm_return_p2p.argtypes = [p, p]
     lib.fm_return_p2p.restype = ctypes.c_int
+    lib.fm_p2p_error_check.argtypes = [p]
+    lib.fm_p2p_error_check.restype = ctypes.c_int
     lib.fm_pack_dispatch.argtypes = [p, p, p]
     lib.fm_pack_dispatch.restype = ctypes.c_int
     lib.fm_expert_ffn_segments.argtypes = [p, p, p, ctypes.c_int32, p, p]

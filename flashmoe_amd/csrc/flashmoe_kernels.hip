@@ -1273,11 +1273,16 @@ __global__ void k_dispatch_p2p(const T* __restrict__ x,
   }
 }
 
-// wait until every (source, local expert) dispatch flag equals seq
+// wait until every (source, local expert) dispatch flag equals seq.
+// On timeout the device error word (host-mapped, zero-copy) is set so
+// the host can detect the poisoned exchange without a stream sync:
+// fm_dispatch_p2p/fm_return_p2p check it at entry and
+// fm_p2p_error_check() checks it definitively after a sync.
 __global__ void k_await_flags(const unsigned long long* __restrict__ flags,
-                              int n, unsigned long long seq) {
+                              int n, unsigned long long seq,
+                              uint32_t* __restrict__ err, long long maxSpin) {
   bool done = false;
-  for (long long spin = 0; !done && spin < (1ll << 26); ++spin) {
+  for (long long spin = 0; !done && spin < maxSpin; ++spin) {
     done = true;
     for (int i = threadIdx.x; i < n; i += blockDim.x) {
       done &= (__hip_atomic_load(flags + i, __ATOMIC_RELAXED,
@@ -1287,8 +1292,12 @@ __global__ void k_await_flags(const unsigned long long* __restrict__ flags,
     if (!done) __builtin_amdgcn_s_sleep(16);
   }
   if (threadIdx.x == 0 && !done) {
-    // bounded spin gave up: poison is preferable to a hang
+    // bounded spin gave up: poison is preferable to a hang, but the
+    // host MUST be able to see it happened
     printf("flashmoe: k_await_flags timeout (seq %llu)\n", seq);
+    if (err)
+      __hip_atomic_store(err, (uint32_t)seq ? (uint32_t)seq : 1u,
+                         __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM);
   }
   __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
   __syncthreads();
@@ -1407,6 +1416,8 @@ struct State {
   uint64_t* dPeerDFlag = nullptr;
   uint64_t* dPeerRFlag = nullptr;
   uint32_t* dArrive = nullptr;     // [2*E] dispatch/return arrival counters
+  uint32_t* hP2pErr = nullptr;     // host-mapped timeout flag (zero-copy)
+  uint32_t* dP2pErr = nullptr;     // device view of hP2pErr
   unsigned long long seq = 0;
   // transparent hipGraph cache for repeated-pointer forwards
   hipGraphExec_t graphExec = nullptr;
@@ -1563,6 +1574,18 @@ int fm_initialize(const fm_config* cfg, int rank, int world_size) {
   }
   g.wesz = (cfg->dtype == 4) ? 1 : g.esz;
   if (g.E > 256) { setErr("E > 256 not supported this round"); return FM_ERR_UNSUPPORTED; }
+  if (world_size > 64) {
+    // the P2P peer tables (State.peerBase, fm_heap_connect locals) are
+    // sized for 64 ranks - a full 8-node envelope; reject beyond it
+    setErr("world_size > 64 not supported");
+    return FM_ERR_UNSUPPORTED;
+  }
+  if (g.S < 128 || g.S % 128) {
+    // the gate tiles tokens in blocks of 128 (launch_gate: tiles = S/128);
+    // a ragged tail would be silently dropped - reject instead
+    setErr("S = sequence_len * mini_batch must be a positive multiple of 128");
+    return FM_ERR_SHAPE;
+  }
   if (cfg->expert_top_k < 1 || cfg->expert_top_k > 8 ||
       cfg->expert_top_k > g.E) {
     setErr("expert_top_k must be in [1, min(E, 8)]");
@@ -1605,6 +1628,7 @@ int fm_finalize(void) {
       if (p != g.rank && g.peerBase[p]) (void)hipIpcCloseMemHandle(g.peerBase[p]);
     }
     (void)hipFree(g.heap); (void)hipFree(g.dPeerRecv); (void)hipFree(g.dArrive);
+    if (g.hP2pErr) (void)hipHostFree(g.hP2pErr);
   }
   g = State{};
   return FM_OK;
@@ -2090,6 +2114,13 @@ int fm_heap_init(void) {
   g.dPeerDFlag = g.dPeerRet + g.world;
   g.dPeerRFlag = g.dPeerDFlag + g.world;
   FM_HIP_CHECK(hipMalloc(&g.dArrive, 2 * (size_t)g.E * 4));
+  // zero-copy timeout flag: k_await_flags sets it on a bounded-spin
+  // give-up; the host reads it without a stream sync
+  FM_HIP_CHECK(hipHostMalloc(&g.hP2pErr, sizeof(uint32_t),
+                             hipHostMallocMapped));
+  *g.hP2pErr = 0;
+  FM_HIP_CHECK(hipHostGetDevicePointer(
+      reinterpret_cast<void**>(&g.dP2pErr), g.hP2pErr, 0));
   return FM_OK;
 }
 
@@ -2139,8 +2170,41 @@ int fm_heap_ptrs(void** recv, void** ret) {
   return FM_OK;
 }
 
+// bounded-spin budget for the in-kernel flag waits; FM_P2P_SPIN_LOG2
+// shrinks it for the forced-timeout test
+static long long p2pSpinBound() {
+  static const long long b = [] {
+    const char* e = getenv("FM_P2P_SPIN_LOG2");
+    const int lg = e ? atoi(e) : 26;
+    return 1ll << (lg < 4 ? 4 : (lg > 40 ? 40 : lg));
+  }();
+  return b;
+}
+
+// a previously-completed k_await_flags timeout poisons the transport:
+// surface it at the next P2P entry (no sync needed - the flag is
+// host-mapped) and definitively via fm_p2p_error_check below
+static int p2pPoisonCheck() {
+  if (g.hP2pErr && *reinterpret_cast<volatile uint32_t*>(g.hP2pErr)) {
+    setErr("p2p exchange timed out (k_await_flags gave up; output poisoned)");
+    return FM_ERR_HIP;
+  }
+  return FM_OK;
+}
+
+int fm_p2p_error_check(void* stream) {
+  if (!g.heap) { setErr("fm_heap_init/connect first"); return FM_ERR_STATE; }
+  hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+  FM_HIP_CHECK(hipStreamSynchronize(st));
+  const int rc = p2pPoisonCheck();
+  if (rc != FM_OK && g.hP2pErr) *g.hP2pErr = 0;  // ack: allow retry
+  return rc;
+}
+
 int fm_dispatch_p2p(void* stream, const void* x) {
   if (!g.heap) { setErr("fm_heap_init/connect first"); return FM_ERR_STATE; }
+  int prc = p2pPoisonCheck();
+  if (prc != FM_OK) return prc;
   hipStream_t st = reinterpret_cast<hipStream_t>(stream);
   g.seq += 1;
   FM_HIP_CHECK(hipMemsetAsync(g.dArrive, 0, 2 * (size_t)g.E * 4, st));
@@ -2160,13 +2224,15 @@ int fm_dispatch_p2p(void* stream, const void* x) {
   hipLaunchKernelGGL(
       k_await_flags, dim3(1), dim3(256), 0, st,
       reinterpret_cast<const unsigned long long*>(g.heap + g.heapDFlagOff),
-      g.E, g.seq);
+      g.E, g.seq, g.dP2pErr, p2pSpinBound());
   FM_HIP_CHECK(hipGetLastError());
   return FM_OK;
 }
 
 int fm_return_p2p(void* stream, const void* ffn_out) {
   if (!g.heap) { setErr("fm_heap_init/connect first"); return FM_ERR_STATE; }
+  int prc = p2pPoisonCheck();
+  if (prc != FM_OK) return prc;
   hipStream_t st = reinterpret_cast<hipStream_t>(stream);
   const int nLx = g.E / g.world;
   dim3 grid(g.EC, g.world * nLx);
@@ -2183,7 +2249,7 @@ int fm_return_p2p(void* stream, const void* ffn_out) {
   hipLaunchKernelGGL(
       k_await_flags, dim3(1), dim3(256), 0, st,
       reinterpret_cast<const unsigned long long*>(g.heap + g.heapRFlagOff),
-      g.E, g.seq);
+      g.E, g.seq, g.dP2pErr, p2pSpinBound());
   FM_HIP_CHECK(hipGetLastError());
   return FM_OK;
 }

@@ -42,7 +42,9 @@ typedef struct fm_config {
   int32_t hidden_act; /* 0 relu, 1 gelu */
   int32_t hidden_size;        /* H */
   int32_t intermediate_size;  /* P */
-  int32_t sequence_len;
+  int32_t sequence_len;  /* sequence_len * mini_batch (= S, the token
+                          * count) must be a positive multiple of 128:
+                          * the gate tiles tokens in blocks of 128 */
   int32_t mini_batch;
   int32_t dtype;
   int32_t is_training;
@@ -191,6 +193,13 @@ int fm_dispatch_p2p(void* stream, const void* x);
 /* write FFN results back into sources' return heaps + signal; wait for
  * my own return cells */
 int fm_return_p2p(void* stream, const void* ffn_out);
+/* A timed-out in-kernel flag wait (bounded spin gave up) poisons the
+ * exchange: the kernel sets a host-mapped error word, the NEXT
+ * fm_dispatch_p2p/fm_return_p2p call returns FM_ERR_HIP without
+ * launching, and this call checks definitively (synchronizes the
+ * stream, returns FM_ERR_HIP if the word is set, then clears it so the
+ * caller may retry). */
+int fm_p2p_error_check(void* stream);
 
 /* Training-mode auxiliary-loss accumulators (gate.cuh:273-299,763-773;
  * types.cuh:936-958): gML[e] = mean softmax prob of expert e over the
