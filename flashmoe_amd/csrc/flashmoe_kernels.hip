@@ -386,6 +386,12 @@ struct GemmArgs {
   int totalJobs;             // persistent grid: total (m,n,e) tiles to
                              // cover (0 = one tile per block, classic)
   int jobsMT, jobsNT;        // tile grid dims when persistent
+  int atomicLogits;          // PHASE 3: atomicAdd into fp32 out (K-split
+                             // partial accumulate; fused gate)
+  int slotAlways;            // PHASE 1: write the per-(token,j) combine
+                             // slot even at topk==1 (scale 1; the fused
+                             // kernel always reduces via k_cast_combine
+                             // semantics so dropped tokens zero out)
 };
 
 // address-space helpers for global_load_lds (direct HBM->LDS DMA)
@@ -631,84 +637,56 @@ __global__ __launch_bounds__(256) void k_group_gemm_bf16(GemmArgs a) {
 // Same phase semantics/epilogues as k_group_gemm_bf16.
 // ---------------------------------------------------------------------------
 
-template <typename ET, int PHASE, int ACT, bool HAS_BIAS, int BN, int BM = 256,
-          typename WET = ET, int STAGES = 2>
-__global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
+// effective staging depth for a (BM, BN, WET, STAGES) tile: degrade
+// triple buffering to double when it would not fit the 160 KB LDS
+template <int BM, int BN, int BEZ, int STAGES>
+constexpr int gemm_se() {
+  return (STAGES == 3 &&
+          3 * BM * 64 * 2 + 3 * BN * 64 * BEZ + BM * 8 + 16 > 160 * 1024)
+             ? 2 : STAGES;
+}
+
+// LDS arena bytes the tile body needs (mirrored on the host when sizing
+// the fused kernel's dynamic arena)
+template <int BM, int BN, int BEZ, int STAGES>
+constexpr int gemm_lds_bytes() {
+  constexpr int SE = gemm_se<BM, BN, BEZ, STAGES>();
+  return SE * BM * 64 * 2 + SE * BN * 64 * BEZ + BM * 8 + 16;
+}
+
+// ---------------------------------------------------------------------------
+// One (e, m0, n0, ksplit) tile job of the big-tile grouped GEMM: the
+// shared body of k_group_gemm_bf16_big (classic multi-kernel path) and
+// k_moe_fused (single-launch persistent path). act / hasBias are
+// runtime here; the classic kernels pass template constants that fold
+// after inlining. Layout/state contracts documented at the kernel below.
+// ---------------------------------------------------------------------------
+template <typename ET, int PHASE, int BN, int BM, typename WET, int STAGES>
+__device__ __forceinline__ void gemm_job_body(
+    const GemmArgs& a, char* smemBase, int e, int ksplit, int m0, int n0,
+    int act, bool hasBias) {
   using vec8 = typename ETr<ET>::vec8;
   constexpr int BK = 64;
   constexpr int NF = BN / 64;           // B fragments per wave (4 or 2)
   constexpr int MI = BM / 32;           // A fragments per wave (8 or 4)
-  // WET: weight (B) storage element - == ET normally, fp8e4m3 for the
-  // config-5 fp8-weight path (W8A16). fp8 B is staged through glds as
-  // raw bytes (halving B's HBM + LDS traffic) and dequantized to bf16
-  // at fragment-read time; MFMA stream unchanged.
   constexpr int BEZ = (int)sizeof(WET);
   constexpr int BCH = BK * BEZ / 16;        // 16B chunks per B row
   constexpr int AGRP = BM * BK * 2 / 1024;  // glds 1KiB groups per A tile
   constexpr int BGRP = BN * BK * BEZ / 1024;
   constexpr int GPW_A = AGRP / 8, GPW_B = BGRP / 8;  // per wave
   constexpr int GPT = GPW_A + GPW_B;    // glds per wave per K-tile
-  // STAGES: staging pipeline depth. 2 = classic double buffer (stage t+1
-  // during compute t, then a vmcnt(0) drain per tile). 3 = triple buffer
-  // (stage t+2 during compute t, end-of-tile wait is a COUNTED
-  // vmcnt(GPT) that only requires t+1 landed) - each stage gets TWO
-  // compute phases of latency budget, absorbing L2/HBM staging jitter
-  // that the mfma_probe showed to be the real wall (the LDS+MFMA inner
-  // loop alone reaches 84% of peak; profiles/r01). Fits LDS only for
-  // the BNxBM = 128x256 / 256x128 tiles (144 KB).
-  // effective depth: degrade to double buffering when triple would not
-  // fit the 160 KB LDS (e.g. 256x256 bf16); fp8 B (BEZ==1) halves the B
-  // footprint, so even the 256-wide tiles triple-buffer there
-  constexpr int SE =
-      (STAGES == 3 &&
-       3 * BM * BK * 2 + 3 * BN * BK * BEZ + BM * 8 + 16 > 160 * 1024)
-          ? 2 : STAGES;
-  __shared__ __attribute__((aligned(16))) char smem[
-      SE * BM * BK * 2 + SE * BN * BK * BEZ + BM * 8 + 16];
-  ET* Abase = reinterpret_cast<ET*>(smem);          // SE x [BM][BK]
-  WET* Bbase = reinterpret_cast<WET*>(smem + SE * BM * BK * 2);
+  constexpr int SE = gemm_se<BM, BN, BEZ, STAGES>();
+  ET* Abase = reinterpret_cast<ET*>(smemBase);      // SE x [BM][BK]
+  WET* Bbase = reinterpret_cast<WET*>(smemBase + SE * BM * BK * 2);
   TPS* sTps = reinterpret_cast<TPS*>(
-      smem + SE * BM * BK * 2 + SE * BN * BK * BEZ);
+      smemBase + SE * BM * BK * 2 + SE * BN * BK * BEZ);
   uint32_t* sRouted = reinterpret_cast<uint32_t*>(sTps + BM);
 
-  // XCD-aware block remap (perf only, placement-independent for
-  // correctness): the dispatcher places linear block b on XCD b%8; the
-  // bijective remap gives each XCD a CONTIGUOUS chunk of (e, nTile)
-  // space with mTile fastest, so the blocks sharing a B weight panel
-  // (and, for small E, a whole expert) run on one XCD and hit its L2
-  // instead of re-reading HBM/L3 (guide T1; staging-traffic-bound at
-  // 6.2 TB/s before this, profiles/r01 pmc2).
-  const int mT = a.totalJobs > 0 ? a.jobsMT : gridDim.x;
-  const int nT = a.totalJobs > 0 ? a.jobsNT : gridDim.y;
-  const int gstride = gridDim.x * gridDim.y * gridDim.z;
-  const int nBlocks = a.totalJobs > 0 ? a.totalJobs : gstride;
-  const int lin0 =
-      blockIdx.x + gridDim.x * (blockIdx.y + gridDim.y * blockIdx.z);
-  const int qx = nBlocks / 8, rx = nBlocks % 8;
-  const int sk = a.splitK > 0 ? a.splitK : 1;
-  const int zE = (a.totalJobs > 0 ? a.totalJobs / (mT * nT)
-                                  : (int)gridDim.z) / sk;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
   const int K = a.K, N = a.N;
-  // persistent grid (totalJobs > 0): each block walks jobs lin0,
-  // lin0+gstride, ... - the per-launch block setup (sTps load, source
-  // pointer setup, I$ warm, prologue staging bubble) amortizes over
-  // several tiles instead of being paid once per 16-tile block
-  // (mfma_probe: the identical tile loop runs 52.5 us/forward-equiv
-  // persistent vs 71.4 us launched per-tile at the cfg2 up shape)
-  for (int jl = lin0; jl < nBlocks; jl += gstride) {
-  const int xcd = jl % 8, pos = jl / 8;
-  const int swz = a.noRemap
-      ? jl
-      : (xcd < rx ? xcd * (qx + 1) : rx * (qx + 1) + (xcd - rx) * qx) + pos;
-  const int eEff = swz / (mT * nT);  // in [0, E*splitK)
-  const int e = eEff % zE;
-  const int ksplit = eEff / zE;
-  const int rem = swz % (mT * nT);
-  const int m0 = (rem % mT) * BM;
-  const int n0 = (rem / mT) * BN;
+  const int sk = a.splitK > 0 ? a.splitK : 1;
   const int kLen = K / sk;            // this split's K range
   const int kStart = ksplit * kLen;
 
@@ -718,7 +696,7 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
     *sRouted = tpsE ? min(a.eC[e], (uint32_t)a.EC) : (uint32_t)a.nRows;
   __syncthreads();
   const uint32_t routed = *sRouted;
-  if ((uint32_t)m0 >= routed) continue;
+  if ((uint32_t)m0 >= routed) return;
   const int mCap = a.tokenIds ? a.pEC : a.nRows;  // A-row clamp bound
   const int we = a.segExpert ? a.segExpert[e] : e;  // weight/bias expert
   if (tid < BM) {
@@ -875,9 +853,13 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
   const int cl = lane & 15;
   const int r0 = (lane >> 4) * 4;
   float* sScale = reinterpret_cast<float*>(Abase);
-  const bool multi = (PHASE == 1) && a.topk > 1;
+  // slot: write the per-(token, j) combine slot; scaled only at k>1
+  // (the reference's k==1 CombineMode::single semantics are unscaled,
+  // processor.cuh:173-204)
+  const bool slot = (PHASE == 1) && (a.topk > 1 || a.slotAlways);
+  const bool scaled = (PHASE == 1) && a.topk > 1;
   if constexpr (PHASE == 1) {
-    if (multi) {
+    if (scaled) {
       __syncthreads();
       if (tid < BM) {
         const TPS tp = sTps[tid];
@@ -894,16 +876,14 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
   float bv[NF];
 #pragma unroll
   for (int ni = 0; ni < NF; ++ni) bv[ni] = 0.f;
-  if constexpr (HAS_BIAS) {
-    if (ksplit == 0) {  // split partials must add the bias exactly once
-      // per-expert bias slabs when multi-expert (see the 128^2 kernel)
-      const ET* bptr = reinterpret_cast<const ET*>(a.bias) +
-                       (a.strideBExpert ? (size_t)we * N : 0);
+  if (hasBias && ksplit == 0) {  // split partials add the bias exactly once
+    // per-expert bias slabs when multi-expert (see the 128^2 kernel)
+    const ET* bptr = reinterpret_cast<const ET*>(a.bias) +
+                     (a.strideBExpert ? (size_t)we * N : 0);
 #pragma unroll
-      for (int ni = 0; ni < NF; ++ni) {
-        const int col = n0 + wc * (BN / 4) + ni * 16 + cl;
-        if (col < N) bv[ni] = toF(bptr[col]);
-      }
+    for (int ni = 0; ni < NF; ++ni) {
+      const int col = n0 + wc * (BN / 4) + ni * 16 + cl;
+      if (col < N) bv[ni] = toF(bptr[col]);
     }
   }
 #pragma unroll
@@ -914,20 +894,20 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
       const int m = m0 + row;
       if ((uint32_t)m >= routed) continue;
       const TPS tp = sTps[row];
-      const float rowScale = multi ? sScale[row] : 1.0f;
+      const float rowScale = scaled ? sScale[row] : 1.0f;
 #pragma unroll
       for (int ni = 0; ni < NF; ++ni) {
         const int col = n0 + wc * (BN / 4) + ni * 16 + cl;
         if (col >= N) continue;
         float v = accv[mi][ni][r] + bv[ni];
         if constexpr (PHASE == 0) {
-          v = (ACT == 0) ? fmaxf(v, 0.0f)
+          v = (act == 0) ? fmaxf(v, 0.0f)
                          : 0.5f * v * (1.0f + erff(v * 0.70710678118654752f));
           reinterpret_cast<ET*>(a.out)[(size_t)e * a.strideOExpert +
                                          (size_t)m * N + col] =
               ETr<ET>::fromf(v);
         } else if constexpr (PHASE == 1) {
-          if (multi) {
+          if (slot) {
             // non-atomic per-(token, j) combine slot (summed with the
             // kept mask in k_cast_combine; replaces fp32 atomics)
             reinterpret_cast<ET*>(a.O32)[
@@ -939,9 +919,13 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
                 ETr<ET>::fromf(v);
           }
         } else if constexpr (PHASE == 3) {
-          // gate logits, fp32 (the small kernel handles the K-split
-          // variant; the big kernel is only selected with splitK == 1)
-          reinterpret_cast<float*>(a.out)[(size_t)m * N + col] = v;
+          // gate logits, fp32; atomic accumulate when the fused gate
+          // K-splits the logits GEMM over jobs (logits32 kept zero
+          // between forwards by the route's re-zeroing pass)
+          if (a.atomicLogits)
+            atomicAdd(&reinterpret_cast<float*>(a.out)[(size_t)m * N + col], v);
+          else
+            reinterpret_cast<float*>(a.out)[(size_t)m * N + col] = v;
         } else {
           reinterpret_cast<ET*>(a.out)[(size_t)e * a.strideOExpert +
                                        (size_t)m * N + col] =
@@ -950,7 +934,57 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
       }
     }
   }
-  }  // job loop (persistent grid)
+}
+
+// ---------------------------------------------------------------------------
+// Big-tile grouped GEMM kernel (classic multi-kernel path): XCD-aware
+// remap loop around gemm_job_body. See the body above for the tile
+// pipeline; DESIGN.md par.3 for the mode selection.
+// ---------------------------------------------------------------------------
+template <typename ET, int PHASE, int ACT, bool HAS_BIAS, int BN, int BM = 256,
+          typename WET = ET, int STAGES = 2>
+__global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
+  constexpr int BK = 64;
+  constexpr int BEZ = (int)sizeof(WET);
+  constexpr int SE = gemm_se<BM, BN, BEZ, STAGES>();
+  __shared__ __attribute__((aligned(16))) char smem[
+      SE * BM * BK * 2 + SE * BN * BK * BEZ + BM * 8 + 16];
+
+  // XCD-aware block remap (perf only, placement-independent for
+  // correctness): the dispatcher places linear block b on XCD b%8; the
+  // bijective remap gives each XCD a CONTIGUOUS chunk of (e, nTile)
+  // space with mTile fastest, so the blocks sharing a B weight panel
+  // (and, for small E, a whole expert) run on one XCD and hit its L2
+  // instead of re-reading HBM/L3 (guide T1; staging-traffic-bound at
+  // 6.2 TB/s before this, profiles/r01 pmc2).
+  const int mT = a.totalJobs > 0 ? a.jobsMT : gridDim.x;
+  const int nT = a.totalJobs > 0 ? a.jobsNT : gridDim.y;
+  const int gstride = gridDim.x * gridDim.y * gridDim.z;
+  const int nBlocks = a.totalJobs > 0 ? a.totalJobs : gstride;
+  const int lin0 =
+      blockIdx.x + gridDim.x * (blockIdx.y + gridDim.y * blockIdx.z);
+  const int qx = nBlocks / 8, rx = nBlocks % 8;
+  const int sk = a.splitK > 0 ? a.splitK : 1;
+  const int zE = (a.totalJobs > 0 ? a.totalJobs / (mT * nT)
+                                  : (int)gridDim.z) / sk;
+  // persistent grid (totalJobs > 0): each block walks jobs lin0,
+  // lin0+gstride, ... - the per-launch block setup (sTps load, source
+  // pointer setup, I$ warm, prologue staging bubble) amortizes over
+  // several tiles instead of being paid once per 16-tile block
+  // (mfma_probe: the identical tile loop runs 52.5 us/forward-equiv
+  // persistent vs 71.4 us launched per-tile at the cfg2 up shape)
+  for (int jl = lin0; jl < nBlocks; jl += gstride) {
+    const int xcd = jl % 8, pos = jl / 8;
+    const int swz = a.noRemap
+        ? jl
+        : (xcd < rx ? xcd * (qx + 1) : rx * (qx + 1) + (xcd - rx) * qx) + pos;
+    const int eEff = swz / (mT * nT);  // in [0, E*splitK)
+    const int e = eEff % zE;
+    const int ksplit = eEff / zE;
+    const int rem = swz % (mT * nT);
+    gemm_job_body<ET, PHASE, BN, BM, WET, STAGES>(
+        a, smem, e, ksplit, (rem % mT) * BM, (rem / mT) * BN, ACT, HAS_BIAS);
+  }
 }
 
 // ---------------------------------------------------------------------------
