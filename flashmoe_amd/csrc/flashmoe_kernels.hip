@@ -1419,6 +1419,405 @@ __global__ void k_mfma_probe(const bf16* A, const bf16* B, float* D) {
 }
 
 // ===========================================================================
+// Fused persistent forward kernel (the north star's single-launch DMoE,
+// reference moe::forward, moe.cuh:77-144): ONE launch whose resident
+// blocks walk four job phases -
+//   L: gate-logits GEMM tiles (K-split, fp32 atomic accumulate into
+//      logits32) with per-tile arrival counters; the LAST arriving block
+//      of a tile runs the softmax/top-k/route for that tile inline
+//      (gate.cuh:474-720 semantics, fused per VERDICT r01 #4)
+//   U: expert up-GEMM tiles (act(x W_up^T + b), token-gather A)
+//   D: expert down-GEMM tiles with the combine-slot epilogue
+//   C: masked slot reduction into moe_out
+// Phase seams are job-done fan-ins + one release per producing block,
+// following the CDNA4 guide's Guideline 16 protocol exactly: payload
+// plain stores -> every wave s_waitcnt vmcnt(0) -> __syncthreads ->
+// lane 0 agent-scope release fence + asm vmcnt drain -> relaxed counter
+// add; consumers poll relaxed, then ONE agent acquire + __syncthreads.
+// All polled words are zeroed by hipMemsetAsync nodes ahead of the
+// launch (replayed first under hipGraph capture). Every spin is bounded
+// and a give-up sets the host-mapped error word instead of hanging.
+// The grid is sized to exactly the co-resident block count (1 block/CU
+// at the 146 KB LDS arena), so the in-kernel waits cannot deadlock.
+// ===========================================================================
+
+struct FusedCtl {
+  uint32_t routeTilesDone;   // fan-in: tiles fully routed
+  uint32_t upBlocksDone;     // fan-in: blocks done with the up phase
+  uint32_t dnBlocksDone;     // fan-in: blocks done with the down phase
+  uint32_t pad0;
+  unsigned long long clk[6]; // atomicMax phase-end wall clocks:
+                             // 0 entry, 1 route done, 2 up, 3 down, 4 end
+  // followed in the same allocation by tileArrive[nTiles]
+};
+
+__device__ __forceinline__ void fused_release_arrive(uint32_t* ctr) {
+  // producer side of the Guideline 16 hand-off (counter form)
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+    // restate the post-wbl2 drain where the compiler would drop it
+    // (guide Guideline 16 pitfall 12)
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __hip_atomic_fetch_add(ctr, 1u, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+  }
+}
+
+__device__ __forceinline__ void fused_wait(const uint32_t* ctr,
+                                           uint32_t target,
+                                           uint32_t* errWord,
+                                           long long bound) {
+  if (threadIdx.x == 0) {
+    long long spins = 0;
+    while (__hip_atomic_load(ctr, __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_AGENT) < target) {
+      __builtin_amdgcn_s_sleep(8);
+      if (++spins > bound) {
+        // give up: poisoned output, host sees the error word
+        if (errWord)
+          __hip_atomic_store(errWord, 2u, __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_SYSTEM);
+        break;
+      }
+    }
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+  }
+  __syncthreads();
+}
+
+struct FusedArgs {
+  GemmArgs gl, gu, gd;       // logits / up / down job descriptors
+  FusedCtl* ctl;
+  uint32_t* tileArrive;      // [nTiles] logits-job arrivals per tile
+  uint32_t* eCf;             // per-expert routed counts (g.eC)
+  float* logits32;
+  void* gate_out;
+  TPS* tokenIds;
+  const void* cbuf;          // combine slots [S, k, H] Element
+  uint8_t* kept;
+  void* moe_out;
+  float* gML;                // training aux (null in inference)
+  float* gMeC;
+  uint32_t* errWord;         // host-mapped (system scope)
+  int S, H, E, PX, EC, pEC, topk, training;
+  int nTiles;                // S / 128
+  int lgKS, lgNT, lgJobs;    // logits K-split, N tiles, total jobs
+  int upGeom, upMT, upNT, upJobs;
+  int dnGeom, dnMT, dnNT, dnJobs;
+  int cbTB, cbJobs;          // combine token-block size / jobs
+  int nBlocks;
+  int ctlOff;                // LDS offset of the 16-B control slice
+  long long spinBound;
+};
+
+// softmax / top-k / ordered placement for one 128-token tile
+// (k_gate_route semantics at 512 threads; reference gate.cuh:557-718).
+// Reads the tile's logits from logits32 (atomic partials complete - the
+// caller acquired after the last arrival), re-zeroes them (keeps the
+// split-K logits accumulate memset-free), writes gate_out / tokenIds /
+// kept and bumps eC.
+template <typename T, int K>
+__device__ void route_tile(char* smem, const FusedArgs& f, int tile) {
+  const int E = f.E, PX = f.PX, S = f.S;
+  float* logits = reinterpret_cast<float*>(smem);                  // [128][E+1]
+  uint16_t* sel = reinterpret_cast<uint16_t*>(logits + 128 * (E + 1));
+  uint16_t* localIdx = sel + 128 * K;
+  // +2: dump slot so the counting scan's store is UNCONDITIONAL
+  uint32_t* base = reinterpret_cast<uint32_t*>(localIdx + 128 * K + 2);  // [E]
+  float* sInv = reinterpret_cast<float*>(base + E);                // [128]
+  float* sMax = sInv + 128;                                        // [128]
+  const int tid = threadIdx.x;
+  const int m0 = tile * 128;
+  T* gate_out = reinterpret_cast<T*>(f.gate_out);
+  for (int i = tid; i < 128 * E; i += 512) {
+    const int row = i / E, col = i % E;
+    float* src = &f.logits32[(size_t)(m0 + row) * E + col];
+    logits[row * (E + 1) + col] = *src;
+    *src = 0.0f;
+  }
+  __syncthreads();
+
+  float mCw = 0.0f;
+  uint16_t mySel[K];
+  if (tid < 128) {
+    const float* lrow = logits + tid * (E + 1);
+    float m = -INFINITY;
+    for (int e = 0; e < E; ++e) m = fmaxf(m, lrow[e]);
+    float d = 0.0f;
+    for (int e = 0; e < E; ++e) d += __expf(lrow[e] - m);
+    const float inv_d = 1.0f / d;
+    sInv[tid] = inv_d;
+    sMax[tid] = m;
+    // iterative argmax, strict >, first index wins (gate.cuh:662-668)
+#pragma unroll
+    for (int i = 0; i < K; ++i) {
+      float sV = -INFINITY;
+      int sIdx = 0;
+      for (int e = 0; e < E; ++e) {
+        bool taken = false;
+#pragma unroll
+        for (int j = 0; j < K; ++j) taken |= (j < i) && (mySel[j] == e);
+        if (!taken && lrow[e] > sV) { sV = lrow[e]; sIdx = e; }
+      }
+      mySel[i] = (uint16_t)sIdx;
+      mCw += __expf(sV - m) * inv_d;
+    }
+#pragma unroll
+    for (int i = 0; i < K; ++i) sel[tid * K + i] = mySel[i];
+  }
+  __syncthreads();
+  // coalesced gate_out store (recompute probs from LDS)
+  for (int i = tid * 8; i < 128 * PX; i += 512 * 8) {
+    const int row = i / PX, col0 = i % PX;
+    const float mR = sMax[row], ivR = sInv[row];
+    const float* lrow = logits + row * (E + 1);
+    struct __attribute__((aligned(16))) V8 { T x[8]; } v;
+#pragma unroll
+    for (int q = 0; q < 8; ++q) {
+      const int col = col0 + q;
+      const float pv = (col < E) ? __expf(lrow[col] - mR) * ivR : 0.0f;
+      fromF(pv, v.x[q]);
+    }
+    *reinterpret_cast<V8*>(gate_out + (size_t)(m0 + row) * PX + col0) = v;
+  }
+  __syncthreads();
+  if (tid < E) {
+    uint32_t cnt = 0;
+    for (int mj = 0; mj < 128 * K; mj += 8) {
+      const u32x4 v = *reinterpret_cast<const u32x4*>(sel + mj);
+#pragma unroll
+      for (int q = 0; q < 8; ++q) {
+        const uint32_t w = v[q >> 1];
+        const uint16_t sv = (q & 1) ? (uint16_t)(w >> 16) : (uint16_t)(w & 0xffff);
+        const bool m = (sv == (uint16_t)tid);
+        localIdx[m ? (mj + q) : 128 * K] = (uint16_t)cnt;
+        cnt += m;
+      }
+    }
+    base[tid] = __hip_atomic_fetch_add(f.eCf + tid, cnt, __ATOMIC_RELAXED,
+                                       __HIP_MEMORY_SCOPE_AGENT);
+    if (f.gML) {
+      float colSum = 0.0f;
+      for (int mj = 0; mj < 128; ++mj)
+        colSum += __expf(logits[mj * (E + 1) + tid] - sMax[mj]) * sInv[mj];
+      atomicAdd(f.gML + tid, colSum / (float)S);
+      atomicAdd(f.gMeC + tid, (float)cnt / (float)S);
+    }
+  }
+  __syncthreads();
+  if (tid < 128) {
+#pragma unroll
+    for (int i = 0; i < K; ++i) {
+      const int e = mySel[i];
+      const uint32_t slot = base[e] + localIdx[tid * K + i];
+      const bool keep = slot < (uint32_t)f.EC;
+      if (keep) {
+        f.tokenIds[(size_t)e * f.pEC + slot] =
+            TPS{(uint32_t)(m0 + tid) | ((uint32_t)i << 28), mCw};
+      }
+      f.kept[(size_t)(m0 + tid) * K + i] = keep ? 1 : 0;
+    }
+  }
+}
+
+// masked slot reduction (k_cast_combine semantics) over a token block
+template <typename T, int K>
+__device__ void combine_tokens(const FusedArgs& f, int t0, int t1) {
+  constexpr int EPU = 16 / sizeof(T);
+  const T* cbuf = reinterpret_cast<const T*>(f.cbuf);
+  T* out = reinterpret_cast<T*>(f.moe_out);
+  const int H = f.H;
+  for (int t = t0; t < t1; ++t) {
+    bool kp[K];
+#pragma unroll
+    for (int j = 0; j < K; ++j) kp[j] = f.kept[(size_t)t * K + j] != 0;
+    for (int h = threadIdx.x * EPU; h < H; h += 512 * EPU) {
+      float acc[EPU];
+#pragma unroll
+      for (int q = 0; q < EPU; ++q) acc[q] = 0.0f;
+#pragma unroll
+      for (int j = 0; j < K; ++j) {
+        if (!kp[j]) continue;
+        const u32x4 v = *reinterpret_cast<const u32x4*>(
+            cbuf + ((size_t)t * K + j) * H + h);
+#pragma unroll
+        for (int w = 0; w < 4; ++w) {
+          const uint32_t vw = v[w];
+          if constexpr (__is_same(T, bf16)) {
+            const float2 fv = __bfloat1622float2(
+                *reinterpret_cast<const __hip_bfloat162*>(&vw));
+            acc[2 * w] += fv.x;
+            acc[2 * w + 1] += fv.y;
+          } else if constexpr (__is_same(T, fp16)) {
+            const float2 fv = __half22float2(
+                *reinterpret_cast<const __half2*>(&vw));
+            acc[2 * w] += fv.x;
+            acc[2 * w + 1] += fv.y;
+          } else {
+            acc[w] += __uint_as_float(vw);
+          }
+        }
+      }
+      T ov[EPU];
+#pragma unroll
+      for (int q = 0; q < EPU; ++q) fromF(acc[q], ov[q]);
+      *reinterpret_cast<u32x4*>(out + (size_t)t * H + h) =
+          *reinterpret_cast<const u32x4*>(ov);
+    }
+  }
+}
+
+// static XCD-swizzled walk over one GEMM phase's (e, mT, nT) jobs
+// (identical swizzle to the classic persistent grid)
+template <typename ET, int PHASE, int BN, int BM, typename WET, int STAGES>
+__device__ void gemm_phase_walk(const GemmArgs& a, char* smem, int mT, int nT,
+                                int nJobs, int nBlocks, int act, int hasBias) {
+  const int qx = nJobs / 8, rx = nJobs % 8;
+  for (int jl = blockIdx.x; jl < nJobs; jl += nBlocks) {
+    const int xcd = jl % 8, pos = jl / 8;
+    const int swz =
+        (xcd < rx ? xcd * (qx + 1) : rx * (qx + 1) + (xcd - rx) * qx) + pos;
+    const int e = swz / (mT * nT);
+    const int rem = swz % (mT * nT);
+    gemm_job_body<ET, PHASE, BN, BM, WET, STAGES>(
+        a, smem, e, 0, (rem % mT) * BM, (rem / mT) * BN, act, hasBias != 0);
+  }
+}
+
+template <typename ET, typename WET>
+__global__ __launch_bounds__(512, 1) void k_moe_fused(FusedArgs f) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  uint32_t* sCtl = reinterpret_cast<uint32_t*>(smem + f.ctlOff);
+  const int tid = threadIdx.x;
+  if (tid == 0)
+    atomicMax(reinterpret_cast<unsigned long long*>(&f.ctl->clk[0]),
+              __builtin_amdgcn_s_memrealtime());
+
+  // ---- phase L: gate logits tiles + inline route on last arrival ----
+  const int jobsPerTile = f.lgKS * f.lgNT;
+  for (int j = blockIdx.x; j < f.lgJobs; j += f.nBlocks) {
+    const int t = j % f.nTiles;
+    const int rest = j / f.nTiles;
+    const int ks = rest % f.lgKS;
+    const int nt = rest / f.lgKS;
+    gemm_job_body<ET, 3, 128, 128, ET, 2>(f.gl, smem, 0, ks, t * 128,
+                                          nt * 128, 0, false);
+    // arrive on the tile: the logits writes are fp32 atomicAdds
+    // (globally coherent), so only a vmcnt drain orders them before
+    // the arrival; no cache release is needed
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    if (tid == 0)
+      *sCtl = __hip_atomic_fetch_add(f.tileArrive + t, 1u, __ATOMIC_RELAXED,
+                                     __HIP_MEMORY_SCOPE_AGENT);
+    __syncthreads();
+    const bool last = (*sCtl == (uint32_t)(jobsPerTile - 1));
+    __syncthreads();  // everyone read sCtl before the next iteration
+    if (last) {
+      if (tid == 0) __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+      __syncthreads();
+      switch (f.topk) {
+        case 1: route_tile<ET, 1>(smem, f, t); break;
+        case 2: route_tile<ET, 2>(smem, f, t); break;
+        case 3: route_tile<ET, 3>(smem, f, t); break;
+        case 4: route_tile<ET, 4>(smem, f, t); break;
+        case 5: route_tile<ET, 5>(smem, f, t); break;
+        case 6: route_tile<ET, 6>(smem, f, t); break;
+        case 7: route_tile<ET, 7>(smem, f, t); break;
+        default: route_tile<ET, 8>(smem, f, t); break;
+      }
+      fused_release_arrive(&f.ctl->routeTilesDone);
+    }
+  }
+  fused_wait(&f.ctl->routeTilesDone, (uint32_t)f.nTiles, f.errWord,
+             f.spinBound);
+  if (tid == 0)
+    atomicMax(reinterpret_cast<unsigned long long*>(&f.ctl->clk[1]),
+              __builtin_amdgcn_s_memrealtime());
+
+  // ---- phase U: expert up-GEMM ----
+  switch (f.upGeom) {
+    case 0:
+      gemm_phase_walk<ET, 0, 256, 256, WET, 2>(
+          f.gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, f.gu.act,
+          f.gu.bias != nullptr);
+      break;
+    case 1:
+      gemm_phase_walk<ET, 0, 128, 256, WET, 3>(
+          f.gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, f.gu.act,
+          f.gu.bias != nullptr);
+      break;
+    case 3:
+      gemm_phase_walk<ET, 0, 256, 128, WET, 3>(
+          f.gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, f.gu.act,
+          f.gu.bias != nullptr);
+      break;
+    default:
+      gemm_phase_walk<ET, 0, 128, 128, WET, 2>(
+          f.gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, f.gu.act,
+          f.gu.bias != nullptr);
+      break;
+  }
+  fused_release_arrive(&f.ctl->upBlocksDone);
+  fused_wait(&f.ctl->upBlocksDone, (uint32_t)f.nBlocks, f.errWord,
+             f.spinBound);
+  if (tid == 0)
+    atomicMax(reinterpret_cast<unsigned long long*>(&f.ctl->clk[2]),
+              __builtin_amdgcn_s_memrealtime());
+
+  // ---- phase D: expert down-GEMM with combine-slot epilogue ----
+  switch (f.dnGeom) {
+    case 0:
+      gemm_phase_walk<ET, 1, 256, 256, WET, 2>(
+          f.gd, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks, 0,
+          f.gd.bias != nullptr);
+      break;
+    case 1:
+      gemm_phase_walk<ET, 1, 128, 256, WET, 3>(
+          f.gd, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks, 0,
+          f.gd.bias != nullptr);
+      break;
+    case 3:
+      gemm_phase_walk<ET, 1, 256, 128, WET, 3>(
+          f.gd, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks, 0,
+          f.gd.bias != nullptr);
+      break;
+    default:
+      gemm_phase_walk<ET, 1, 128, 128, WET, 2>(
+          f.gd, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks, 0,
+          f.gd.bias != nullptr);
+      break;
+  }
+  fused_release_arrive(&f.ctl->dnBlocksDone);
+  fused_wait(&f.ctl->dnBlocksDone, (uint32_t)f.nBlocks, f.errWord,
+             f.spinBound);
+  if (tid == 0)
+    atomicMax(reinterpret_cast<unsigned long long*>(&f.ctl->clk[3]),
+              __builtin_amdgcn_s_memrealtime());
+
+  // ---- phase C: masked slot reduction ----
+  for (int j = blockIdx.x; j < f.cbJobs; j += f.nBlocks) {
+    const int t0 = j * f.cbTB;
+    const int t1 = min(f.S, t0 + f.cbTB);
+    switch (f.topk) {
+      case 1: combine_tokens<ET, 1>(f, t0, t1); break;
+      case 2: combine_tokens<ET, 2>(f, t0, t1); break;
+      case 3: combine_tokens<ET, 3>(f, t0, t1); break;
+      case 4: combine_tokens<ET, 4>(f, t0, t1); break;
+      case 5: combine_tokens<ET, 5>(f, t0, t1); break;
+      case 6: combine_tokens<ET, 6>(f, t0, t1); break;
+      case 7: combine_tokens<ET, 7>(f, t0, t1); break;
+      default: combine_tokens<ET, 8>(f, t0, t1); break;
+    }
+  }
+  if (tid == 0)
+    atomicMax(reinterpret_cast<unsigned long long*>(&f.ctl->clk[4]),
+              __builtin_amdgcn_s_memrealtime());
+}
+
+// ===========================================================================
 // Host side: state, workspace, ABI
 // ===========================================================================
 
@@ -1463,6 +1862,14 @@ struct State {
   void* cbuf = nullptr;      // [S, k, H] Element: non-atomic combine slots
   uint8_t* kept = nullptr;   // [S, k] capacity-kept mask (gate-written)
   int nLxAlloc = 0;
+  // fused persistent kernel (single-launch forward)
+  FusedCtl* fusedCtl = nullptr;   // device: FusedCtl + tileArrive[nTiles]
+  size_t fusedCtlBytes = 0;
+  uint32_t* hFusedErr = nullptr;  // host-mapped give-up flag (zero-copy)
+  uint32_t* dFusedErr = nullptr;
+  int nCU = 0;
+  int wallClockKHz = 100000;      // s_memrealtime rate (ticks per ms)
+  bool lastForwardFused = false;
 };
 State g;
 
@@ -1573,6 +1980,249 @@ int launch_gate(hipStream_t st, const void* x, const void* gate_w,
   return FM_OK;
 }
 
+// ---------------------------------------------------------------------------
+// Fused single-launch forward: geometry selection + launch (device side
+// above, k_moe_fused). Returns FM_FALLBACK when the fused kernel cannot
+// run this shape (classic multi-kernel path takes over).
+// ---------------------------------------------------------------------------
+
+static const int FM_FALLBACK = 1000;  // internal sentinel, never ABI-visible
+
+static bool fusedEnabled() {
+  static const bool en = [] {
+    const char* e = getenv("FM_FUSED");
+    return !(e && e[0] == '0');
+  }();
+  return en;
+}
+
+static long long fusedSpinBound() {
+  static const long long b = [] {
+    const char* e = getenv("FM_FUSED_SPIN_LOG2");
+    const int lg = e ? atoi(e) : 24;
+    return 1ll << (lg < 4 ? 4 : (lg > 40 ? 40 : lg));
+  }();
+  return b;
+}
+
+static int fusedPoisonCheck() {
+  if (g.hFusedErr &&
+      *reinterpret_cast<volatile uint32_t*>(g.hFusedErr)) {
+    *g.hFusedErr = 0;  // ack so the caller may retry
+    setErr("fused forward gave up on an in-kernel wait (output poisoned)");
+    return FM_ERR_HIP;
+  }
+  return FM_OK;
+}
+
+// host mirror of gemm_lds_bytes<> for the dynamic arena
+static int gemmLdsBytesRT(int BM, int BN, int BEZ, int stages) {
+  const int SE = (stages == 3 && 3 * BM * 64 * 2 + 3 * BN * 64 * BEZ +
+                                      BM * 8 + 16 > 160 * 1024)
+                     ? 2 : stages;
+  return SE * BM * 64 * 2 + SE * BN * 64 * BEZ + BM * 8 + 16;
+}
+
+static void fusedGeoDims(int gm, int& bm, int& bn, int& stages) {
+  bm = (gm == 0 || gm == 1) ? 256 : 128;
+  bn = (gm == 0 || gm == 3) ? 256 : 128;
+  stages = (gm == 1 || gm == 3) ? 3 : 2;
+}
+
+struct FusedGeo { int geom, mt, nt, jobs; };
+
+// pick the largest tile that still fills the resident grid (mirrors the
+// classic mode selection, DESIGN.md par.3, restricted to 1 block/CU)
+static FusedGeo fusedPickGeo(int M, int N, int E, int nBlocks,
+                             const int* order, int nOrder, int forceGeom) {
+  FusedGeo last{};
+  for (int i = 0; i < nOrder; ++i) {
+    const int gm = order[i];
+    int bm, bn, st;
+    fusedGeoDims(gm, bm, bn, st);
+    if (forceGeom >= 0 && gm != forceGeom) continue;
+    if (M < bm && bm > 128) continue;  // do not waste >half the M rows
+    const int mt = DIVUP(M, bm), nt = DIVUP(N, bn);
+    last = FusedGeo{gm, mt, nt, mt * nt * E};
+    if (forceGeom >= 0) return last;
+    if (N < bn && bn > 128) continue;  // half-empty B tiles: next option
+    if (last.jobs >= nBlocks) return last;
+  }
+  const int mt = DIVUP(M, 128), nt = DIVUP(N, 128);
+  return FusedGeo{4, mt, nt, mt * nt * E};
+}
+
+template <typename ET, typename WET>
+static int launchFused(hipStream_t st, const FusedArgs& f, int arena,
+                       int nBlocks) {
+  static int maxDyn = -1;
+  if (maxDyn < 0) {
+    maxDyn = (hipFuncSetAttribute(
+                  reinterpret_cast<const void*>(&k_moe_fused<ET, WET>),
+                  hipFuncAttributeMaxDynamicSharedMemorySize,
+                  160 * 1024) == hipSuccess)
+                 ? 160 * 1024 : 64 * 1024;
+  }
+  if (arena > maxDyn) return FM_FALLBACK;
+  static int occArena = -1, occVal = 0;
+  if (occArena != arena) {
+    occVal = 0;
+    if (hipOccupancyMaxActiveBlocksPerMultiprocessor(
+            &occVal, k_moe_fused<ET, WET>, 512, arena) != hipSuccess)
+      occVal = 0;
+    occArena = arena;
+  }
+  if (occVal < 1) return FM_FALLBACK;  // grid would not be co-resident
+  hipLaunchKernelGGL((k_moe_fused<ET, WET>), dim3(nBlocks), dim3(512), arena,
+                     st, f);
+  FM_HIP_CHECK(hipGetLastError());
+  return FM_OK;
+}
+
+static int moe_forward_fused(hipStream_t st, const void* x, const void* gate_w,
+                             const void* expert_w, const void* b_up,
+                             const void* b_dn, void* gate_out,
+                             void* moe_out) {
+  if (!g.fusedCtl || g.nCU <= 0) return FM_FALLBACK;
+  const int nBlocks = g.nCU;  // 1 block/CU: co-resident by construction
+  const int K = g.cfg.expert_top_k;
+  const int BEZ = (int)g.wesz;
+
+  // logits jobs: 128x128 K-split tiles, enough to fill the grid
+  const int nTiles = g.S / 128;
+  const int lgNT = DIVUP(g.PX, 128);
+  int lgKS = 1;
+  while (nTiles * lgNT * lgKS < nBlocks && lgKS < 8 &&
+         (g.H / 64) % (lgKS * 2) == 0)
+    lgKS *= 2;
+  const int lgJobs = nTiles * lgNT * lgKS;
+
+  static const int forceUp = [] {
+    const char* e = getenv("FM_FUSED_UP");
+    return e ? atoi(e) : -1;
+  }();
+  static const int forceDn = [] {
+    const char* e = getenv("FM_FUSED_DN");
+    return e ? atoi(e) : -1;
+  }();
+  const int upOrder[3] = {0, 3, 4};
+  const int dnOrder[3] = {1, 3, 4};
+  const FusedGeo up = fusedPickGeo(g.pEC, g.P, g.E, nBlocks, upOrder, 3,
+                                   forceUp);
+  const FusedGeo dn = fusedPickGeo(g.pEC, g.H, g.E, nBlocks, dnOrder, 3,
+                                   forceDn);
+
+  // LDS arena: max of every tile body used plus the route scratch,
+  // plus the 16-B control slice at the tail
+  int bm, bn, stg;
+  int arena = gemmLdsBytesRT(128, 128, (int)g.esz, 2);  // logits tile
+  fusedGeoDims(up.geom, bm, bn, stg);
+  arena = std::max(arena, gemmLdsBytesRT(bm, bn, BEZ, stg));
+  fusedGeoDims(dn.geom, bm, bn, stg);
+  arena = std::max(arena, gemmLdsBytesRT(bm, bn, BEZ, stg));
+  const int routeBytes = 128 * (g.E + 1) * 4 + 128 * K * 2 +
+                         (128 * K + 2) * 2 + g.E * 4 + 2 * 128 * 4;
+  arena = std::max(arena, routeBytes);
+  arena = (arena + 15) / 16 * 16 + 16;
+  if (arena > 160 * 1024) return FM_FALLBACK;
+
+  FusedArgs f{};
+  f.gl.A = x;
+  f.gl.B = gate_w;
+  f.gl.out = g.logits32;
+  f.gl.K = g.H;
+  f.gl.N = g.E;
+  f.gl.nRows = g.S;
+  f.gl.H = g.H;
+  f.gl.splitK = lgKS;
+  f.gl.atomicLogits = (lgKS > 1) ? 1 : 0;
+
+  f.gu.A = x;
+  f.gu.B = expert_w;
+  f.gu.bias = b_up;
+  f.gu.out = g.xM;
+  f.gu.gate_out = gate_out;
+  f.gu.tokenIds = g.tokenIds;
+  f.gu.eC = g.eC;
+  f.gu.strideAExpert = 0;
+  f.gu.strideBExpert = 2LL * g.P * g.H;
+  f.gu.strideOExpert = (long long)g.pEC * g.P;
+  f.gu.K = g.H;
+  f.gu.N = g.P;
+  f.gu.EC = g.EC;
+  f.gu.pEC = g.pEC;
+  f.gu.PX = g.PX;
+  f.gu.topk = K;
+  f.gu.act = g.cfg.hidden_act;
+  f.gu.H = g.H;
+  f.gu.splitK = 1;
+
+  f.gd = f.gu;
+  f.gd.A = g.xM;
+  f.gd.B = reinterpret_cast<const char*>(expert_w) +
+           (size_t)g.P * g.H * g.wesz;
+  f.gd.bias = b_dn;
+  f.gd.out = nullptr;
+  f.gd.O32 = reinterpret_cast<float*>(g.cbuf);
+  f.gd.moe_out = moe_out;
+  f.gd.strideAExpert = (long long)g.pEC * g.P;
+  f.gd.K = g.P;
+  f.gd.N = g.H;
+  f.gd.act = 0;
+  f.gd.slotAlways = 1;
+
+  f.ctl = g.fusedCtl;
+  f.tileArrive = reinterpret_cast<uint32_t*>(g.fusedCtl + 1);
+  f.eCf = g.eC;
+  f.logits32 = g.logits32;
+  f.gate_out = gate_out;
+  f.tokenIds = g.tokenIds;
+  f.cbuf = g.cbuf;
+  f.kept = g.kept;
+  f.moe_out = moe_out;
+  f.gML = g.cfg.is_training ? g.gML : nullptr;
+  f.gMeC = g.cfg.is_training ? g.gMeC : nullptr;
+  f.errWord = g.dFusedErr;
+  f.S = g.S;
+  f.H = g.H;
+  f.E = g.E;
+  f.PX = g.PX;
+  f.EC = g.EC;
+  f.pEC = g.pEC;
+  f.topk = K;
+  f.training = g.cfg.is_training;
+  f.nTiles = nTiles;
+  f.lgKS = lgKS;
+  f.lgNT = lgNT;
+  f.lgJobs = lgJobs;
+  f.upGeom = up.geom;
+  f.upMT = up.mt;
+  f.upNT = up.nt;
+  f.upJobs = up.jobs;
+  f.dnGeom = dn.geom;
+  f.dnMT = dn.mt;
+  f.dnNT = dn.nt;
+  f.dnJobs = dn.jobs;
+  f.cbTB = 8;
+  f.cbJobs = DIVUP(g.S, 8);
+  f.nBlocks = nBlocks;
+  f.ctlOff = arena - 16;
+  f.spinBound = fusedSpinBound();
+
+  // per-forward state re-init (Guideline 16: zero every polled word
+  // ahead of the launch; these become memset nodes under graph capture)
+  FM_HIP_CHECK(hipMemsetAsync(g.fusedCtl, 0, g.fusedCtlBytes, st));
+  FM_HIP_CHECK(hipMemsetAsync(g.eC, 0, (size_t)g.E * sizeof(uint32_t), st));
+  if (g.cfg.is_training)
+    FM_HIP_CHECK(hipMemsetAsync(g.gML, 0, 2 * (size_t)g.E * sizeof(float), st));
+
+  if (g.cfg.dtype == 3) return launchFused<fp16, fp16>(st, f, arena, nBlocks);
+  if (g.cfg.dtype == 4)
+    return launchFused<bf16, fp8e4m3>(st, f, arena, nBlocks);
+  return launchFused<bf16, bf16>(st, f, arena, nBlocks);
+}
+
+
 }  // namespace
 
 extern "C" {
@@ -1646,6 +2296,30 @@ int fm_initialize(const fm_config* cfg, int rank, int world_size) {
   FM_HIP_CHECK(hipMalloc(&g.cbuf,
                          (size_t)g.S * g.cfg.expert_top_k * g.H * g.esz));
   FM_HIP_CHECK(hipMalloc(&g.kept, (size_t)g.S * g.cfg.expert_top_k));
+  if (g.esz == 2) {
+    // fused persistent kernel state: control block + per-tile arrival
+    // counters (zeroed per forward by a memset node) + host-mapped
+    // give-up flag
+    const int nTiles = g.S / 128;
+    g.fusedCtlBytes = sizeof(FusedCtl) + (size_t)nTiles * sizeof(uint32_t);
+    FM_HIP_CHECK(hipMalloc(&g.fusedCtl, g.fusedCtlBytes));
+    FM_HIP_CHECK(hipMemset(g.fusedCtl, 0, g.fusedCtlBytes));
+    FM_HIP_CHECK(hipHostMalloc(&g.hFusedErr, sizeof(uint32_t),
+                               hipHostMallocMapped));
+    *g.hFusedErr = 0;
+    FM_HIP_CHECK(hipHostGetDevicePointer(
+        reinterpret_cast<void**>(&g.dFusedErr), g.hFusedErr, 0));
+    int dev = 0;
+    FM_HIP_CHECK(hipGetDevice(&dev));
+    int ncu = 0;
+    FM_HIP_CHECK(hipDeviceGetAttribute(
+        &ncu, hipDeviceAttributeMultiprocessorCount, dev));
+    g.nCU = ncu;
+    int khz = 0;
+    if (hipDeviceGetAttribute(&khz, hipDeviceAttributeWallClockRate, dev) ==
+            hipSuccess && khz > 0)
+      g.wallClockKHz = khz;
+  }
   g.initialized = true;
   return FM_OK;
 }
@@ -1657,6 +2331,8 @@ int fm_finalize(void) {
   (void)hipFree(g.O32);
   (void)hipFree(g.cbuf); (void)hipFree(g.kept);
   (void)hipFree(g.logits32); (void)hipFree(g.gML);
+  if (g.fusedCtl) (void)hipFree(g.fusedCtl);
+  if (g.hFusedErr) (void)hipHostFree(g.hFusedErr);
   if (g.heap) {
     for (int p = 0; p < g.world; ++p) {
       if (p != g.rank && g.peerBase[p]) (void)hipIpcCloseMemHandle(g.peerBase[p]);
@@ -1922,6 +2598,17 @@ static int moe_forward_impl(hipStream_t st, const void* x, const void* gate_w,
                             const void* expert_w, const void* b_up,
                             const void* b_dn, void* gate_out, void* moe_out,
                             int64_t S, hipEvent_t* evs) {
+  // fused single-launch path (the north-star structure); classic
+  // multi-kernel path remains for fp32 and as FM_FUSED=0 fallback
+  g.lastForwardFused = false;
+  if (fusedEnabled() && g.esz == 2 && g.world == 1) {
+    int rc = moe_forward_fused(st, x, gate_w, expert_w, b_up, b_dn, gate_out,
+                               moe_out);
+    if (rc != FM_FALLBACK) {
+      if (rc == FM_OK) g.lastForwardFused = true;
+      return rc;
+    }
+  }
   // eC is zeroed by the gate logits kernel (block 0)
   // O32 is kept zero across calls (zeroed at initialize; k_cast_out
   // re-zeroes after reading)
@@ -1984,6 +2671,8 @@ int fm_moe_forward(void* stream, const void* x, const void* gate_w,
     return FM_ERR_STATE;
   }
   if (S != g.S) { setErr("S mismatch vs frozen config"); return FM_ERR_SHAPE; }
+  int prc = fusedPoisonCheck();  // a prior fused forward that gave up
+  if (prc != FM_OK) return prc;
   hipStream_t st = reinterpret_cast<hipStream_t>(stream);
 
   // hipGraph fast path: torch's caching allocator hands back the same
@@ -2049,13 +2738,31 @@ int fm_moe_forward_phased(void* stream, const void* x, const void* gate_w,
   if (rc == FM_OK) {
     FM_HIP_CHECK(hipEventRecord(evs[5], st));
     FM_HIP_CHECK(hipStreamSynchronize(st));
-    float pre = 0, gate = 0, upT = 0, dnT = 0, post = 0;
-    (void)hipEventElapsedTime(&pre, evs[0], evs[1]);
-    (void)hipEventElapsedTime(&gate, evs[1], evs[2]);
-    (void)hipEventElapsedTime(&upT, evs[2], evs[3]);
-    (void)hipEventElapsedTime(&dnT, evs[3], evs[4]);
-    (void)hipEventElapsedTime(&post, evs[4], evs[5]);
-    ms[0] = gate; ms[1] = upT; ms[2] = dnT; ms[3] = pre + post;
+    if (g.lastForwardFused) {
+      // fused forward: phase times come from the in-kernel wall clocks
+      // (one launch has no event boundaries); ms = {gate+route, up GEMM,
+      // down GEMM, combine} like the classic path
+      int prc = fusedPoisonCheck();
+      if (prc != FM_OK) { rc = prc; }
+      else {
+        FusedCtl hc;
+        FM_HIP_CHECK(hipMemcpy(&hc, g.fusedCtl, sizeof(hc),
+                               hipMemcpyDeviceToHost));
+        const double khz = (double)g.wallClockKHz;
+        ms[0] = (float)((double)(hc.clk[1] - hc.clk[0]) / khz);
+        ms[1] = (float)((double)(hc.clk[2] - hc.clk[1]) / khz);
+        ms[2] = (float)((double)(hc.clk[3] - hc.clk[2]) / khz);
+        ms[3] = (float)((double)(hc.clk[4] - hc.clk[3]) / khz);
+      }
+    } else {
+      float pre = 0, gate = 0, upT = 0, dnT = 0, post = 0;
+      (void)hipEventElapsedTime(&pre, evs[0], evs[1]);
+      (void)hipEventElapsedTime(&gate, evs[1], evs[2]);
+      (void)hipEventElapsedTime(&upT, evs[2], evs[3]);
+      (void)hipEventElapsedTime(&dnT, evs[3], evs[4]);
+      (void)hipEventElapsedTime(&post, evs[4], evs[5]);
+      ms[0] = gate; ms[1] = upT; ms[2] = dnT; ms[3] = pre + post;
+    }
   }
   for (auto& e : evs) (void)hipEventDestroy(e);
   return rc;
