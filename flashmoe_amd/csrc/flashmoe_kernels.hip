@@ -1487,8 +1487,7 @@ __device__ __forceinline__ void fused_wait(const uint32_t* ctr,
   __syncthreads();
 }
 
-struct FusedArgs {
-  GemmArgs gl, gu, gd;       // logits / up / down job descriptors
+struct FusedMeta {
   FusedCtl* ctl;
   uint32_t* tileArrive;      // [nTiles] logits-job arrivals per tile
   uint32_t* eCf;             // per-expert routed counts (g.eC)
@@ -1519,7 +1518,8 @@ struct FusedArgs {
 // split-K logits accumulate memset-free), writes gate_out / tokenIds /
 // kept and bumps eC.
 template <typename T, int K>
-__device__ void route_tile(char* smem, const FusedArgs& f, int tile) {
+__device__ __forceinline__ void route_tile(char* smem, const FusedMeta& f,
+                                           int tile) {
   const int E = f.E, PX = f.PX, S = f.S;
   float* logits = reinterpret_cast<float*>(smem);                  // [128][E+1]
   uint16_t* sel = reinterpret_cast<uint16_t*>(logits + 128 * (E + 1));
@@ -1624,22 +1624,26 @@ __device__ void route_tile(char* smem, const FusedArgs& f, int tile) {
 
 // masked slot reduction (k_cast_combine semantics) over a token block
 template <typename T, int K>
-__device__ void combine_tokens(const FusedArgs& f, int t0, int t1) {
+__device__ __forceinline__ void combine_tokens(const FusedMeta& f, int t0,
+                                               int t1) {
   constexpr int EPU = 16 / sizeof(T);
   const T* cbuf = reinterpret_cast<const T*>(f.cbuf);
   T* out = reinterpret_cast<T*>(f.moe_out);
   const int H = f.H;
-  for (int t = t0; t < t1; ++t) {
-    bool kp[K];
-#pragma unroll
-    for (int j = 0; j < K; ++j) kp[j] = f.kept[(size_t)t * K + j] != 0;
-    for (int h = threadIdx.x * EPU; h < H; h += 512 * EPU) {
+  // flat (token, h-unit) sweep: keeps all 512 threads busy even when
+  // H/EPU < 512 (the per-token h loop left 3/4 of the block idle)
+  const int units = H / EPU;
+  const int total = (t1 - t0) * units;
+  for (int u = threadIdx.x; u < total; u += 512) {
+    const int t = t0 + u / units;
+    const int h = (u % units) * EPU;
+    {
       float acc[EPU];
 #pragma unroll
       for (int q = 0; q < EPU; ++q) acc[q] = 0.0f;
 #pragma unroll
       for (int j = 0; j < K; ++j) {
-        if (!kp[j]) continue;
+        if (f.kept[(size_t)t * K + j] == 0) continue;
         const u32x4 v = *reinterpret_cast<const u32x4*>(
             cbuf + ((size_t)t * K + j) * H + h);
 #pragma unroll
@@ -1672,7 +1676,7 @@ __device__ void combine_tokens(const FusedArgs& f, int t0, int t1) {
 // static XCD-swizzled walk over one GEMM phase's (e, mT, nT) jobs
 // (identical swizzle to the classic persistent grid)
 template <typename ET, int PHASE, int BN, int BM, typename WET, int STAGES>
-__device__ void gemm_phase_walk(const GemmArgs& a, char* smem, int mT, int nT,
+__device__ __forceinline__ void gemm_phase_walk(const GemmArgs& a, char* smem, int mT, int nT,
                                 int nJobs, int nBlocks, int act, int hasBias) {
   const int qx = nJobs / 8, rx = nJobs % 8;
   for (int jl = blockIdx.x; jl < nJobs; jl += nBlocks) {
@@ -1687,7 +1691,9 @@ __device__ void gemm_phase_walk(const GemmArgs& a, char* smem, int mT, int nT,
 }
 
 template <typename ET, typename WET>
-__global__ __launch_bounds__(512, 1) void k_moe_fused(FusedArgs f) {
+__global__ __launch_bounds__(512, 1) void k_moe_fused(GemmArgs gl, GemmArgs gu,
+                                                      GemmArgs gd,
+                                                      FusedMeta f) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   uint32_t* sCtl = reinterpret_cast<uint32_t*>(smem + f.ctlOff);
   const int tid = threadIdx.x;
@@ -1702,7 +1708,7 @@ __global__ __launch_bounds__(512, 1) void k_moe_fused(FusedArgs f) {
     const int rest = j / f.nTiles;
     const int ks = rest % f.lgKS;
     const int nt = rest / f.lgKS;
-    gemm_job_body<ET, 3, 128, 128, ET, 2>(f.gl, smem, 0, ks, t * 128,
+    gemm_job_body<ET, 3, 128, 128, ET, 2>(gl, smem, 0, ks, t * 128,
                                           nt * 128, 0, false);
     // arrive on the tile: the logits writes are fp32 atomicAdds
     // (globally coherent), so only a vmcnt drain orders them before
@@ -1741,23 +1747,23 @@ __global__ __launch_bounds__(512, 1) void k_moe_fused(FusedArgs f) {
   switch (f.upGeom) {
     case 0:
       gemm_phase_walk<ET, 0, 256, 256, WET, 2>(
-          f.gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, f.gu.act,
-          f.gu.bias != nullptr);
+          gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, gu.act,
+          gu.bias != nullptr);
       break;
     case 1:
       gemm_phase_walk<ET, 0, 128, 256, WET, 3>(
-          f.gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, f.gu.act,
-          f.gu.bias != nullptr);
+          gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, gu.act,
+          gu.bias != nullptr);
       break;
     case 3:
       gemm_phase_walk<ET, 0, 256, 128, WET, 3>(
-          f.gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, f.gu.act,
-          f.gu.bias != nullptr);
+          gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, gu.act,
+          gu.bias != nullptr);
       break;
     default:
       gemm_phase_walk<ET, 0, 128, 128, WET, 2>(
-          f.gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, f.gu.act,
-          f.gu.bias != nullptr);
+          gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, gu.act,
+          gu.bias != nullptr);
       break;
   }
   fused_release_arrive(&f.ctl->upBlocksDone);
@@ -1771,23 +1777,23 @@ __global__ __launch_bounds__(512, 1) void k_moe_fused(FusedArgs f) {
   switch (f.dnGeom) {
     case 0:
       gemm_phase_walk<ET, 1, 256, 256, WET, 2>(
-          f.gd, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks, 0,
-          f.gd.bias != nullptr);
+          gd, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks, 0,
+          gd.bias != nullptr);
       break;
     case 1:
       gemm_phase_walk<ET, 1, 128, 256, WET, 3>(
-          f.gd, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks, 0,
-          f.gd.bias != nullptr);
+          gd, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks, 0,
+          gd.bias != nullptr);
       break;
     case 3:
       gemm_phase_walk<ET, 1, 256, 128, WET, 3>(
-          f.gd, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks, 0,
-          f.gd.bias != nullptr);
+          gd, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks, 0,
+          gd.bias != nullptr);
       break;
     default:
       gemm_phase_walk<ET, 1, 128, 128, WET, 2>(
-          f.gd, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks, 0,
-          f.gd.bias != nullptr);
+          gd, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks, 0,
+          gd.bias != nullptr);
       break;
   }
   fused_release_arrive(&f.ctl->dnBlocksDone);
@@ -2053,8 +2059,9 @@ static FusedGeo fusedPickGeo(int M, int N, int E, int nBlocks,
 }
 
 template <typename ET, typename WET>
-static int launchFused(hipStream_t st, const FusedArgs& f, int arena,
-                       int nBlocks) {
+static int launchFused(hipStream_t st, const GemmArgs& gl,
+                       const GemmArgs& gu, const GemmArgs& gd,
+                       const FusedMeta& f, int arena, int nBlocks) {
   static int maxDyn = -1;
   if (maxDyn < 0) {
     maxDyn = (hipFuncSetAttribute(
@@ -2074,7 +2081,7 @@ static int launchFused(hipStream_t st, const FusedArgs& f, int arena,
   }
   if (occVal < 1) return FM_FALLBACK;  // grid would not be co-resident
   hipLaunchKernelGGL((k_moe_fused<ET, WET>), dim3(nBlocks), dim3(512), arena,
-                     st, f);
+                     st, gl, gu, gd, f);
   FM_HIP_CHECK(hipGetLastError());
   return FM_OK;
 }
@@ -2126,50 +2133,51 @@ static int moe_forward_fused(hipStream_t st, const void* x, const void* gate_w,
   arena = (arena + 15) / 16 * 16 + 16;
   if (arena > 160 * 1024) return FM_FALLBACK;
 
-  FusedArgs f{};
-  f.gl.A = x;
-  f.gl.B = gate_w;
-  f.gl.out = g.logits32;
-  f.gl.K = g.H;
-  f.gl.N = g.E;
-  f.gl.nRows = g.S;
-  f.gl.H = g.H;
-  f.gl.splitK = lgKS;
-  f.gl.atomicLogits = (lgKS > 1) ? 1 : 0;
+  GemmArgs gl{}, gu{}, gd{};
+  FusedMeta f{};
+  gl.A = x;
+  gl.B = gate_w;
+  gl.out = g.logits32;
+  gl.K = g.H;
+  gl.N = g.E;
+  gl.nRows = g.S;
+  gl.H = g.H;
+  gl.splitK = lgKS;
+  gl.atomicLogits = (lgKS > 1) ? 1 : 0;
 
-  f.gu.A = x;
-  f.gu.B = expert_w;
-  f.gu.bias = b_up;
-  f.gu.out = g.xM;
-  f.gu.gate_out = gate_out;
-  f.gu.tokenIds = g.tokenIds;
-  f.gu.eC = g.eC;
-  f.gu.strideAExpert = 0;
-  f.gu.strideBExpert = 2LL * g.P * g.H;
-  f.gu.strideOExpert = (long long)g.pEC * g.P;
-  f.gu.K = g.H;
-  f.gu.N = g.P;
-  f.gu.EC = g.EC;
-  f.gu.pEC = g.pEC;
-  f.gu.PX = g.PX;
-  f.gu.topk = K;
-  f.gu.act = g.cfg.hidden_act;
-  f.gu.H = g.H;
-  f.gu.splitK = 1;
+  gu.A = x;
+  gu.B = expert_w;
+  gu.bias = b_up;
+  gu.out = g.xM;
+  gu.gate_out = gate_out;
+  gu.tokenIds = g.tokenIds;
+  gu.eC = g.eC;
+  gu.strideAExpert = 0;
+  gu.strideBExpert = 2LL * g.P * g.H;
+  gu.strideOExpert = (long long)g.pEC * g.P;
+  gu.K = g.H;
+  gu.N = g.P;
+  gu.EC = g.EC;
+  gu.pEC = g.pEC;
+  gu.PX = g.PX;
+  gu.topk = K;
+  gu.act = g.cfg.hidden_act;
+  gu.H = g.H;
+  gu.splitK = 1;
 
-  f.gd = f.gu;
-  f.gd.A = g.xM;
-  f.gd.B = reinterpret_cast<const char*>(expert_w) +
+  gd = gu;
+  gd.A = g.xM;
+  gd.B = reinterpret_cast<const char*>(expert_w) +
            (size_t)g.P * g.H * g.wesz;
-  f.gd.bias = b_dn;
-  f.gd.out = nullptr;
-  f.gd.O32 = reinterpret_cast<float*>(g.cbuf);
-  f.gd.moe_out = moe_out;
-  f.gd.strideAExpert = (long long)g.pEC * g.P;
-  f.gd.K = g.P;
-  f.gd.N = g.H;
-  f.gd.act = 0;
-  f.gd.slotAlways = 1;
+  gd.bias = b_dn;
+  gd.out = nullptr;
+  gd.O32 = reinterpret_cast<float*>(g.cbuf);
+  gd.moe_out = moe_out;
+  gd.strideAExpert = (long long)g.pEC * g.P;
+  gd.K = g.P;
+  gd.N = g.H;
+  gd.act = 0;
+  gd.slotAlways = 1;
 
   f.ctl = g.fusedCtl;
   f.tileArrive = reinterpret_cast<uint32_t*>(g.fusedCtl + 1);
@@ -2216,10 +2224,11 @@ static int moe_forward_fused(hipStream_t st, const void* x, const void* gate_w,
   if (g.cfg.is_training)
     FM_HIP_CHECK(hipMemsetAsync(g.gML, 0, 2 * (size_t)g.E * sizeof(float), st));
 
-  if (g.cfg.dtype == 3) return launchFused<fp16, fp16>(st, f, arena, nBlocks);
+  if (g.cfg.dtype == 3)
+    return launchFused<fp16, fp16>(st, gl, gu, gd, f, arena, nBlocks);
   if (g.cfg.dtype == 4)
-    return launchFused<bf16, fp8e4m3>(st, f, arena, nBlocks);
-  return launchFused<bf16, bf16>(st, f, arena, nBlocks);
+    return launchFused<bf16, fp8e4m3>(st, gl, gu, gd, f, arena, nBlocks);
+  return launchFused<bf16, bf16>(st, gl, gu, gd, f, arena, nBlocks);
 }
 
 
