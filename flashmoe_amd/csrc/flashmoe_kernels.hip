@@ -79,6 +79,21 @@ __device__ __forceinline__ bf16x8 dequant_fp8x8_bf16(uint32_t lo, uint32_t hi) {
   return r;
 }
 
+// 2-byte element store, optionally agent-scope write-through (sc1) so
+// same-launch consumers on other CUs/XCDs observe it without the
+// producer issuing a cache-flushing release fence (Guideline 16 R1)
+template <typename ET>
+__device__ __forceinline__ void storeElem(ET* p, ET v, bool sc1) {
+  if (sc1) {
+    uint16_t b;
+    __builtin_memcpy(&b, &v, 2);
+    __hip_atomic_store(reinterpret_cast<uint16_t*>(p), b, __ATOMIC_RELAXED,
+                       __HIP_MEMORY_SCOPE_AGENT);
+  } else {
+    *p = v;
+  }
+}
+
 __device__ __forceinline__ float toF(float v) { return v; }
 __device__ __forceinline__ float toF(bf16 v) { return __bfloat162float(v); }
 __device__ __forceinline__ float toF(fp16 v) { return __half2float(v); }
@@ -392,6 +407,11 @@ struct GemmArgs {
                              // slot even at topk==1 (scale 1; the fused
                              // kernel always reduces via k_cast_combine
                              // semantics so dropped tokens zero out)
+  int sc1Out;                // epilogue stores agent-scope write-through
+                             // (Guideline 16 R1 publish-large): in-launch
+                             // consumers on other CUs see them after a
+                             // vmcnt drain + job-count arrival, with no
+                             // per-block release fence
 };
 
 // address-space helpers for global_load_lds (direct HBM->LDS DMA)
@@ -903,16 +923,17 @@ __device__ __forceinline__ void gemm_job_body(
         if constexpr (PHASE == 0) {
           v = (act == 0) ? fmaxf(v, 0.0f)
                          : 0.5f * v * (1.0f + erff(v * 0.70710678118654752f));
-          reinterpret_cast<ET*>(a.out)[(size_t)e * a.strideOExpert +
-                                         (size_t)m * N + col] =
-              ETr<ET>::fromf(v);
+          storeElem(&reinterpret_cast<ET*>(a.out)[(size_t)e * a.strideOExpert +
+                                                  (size_t)m * N + col],
+                    ETr<ET>::fromf(v), a.sc1Out != 0);
         } else if constexpr (PHASE == 1) {
           if (slot) {
             // non-atomic per-(token, j) combine slot (summed with the
             // kept mask in k_cast_combine; replaces fp32 atomics)
-            reinterpret_cast<ET*>(a.O32)[
-                ((size_t)tpsTok(tp.tokenIdx) * a.topk + tpsJ(tp.tokenIdx)) *
-                    a.H + col] = ETr<ET>::fromf(v * rowScale);
+            storeElem(&reinterpret_cast<ET*>(a.O32)[
+                          ((size_t)tpsTok(tp.tokenIdx) * a.topk +
+                           tpsJ(tp.tokenIdx)) * a.H + col],
+                      ETr<ET>::fromf(v * rowScale), a.sc1Out != 0);
           } else {
             reinterpret_cast<ET*>(
                 a.moe_out)[(size_t)tpsTok(tp.tokenIdx) * a.H + col] =
@@ -1443,8 +1464,8 @@ __global__ void k_mfma_probe(const bf16* A, const bf16* B, float* D) {
 
 struct FusedCtl {
   uint32_t routeTilesDone;   // fan-in: tiles fully routed
-  uint32_t upBlocksDone;     // fan-in: blocks done with the up phase
-  uint32_t dnBlocksDone;     // fan-in: blocks done with the down phase
+  uint32_t upJobsDone;       // fan-in: up-GEMM jobs completed
+  uint32_t dnJobsDone;       // fan-in: down-GEMM jobs completed
   uint32_t pad0;
   unsigned long long clk[6]; // atomicMax phase-end wall clocks:
                              // 0 entry, 1 route done, 2 up, 3 down, 4 end
@@ -1676,8 +1697,11 @@ __device__ __forceinline__ void combine_tokens(const FusedMeta& f, int t0,
 // static XCD-swizzled walk over one GEMM phase's (e, mT, nT) jobs
 // (identical swizzle to the classic persistent grid)
 template <typename ET, int PHASE, int BN, int BM, typename WET, int STAGES>
-__device__ __forceinline__ void gemm_phase_walk(const GemmArgs& a, char* smem, int mT, int nT,
-                                int nJobs, int nBlocks, int act, int hasBias) {
+__device__ __forceinline__ void gemm_phase_walk(const GemmArgs& a, char* smem,
+                                                int mT, int nT, int nJobs,
+                                                int nBlocks, int act,
+                                                int hasBias,
+                                                uint32_t* doneCtr) {
   const int qx = nJobs / 8, rx = nJobs % 8;
   for (int jl = blockIdx.x; jl < nJobs; jl += nBlocks) {
     const int xcd = jl % 8, pos = jl / 8;
@@ -1687,6 +1711,14 @@ __device__ __forceinline__ void gemm_phase_walk(const GemmArgs& a, char* smem, i
     const int rem = swz % (mT * nT);
     gemm_job_body<ET, PHASE, BN, BM, WET, STAGES>(
         a, smem, e, 0, (rem % mT) * BM, (rem / mT) * BN, act, hasBias != 0);
+    // job-count arrival: the epilogue's sc1 write-through stores need
+    // only a per-wave drain before the relaxed arrival (Guideline 16
+    // R1) - no cache-flushing release fence per block
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    if (threadIdx.x == 0)
+      __hip_atomic_fetch_add(doneCtr, 1u, __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_AGENT);
   }
 }
 
@@ -1748,26 +1780,25 @@ __global__ __launch_bounds__(512, 1) void k_moe_fused(GemmArgs gl, GemmArgs gu,
     case 0:
       gemm_phase_walk<ET, 0, 256, 256, WET, 2>(
           gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, gu.act,
-          gu.bias != nullptr);
+          gu.bias != nullptr, &f.ctl->upJobsDone);
       break;
     case 1:
       gemm_phase_walk<ET, 0, 128, 256, WET, 3>(
           gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, gu.act,
-          gu.bias != nullptr);
+          gu.bias != nullptr, &f.ctl->upJobsDone);
       break;
     case 3:
       gemm_phase_walk<ET, 0, 256, 128, WET, 3>(
           gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, gu.act,
-          gu.bias != nullptr);
+          gu.bias != nullptr, &f.ctl->upJobsDone);
       break;
     default:
       gemm_phase_walk<ET, 0, 128, 128, WET, 2>(
           gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, gu.act,
-          gu.bias != nullptr);
+          gu.bias != nullptr, &f.ctl->upJobsDone);
       break;
   }
-  fused_release_arrive(&f.ctl->upBlocksDone);
-  fused_wait(&f.ctl->upBlocksDone, (uint32_t)f.nBlocks, f.errWord,
+  fused_wait(&f.ctl->upJobsDone, (uint32_t)f.upJobs, f.errWord,
              f.spinBound);
   if (tid == 0)
     atomicMax(reinterpret_cast<unsigned long long*>(&f.ctl->clk[2]),
@@ -1778,26 +1809,25 @@ __global__ __launch_bounds__(512, 1) void k_moe_fused(GemmArgs gl, GemmArgs gu,
     case 0:
       gemm_phase_walk<ET, 1, 256, 256, WET, 2>(
           gd, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks, 0,
-          gd.bias != nullptr);
+          gd.bias != nullptr, &f.ctl->dnJobsDone);
       break;
     case 1:
       gemm_phase_walk<ET, 1, 128, 256, WET, 3>(
           gd, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks, 0,
-          gd.bias != nullptr);
+          gd.bias != nullptr, &f.ctl->dnJobsDone);
       break;
     case 3:
       gemm_phase_walk<ET, 1, 256, 128, WET, 3>(
           gd, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks, 0,
-          gd.bias != nullptr);
+          gd.bias != nullptr, &f.ctl->dnJobsDone);
       break;
     default:
       gemm_phase_walk<ET, 1, 128, 128, WET, 2>(
           gd, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks, 0,
-          gd.bias != nullptr);
+          gd.bias != nullptr, &f.ctl->dnJobsDone);
       break;
   }
-  fused_release_arrive(&f.ctl->dnBlocksDone);
-  fused_wait(&f.ctl->dnBlocksDone, (uint32_t)f.nBlocks, f.errWord,
+  fused_wait(&f.ctl->dnJobsDone, (uint32_t)f.dnJobs, f.errWord,
              f.spinBound);
   if (tid == 0)
     atomicMax(reinterpret_cast<unsigned long long*>(&f.ctl->clk[3]),
@@ -2058,10 +2088,11 @@ static FusedGeo fusedPickGeo(int M, int N, int E, int nBlocks,
   return FusedGeo{4, mt, nt, mt * nt * E};
 }
 
+// co-resident blocks/CU for the fused kernel at a given arena size
+// (0 = cannot run). The grid is sized to exactly occ x nCU so every
+// block is resident and the in-kernel fan-in waits cannot deadlock.
 template <typename ET, typename WET>
-static int launchFused(hipStream_t st, const GemmArgs& gl,
-                       const GemmArgs& gu, const GemmArgs& gd,
-                       const FusedMeta& f, int arena, int nBlocks) {
+static int fusedOccT(int arena) {
   static int maxDyn = -1;
   if (maxDyn < 0) {
     maxDyn = (hipFuncSetAttribute(
@@ -2070,7 +2101,7 @@ static int launchFused(hipStream_t st, const GemmArgs& gl,
                   160 * 1024) == hipSuccess)
                  ? 160 * 1024 : 64 * 1024;
   }
-  if (arena > maxDyn) return FM_FALLBACK;
+  if (arena > maxDyn) return 0;
   static int occArena = -1, occVal = 0;
   if (occArena != arena) {
     occVal = 0;
@@ -2079,7 +2110,19 @@ static int launchFused(hipStream_t st, const GemmArgs& gl,
       occVal = 0;
     occArena = arena;
   }
-  if (occVal < 1) return FM_FALLBACK;  // grid would not be co-resident
+  return occVal;
+}
+
+static int fusedOcc(int arena) {
+  if (g.cfg.dtype == 3) return fusedOccT<fp16, fp16>(arena);
+  if (g.cfg.dtype == 4) return fusedOccT<bf16, fp8e4m3>(arena);
+  return fusedOccT<bf16, bf16>(arena);
+}
+
+template <typename ET, typename WET>
+static int launchFused(hipStream_t st, const GemmArgs& gl,
+                       const GemmArgs& gu, const GemmArgs& gd,
+                       const FusedMeta& f, int arena, int nBlocks) {
   hipLaunchKernelGGL((k_moe_fused<ET, WET>), dim3(nBlocks), dim3(512), arena,
                      st, gl, gu, gd, f);
   FM_HIP_CHECK(hipGetLastError());
@@ -2091,7 +2134,7 @@ static int moe_forward_fused(hipStream_t st, const void* x, const void* gate_w,
                              const void* b_dn, void* gate_out,
                              void* moe_out) {
   if (!g.fusedCtl || g.nCU <= 0) return FM_FALLBACK;
-  const int nBlocks = g.nCU;  // 1 block/CU: co-resident by construction
+  int nBlocks = g.nCU;  // refined to occ x nCU once the arena is known
   const int K = g.cfg.expert_top_k;
   const int BEZ = (int)g.wesz;
 
@@ -2132,6 +2175,9 @@ static int moe_forward_fused(hipStream_t st, const void* x, const void* gate_w,
   arena = std::max(arena, routeBytes);
   arena = (arena + 15) / 16 * 16 + 16;
   if (arena > 160 * 1024) return FM_FALLBACK;
+  const int occ = fusedOcc(arena);
+  if (occ < 1) return FM_FALLBACK;
+  nBlocks = std::min(occ, 2) * g.nCU;  // all blocks co-resident
 
   GemmArgs gl{}, gu{}, gd{};
   FusedMeta f{};
@@ -2143,7 +2189,9 @@ static int moe_forward_fused(hipStream_t st, const void* x, const void* gate_w,
   gl.nRows = g.S;
   gl.H = g.H;
   gl.splitK = lgKS;
-  gl.atomicLogits = (lgKS > 1) ? 1 : 0;
+  // always atomic: the route runs in a DIFFERENT block than the logits
+  // tiles, and atomics are globally coherent without a release fence
+  gl.atomicLogits = 1;
 
   gu.A = x;
   gu.B = expert_w;
@@ -2164,6 +2212,7 @@ static int moe_forward_fused(hipStream_t st, const void* x, const void* gate_w,
   gu.act = g.cfg.hidden_act;
   gu.H = g.H;
   gu.splitK = 1;
+  gu.sc1Out = 1;  // xM consumed by down-GEMM blocks in this launch
 
   gd = gu;
   gd.A = g.xM;
