@@ -682,7 +682,7 @@ constexpr int gemm_lds_bytes() {
 // after inlining. Layout/state contracts documented at the kernel below.
 // ---------------------------------------------------------------------------
 template <typename ET, int PHASE, int BN, int BM, typename WET, int STAGES>
-__device__ __forceinline__ void gemm_job_body(
+__device__ __forceinline__ bool gemm_job_body(
     const GemmArgs& a, char* smemBase, int e, int ksplit, int m0, int n0,
     int act, bool hasBias) {
   using vec8 = typename ETr<ET>::vec8;
@@ -716,7 +716,7 @@ __device__ __forceinline__ void gemm_job_body(
     *sRouted = tpsE ? min(a.eC[e], (uint32_t)a.EC) : (uint32_t)a.nRows;
   __syncthreads();
   const uint32_t routed = *sRouted;
-  if ((uint32_t)m0 >= routed) return;
+  if ((uint32_t)m0 >= routed) return false;  // empty tile
   const int mCap = a.tokenIds ? a.pEC : a.nRows;  // A-row clamp bound
   const int we = a.segExpert ? a.segExpert[e] : e;  // weight/bias expert
   if (tid < BM) {
@@ -955,6 +955,7 @@ __device__ __forceinline__ void gemm_job_body(
       }
     }
   }
+  return true;
 }
 
 // ---------------------------------------------------------------------------
@@ -1003,7 +1004,7 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
     const int e = eEff % zE;
     const int ksplit = eEff / zE;
     const int rem = swz % (mT * nT);
-    gemm_job_body<ET, PHASE, BN, BM, WET, STAGES>(
+    (void)gemm_job_body<ET, PHASE, BN, BM, WET, STAGES>(
         a, smem, e, ksplit, (rem % mT) * BM, (rem / mT) * BN, ACT, HAS_BIAS);
   }
 }
@@ -1511,6 +1512,8 @@ __device__ __forceinline__ void fused_wait(const uint32_t* ctr,
 struct FusedMeta {
   FusedCtl* ctl;
   uint32_t* tileArrive;      // [nTiles] logits-job arrivals per tile
+  uint32_t* upRowDone;       // [E, upMT] completed up n-tiles per row-tile
+  int upBM, upRowTarget;     // up tile rows / n-tiles per complete row
   uint32_t* eCf;             // per-expert routed counts (g.eC)
   float* logits32;
   void* gate_out;
@@ -1701,7 +1704,8 @@ __device__ __forceinline__ void gemm_phase_walk(const GemmArgs& a, char* smem,
                                                 int mT, int nT, int nJobs,
                                                 int nBlocks, int act,
                                                 int hasBias,
-                                                uint32_t* doneCtr) {
+                                                uint32_t* doneCtr,
+                                                uint32_t* rowDone) {
   const int qx = nJobs / 8, rx = nJobs % 8;
   for (int jl = blockIdx.x; jl < nJobs; jl += nBlocks) {
     const int xcd = jl % 8, pos = jl / 8;
@@ -1709,13 +1713,78 @@ __device__ __forceinline__ void gemm_phase_walk(const GemmArgs& a, char* smem,
         (xcd < rx ? xcd * (qx + 1) : rx * (qx + 1) + (xcd - rx) * qx) + pos;
     const int e = swz / (mT * nT);
     const int rem = swz % (mT * nT);
-    gemm_job_body<ET, PHASE, BN, BM, WET, STAGES>(
+    const bool ran = gemm_job_body<ET, PHASE, BN, BM, WET, STAGES>(
         a, smem, e, 0, (rem % mT) * BM, (rem / mT) * BN, act, hasBias != 0);
     // job-count arrival: the epilogue's sc1 write-through stores need
     // only a per-wave drain before the relaxed arrival (Guideline 16
-    // R1) - no cache-flushing release fence per block
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
+    // R1) - no cache-flushing release fence per block. Row arrivals
+    // (rowDone) progressively unlock the down phase's dependent tiles.
+    if (ran) {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+      __hip_atomic_fetch_add(doneCtr, 1u, __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_AGENT);
+      if (ran && rowDone)
+        __hip_atomic_fetch_add(rowDone + (size_t)e * mT + (rem % mT), 1u,
+                               __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    }
+  }
+}
+
+// down-phase walk: each job first waits (bounded) for the up row-tiles
+// covering its A rows, takes ONE agent acquire, then runs - the down
+// phase overlaps the up phase's tail instead of waiting for all of it
+template <typename ET, int BN, int BM, typename WET, int STAGES>
+__device__ __forceinline__ void gemm_phase_walk_dn(
+    const GemmArgs& a, const FusedMeta& f, char* smem, int mT, int nT,
+    int nJobs, int nBlocks, int hasBias, uint32_t* doneCtr) {
+  const int qx = nJobs / 8, rx = nJobs % 8;
+  for (int jl = blockIdx.x; jl < nJobs; jl += nBlocks) {
+    const int xcd = jl % 8, pos = jl / 8;
+    const int swz =
+        (xcd < rx ? xcd * (qx + 1) : rx * (qx + 1) + (xcd - rx) * qx) + pos;
+    const int e = swz / (mT * nT);
+    const int rem = swz % (mT * nT);
+    const int m0 = (rem % mT) * BM;
+    // routed counts are final (route fan-in acquired earlier)
+    const uint32_t rClip =
+        min(__hip_atomic_load(f.eCf + e, __ATOMIC_RELAXED,
+                              __HIP_MEMORY_SCOPE_AGENT),
+            (uint32_t)a.EC);
+    bool ran = false;
+    if ((uint32_t)m0 < rClip) {
+      if (threadIdx.x == 0) {
+        // wait for every up row-tile covering rows [m0, min(m0+BM, rClip))
+        const int rowHi =
+            ((int)rClip - 1 < m0 + BM - 1) ? (int)rClip - 1 : m0 + BM - 1;
+        for (int r = m0 / f.upBM; r <= rowHi / f.upBM; ++r) {
+          const uint32_t* ctr = f.upRowDone + (size_t)e * f.upMT + r;
+          long long spins = 0;
+          while (__hip_atomic_load(ctr, __ATOMIC_RELAXED,
+                                   __HIP_MEMORY_SCOPE_AGENT) <
+                 (uint32_t)f.upRowTarget) {
+            __builtin_amdgcn_s_sleep(8);
+            if (++spins > f.spinBound) {
+              if (f.errWord)
+                __hip_atomic_store(f.errWord, 3u, __ATOMIC_RELAXED,
+                                   __HIP_MEMORY_SCOPE_SYSTEM);
+              break;
+            }
+          }
+        }
+        __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+      }
+      __syncthreads();
+      ran = gemm_job_body<ET, 1, BN, BM, WET, STAGES>(a, smem, e, 0, m0,
+                                                      (rem / mT) * BN, 0,
+                                                      hasBias != 0);
+    }
+    if (ran) {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+    }
     if (threadIdx.x == 0)
       __hip_atomic_fetch_add(doneCtr, 1u, __ATOMIC_RELAXED,
                              __HIP_MEMORY_SCOPE_AGENT);
@@ -1740,8 +1809,8 @@ __global__ __launch_bounds__(512, 1) void k_moe_fused(GemmArgs gl, GemmArgs gu,
     const int rest = j / f.nTiles;
     const int ks = rest % f.lgKS;
     const int nt = rest / f.lgKS;
-    gemm_job_body<ET, 3, 128, 128, ET, 2>(gl, smem, 0, ks, t * 128,
-                                          nt * 128, 0, false);
+    (void)gemm_job_body<ET, 3, 128, 128, ET, 2>(gl, smem, 0, ks, t * 128,
+                                                nt * 128, 0, false);
     // arrive on the tile: the logits writes are fp32 atomicAdds
     // (globally coherent), so only a vmcnt drain orders them before
     // the arrival; no cache release is needed
@@ -1780,26 +1849,27 @@ __global__ __launch_bounds__(512, 1) void k_moe_fused(GemmArgs gl, GemmArgs gu,
     case 0:
       gemm_phase_walk<ET, 0, 256, 256, WET, 2>(
           gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, gu.act,
-          gu.bias != nullptr, &f.ctl->upJobsDone);
+          gu.bias != nullptr, &f.ctl->upJobsDone, f.upRowDone);
       break;
     case 1:
       gemm_phase_walk<ET, 0, 128, 256, WET, 3>(
           gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, gu.act,
-          gu.bias != nullptr, &f.ctl->upJobsDone);
+          gu.bias != nullptr, &f.ctl->upJobsDone, f.upRowDone);
       break;
     case 3:
       gemm_phase_walk<ET, 0, 256, 128, WET, 3>(
           gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, gu.act,
-          gu.bias != nullptr, &f.ctl->upJobsDone);
+          gu.bias != nullptr, &f.ctl->upJobsDone, f.upRowDone);
       break;
     default:
       gemm_phase_walk<ET, 0, 128, 128, WET, 2>(
           gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, gu.act,
-          gu.bias != nullptr, &f.ctl->upJobsDone);
+          gu.bias != nullptr, &f.ctl->upJobsDone, f.upRowDone);
       break;
   }
-  fused_wait(&f.ctl->upJobsDone, (uint32_t)f.upJobs, f.errWord,
-             f.spinBound);
+  // no global up->down seam: each down job waits only for the up
+  // row-tiles covering its A rows (gemm_phase_walk_dn); clk[2] marks
+  // this block's transition into the down walk
   if (tid == 0)
     atomicMax(reinterpret_cast<unsigned long long*>(&f.ctl->clk[2]),
               __builtin_amdgcn_s_memrealtime());
@@ -1807,23 +1877,23 @@ __global__ __launch_bounds__(512, 1) void k_moe_fused(GemmArgs gl, GemmArgs gu,
   // ---- phase D: expert down-GEMM with combine-slot epilogue ----
   switch (f.dnGeom) {
     case 0:
-      gemm_phase_walk<ET, 1, 256, 256, WET, 2>(
-          gd, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks, 0,
+      gemm_phase_walk_dn<ET, 256, 256, WET, 2>(
+          gd, f, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks,
           gd.bias != nullptr, &f.ctl->dnJobsDone);
       break;
     case 1:
-      gemm_phase_walk<ET, 1, 128, 256, WET, 3>(
-          gd, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks, 0,
+      gemm_phase_walk_dn<ET, 128, 256, WET, 3>(
+          gd, f, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks,
           gd.bias != nullptr, &f.ctl->dnJobsDone);
       break;
     case 3:
-      gemm_phase_walk<ET, 1, 256, 128, WET, 3>(
-          gd, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks, 0,
+      gemm_phase_walk_dn<ET, 256, 128, WET, 3>(
+          gd, f, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks,
           gd.bias != nullptr, &f.ctl->dnJobsDone);
       break;
     default:
-      gemm_phase_walk<ET, 1, 128, 128, WET, 2>(
-          gd, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks, 0,
+      gemm_phase_walk_dn<ET, 128, 128, WET, 2>(
+          gd, f, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks,
           gd.bias != nullptr, &f.ctl->dnJobsDone);
       break;
   }
@@ -2230,6 +2300,10 @@ static int moe_forward_fused(hipStream_t st, const void* x, const void* gate_w,
 
   f.ctl = g.fusedCtl;
   f.tileArrive = reinterpret_cast<uint32_t*>(g.fusedCtl + 1);
+  f.upRowDone = f.tileArrive + nTiles;
+  fusedGeoDims(up.geom, bm, bn, stg);
+  f.upBM = bm;
+  f.upRowTarget = up.nt;
   f.eCf = g.eC;
   f.logits32 = g.logits32;
   f.gate_out = gate_out;
@@ -2359,7 +2433,10 @@ int fm_initialize(const fm_config* cfg, int rank, int world_size) {
     // counters (zeroed per forward by a memset node) + host-mapped
     // give-up flag
     const int nTiles = g.S / 128;
-    g.fusedCtlBytes = sizeof(FusedCtl) + (size_t)nTiles * sizeof(uint32_t);
+    // + per-tile logits arrivals + per-(expert, row-tile) up completion
+    // counters (worst case BM=128 row tiles)
+    g.fusedCtlBytes = sizeof(FusedCtl) + (size_t)nTiles * sizeof(uint32_t) +
+                      (size_t)g.E * (g.pEC / 128) * sizeof(uint32_t);
     FM_HIP_CHECK(hipMalloc(&g.fusedCtl, g.fusedCtlBytes));
     FM_HIP_CHECK(hipMemset(g.fusedCtl, 0, g.fusedCtlBytes));
     FM_HIP_CHECK(hipHostMalloc(&g.hFusedErr, sizeof(uint32_t),
