@@ -681,7 +681,8 @@ constexpr int gemm_lds_bytes() {
 // runtime here; the classic kernels pass template constants that fold
 // after inlining. Layout/state contracts documented at the kernel below.
 // ---------------------------------------------------------------------------
-template <typename ET, int PHASE, int BN, int BM, typename WET, int STAGES>
+template <typename ET, int PHASE, int BN, int BM, typename WET, int STAGES,
+          int CK = 0, int CN = 0>
 __device__ __forceinline__ bool gemm_job_body(
     const GemmArgs& a, char* smemBase, int e, int ksplit, int m0, int n0,
     int act, bool hasBias) {
@@ -705,7 +706,13 @@ __device__ __forceinline__ bool gemm_job_body(
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  const int K = a.K, N = a.N;
+  // CK/CN: compile-time K/N for the BASELINE shape dictionary (the
+  // reference's own architecture freezes these in flashmoe_config.json
+  // as compile-time constants); 0 = runtime (generic path). Constant
+  // shapes fold the staging address math and fully determine the
+  // K-tile trip count (the probe ladder's "template generality" gap).
+  const int K = CK ? CK : a.K;
+  const int N = CN ? CN : a.N;
   const int sk = a.splitK > 0 ? a.splitK : 1;
   const int kLen = K / sk;            // this split's K range
   const int kStart = ksplit * kLen;
@@ -738,7 +745,7 @@ __device__ __forceinline__ bool gemm_job_body(
   // fp8 swizzle is on the 16B chunk16 index: ^ ((row >> 2) & 3)
   const int bgrow = lane / BCH;
   const int bsc = (lane % BCH) ^ (BEZ == 1 ? ((bgrow >> 2) & 3) : bgrow);
-  const int aRowStride = (PHASE == 0) ? a.H : K;
+  const int aRowStride = K;  // == a.H (up, logits) / a.K (down) by layout
   const size_t aBase = (size_t)e * a.strideAExpert;  // 0 for the x-gather up phase
   const ET* aSrc[GPW_A];
   const WET* bSrc[GPW_B];
@@ -1699,7 +1706,8 @@ __device__ __forceinline__ void combine_tokens(const FusedMeta& f, int t0,
 
 // static XCD-swizzled walk over one GEMM phase's (e, mT, nT) jobs
 // (identical swizzle to the classic persistent grid)
-template <typename ET, int PHASE, int BN, int BM, typename WET, int STAGES>
+template <typename ET, int PHASE, int BN, int BM, typename WET, int STAGES,
+          int CK = 0, int CN = 0>
 __device__ __forceinline__ void gemm_phase_walk(const GemmArgs& a, char* smem,
                                                 int mT, int nT, int nJobs,
                                                 int nBlocks, int act,
@@ -1713,7 +1721,7 @@ __device__ __forceinline__ void gemm_phase_walk(const GemmArgs& a, char* smem,
         (xcd < rx ? xcd * (qx + 1) : rx * (qx + 1) + (xcd - rx) * qx) + pos;
     const int e = swz / (mT * nT);
     const int rem = swz % (mT * nT);
-    const bool ran = gemm_job_body<ET, PHASE, BN, BM, WET, STAGES>(
+    const bool ran = gemm_job_body<ET, PHASE, BN, BM, WET, STAGES, CK, CN>(
         a, smem, e, 0, (rem % mT) * BM, (rem / mT) * BN, act, hasBias != 0);
     // job-count arrival: the epilogue's sc1 write-through stores need
     // only a per-wave drain before the relaxed arrival (Guideline 16
@@ -1736,7 +1744,8 @@ __device__ __forceinline__ void gemm_phase_walk(const GemmArgs& a, char* smem,
 // down-phase walk: each job first waits (bounded) for the up row-tiles
 // covering its A rows, takes ONE agent acquire, then runs - the down
 // phase overlaps the up phase's tail instead of waiting for all of it
-template <typename ET, int BN, int BM, typename WET, int STAGES>
+template <typename ET, int BN, int BM, typename WET, int STAGES,
+          int CK = 0, int CN = 0>
 __device__ __forceinline__ void gemm_phase_walk_dn(
     const GemmArgs& a, const FusedMeta& f, char* smem, int mT, int nT,
     int nJobs, int nBlocks, int hasBias, uint32_t* doneCtr) {
@@ -1777,9 +1786,8 @@ __device__ __forceinline__ void gemm_phase_walk_dn(
         __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
       }
       __syncthreads();
-      ran = gemm_job_body<ET, 1, BN, BM, WET, STAGES>(a, smem, e, 0, m0,
-                                                      (rem / mT) * BN, 0,
-                                                      hasBias != 0);
+      ran = gemm_job_body<ET, 1, BN, BM, WET, STAGES, CK, CN>(
+          a, smem, e, 0, m0, (rem / mT) * BN, 0, hasBias != 0);
     }
     if (ran) {
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -1791,7 +1799,7 @@ __device__ __forceinline__ void gemm_phase_walk_dn(
   }
 }
 
-template <typename ET, typename WET>
+template <typename ET, typename WET, int CH = 0, int CP = 0>
 __global__ __launch_bounds__(512, 1) void k_moe_fused(GemmArgs gl, GemmArgs gu,
                                                       GemmArgs gd,
                                                       FusedMeta f) {
@@ -1809,8 +1817,9 @@ __global__ __launch_bounds__(512, 1) void k_moe_fused(GemmArgs gl, GemmArgs gu,
     const int rest = j / f.nTiles;
     const int ks = rest % f.lgKS;
     const int nt = rest / f.lgKS;
-    (void)gemm_job_body<ET, 3, 128, 128, ET, 2>(gl, smem, 0, ks, t * 128,
-                                                nt * 128, 0, false);
+    (void)gemm_job_body<ET, 3, 128, 128, ET, 2, CH, 0>(gl, smem, 0, ks,
+                                                        t * 128, nt * 128, 0,
+                                                        false);
     // arrive on the tile: the logits writes are fp32 atomicAdds
     // (globally coherent), so only a vmcnt drain orders them before
     // the arrival; no cache release is needed
@@ -1847,22 +1856,22 @@ __global__ __launch_bounds__(512, 1) void k_moe_fused(GemmArgs gl, GemmArgs gu,
   // ---- phase U: expert up-GEMM ----
   switch (f.upGeom) {
     case 0:
-      gemm_phase_walk<ET, 0, 256, 256, WET, 2>(
+      gemm_phase_walk<ET, 0, 256, 256, WET, 2, CH, CP>(
           gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, gu.act,
           gu.bias != nullptr, &f.ctl->upJobsDone, f.upRowDone);
       break;
     case 1:
-      gemm_phase_walk<ET, 0, 128, 256, WET, 3>(
+      gemm_phase_walk<ET, 0, 128, 256, WET, 3, CH, CP>(
           gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, gu.act,
           gu.bias != nullptr, &f.ctl->upJobsDone, f.upRowDone);
       break;
     case 3:
-      gemm_phase_walk<ET, 0, 256, 128, WET, 3>(
+      gemm_phase_walk<ET, 0, 256, 128, WET, 3, CH, CP>(
           gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, gu.act,
           gu.bias != nullptr, &f.ctl->upJobsDone, f.upRowDone);
       break;
     default:
-      gemm_phase_walk<ET, 0, 128, 128, WET, 2>(
+      gemm_phase_walk<ET, 0, 128, 128, WET, 2, CH, CP>(
           gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, gu.act,
           gu.bias != nullptr, &f.ctl->upJobsDone, f.upRowDone);
       break;
@@ -1877,22 +1886,22 @@ __global__ __launch_bounds__(512, 1) void k_moe_fused(GemmArgs gl, GemmArgs gu,
   // ---- phase D: expert down-GEMM with combine-slot epilogue ----
   switch (f.dnGeom) {
     case 0:
-      gemm_phase_walk_dn<ET, 256, 256, WET, 2>(
+      gemm_phase_walk_dn<ET, 256, 256, WET, 2, CP, CH>(
           gd, f, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks,
           gd.bias != nullptr, &f.ctl->dnJobsDone);
       break;
     case 1:
-      gemm_phase_walk_dn<ET, 128, 256, WET, 3>(
+      gemm_phase_walk_dn<ET, 128, 256, WET, 3, CP, CH>(
           gd, f, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks,
           gd.bias != nullptr, &f.ctl->dnJobsDone);
       break;
     case 3:
-      gemm_phase_walk_dn<ET, 256, 128, WET, 3>(
+      gemm_phase_walk_dn<ET, 256, 128, WET, 3, CP, CH>(
           gd, f, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks,
           gd.bias != nullptr, &f.ctl->dnJobsDone);
       break;
     default:
-      gemm_phase_walk_dn<ET, 128, 128, WET, 2>(
+      gemm_phase_walk_dn<ET, 128, 128, WET, 2, CP, CH>(
           gd, f, smem, f.dnMT, f.dnNT, f.dnJobs, f.nBlocks,
           gd.bias != nullptr, &f.ctl->dnJobsDone);
       break;
@@ -2161,12 +2170,12 @@ static FusedGeo fusedPickGeo(int M, int N, int E, int nBlocks,
 // co-resident blocks/CU for the fused kernel at a given arena size
 // (0 = cannot run). The grid is sized to exactly occ x nCU so every
 // block is resident and the in-kernel fan-in waits cannot deadlock.
-template <typename ET, typename WET>
+template <typename ET, typename WET, int CH, int CP>
 static int fusedOccT(int arena) {
   static int maxDyn = -1;
   if (maxDyn < 0) {
     maxDyn = (hipFuncSetAttribute(
-                  reinterpret_cast<const void*>(&k_moe_fused<ET, WET>),
+                  reinterpret_cast<const void*>(&k_moe_fused<ET, WET, CH, CP>),
                   hipFuncAttributeMaxDynamicSharedMemorySize,
                   160 * 1024) == hipSuccess)
                  ? 160 * 1024 : 64 * 1024;
@@ -2176,27 +2185,48 @@ static int fusedOccT(int arena) {
   if (occArena != arena) {
     occVal = 0;
     if (hipOccupancyMaxActiveBlocksPerMultiprocessor(
-            &occVal, k_moe_fused<ET, WET>, 512, arena) != hipSuccess)
+            &occVal, k_moe_fused<ET, WET, CH, CP>, 512, arena) != hipSuccess)
       occVal = 0;
     occArena = arena;
   }
   return occVal;
 }
 
+template <typename ET, typename WET>
+static int launchFusedT(hipStream_t st, const GemmArgs& gl,
+                        const GemmArgs& gu, const GemmArgs& gd,
+                        const FusedMeta& f, int arena, int nBlocks) {
+  // BASELINE shape dictionary: compile-time (H, P) fold the tile loop's
+  // address math and trip counts (the reference freezes these in its
+  // compile-time config contract too); generic fallback otherwise
+#define FM_LAUNCH_SHAPE(HH, PP)                                                 do {                                                                            hipLaunchKernelGGL((k_moe_fused<ET, WET, HH, PP>), dim3(nBlocks),                                dim3(512), arena, st, gl, gu, gd, f);                      FM_HIP_CHECK(hipGetLastError());                                              return FM_OK;                                                               } while (0)
+  if (g.H == 1024 && g.P == 4096) FM_LAUNCH_SHAPE(1024, 4096);
+  if (g.H == 2048 && g.P == 8192) FM_LAUNCH_SHAPE(2048, 8192);
+  if (g.H == 4096 && g.P == 14336) FM_LAUNCH_SHAPE(4096, 14336);
+  FM_LAUNCH_SHAPE(0, 0);
+#undef FM_LAUNCH_SHAPE
+}
+
+template <typename ET, typename WET>
+static int fusedOccShape(int arena) {
+  if (g.H == 1024 && g.P == 4096) return fusedOccT<ET, WET, 1024, 4096>(arena);
+  if (g.H == 2048 && g.P == 8192) return fusedOccT<ET, WET, 2048, 8192>(arena);
+  if (g.H == 4096 && g.P == 14336)
+    return fusedOccT<ET, WET, 4096, 14336>(arena);
+  return fusedOccT<ET, WET, 0, 0>(arena);
+}
+
 static int fusedOcc(int arena) {
-  if (g.cfg.dtype == 3) return fusedOccT<fp16, fp16>(arena);
-  if (g.cfg.dtype == 4) return fusedOccT<bf16, fp8e4m3>(arena);
-  return fusedOccT<bf16, bf16>(arena);
+  if (g.cfg.dtype == 3) return fusedOccShape<fp16, fp16>(arena);
+  if (g.cfg.dtype == 4) return fusedOccShape<bf16, fp8e4m3>(arena);
+  return fusedOccShape<bf16, bf16>(arena);
 }
 
 template <typename ET, typename WET>
 static int launchFused(hipStream_t st, const GemmArgs& gl,
                        const GemmArgs& gu, const GemmArgs& gd,
                        const FusedMeta& f, int arena, int nBlocks) {
-  hipLaunchKernelGGL((k_moe_fused<ET, WET>), dim3(nBlocks), dim3(512), arena,
-                     st, gl, gu, gd, f);
-  FM_HIP_CHECK(hipGetLastError());
-  return FM_OK;
+  return launchFusedT<ET, WET>(st, gl, gu, gd, f, arena, nBlocks);
 }
 
 static int moe_forward_fused(hipStream_t st, const void* x, const void* gate_w,
