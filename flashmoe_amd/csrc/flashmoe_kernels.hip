@@ -1497,7 +1497,8 @@ __device__ __forceinline__ void fused_release_arrive(uint32_t* ctr) {
 __device__ __forceinline__ void fused_wait(const uint32_t* ctr,
                                            uint32_t target,
                                            uint32_t* errWord,
-                                           long long bound) {
+                                           long long bound,
+                                           bool noAcq = false) {
   if (threadIdx.x == 0) {
     long long spins = 0;
     while (__hip_atomic_load(ctr, __ATOMIC_RELAXED,
@@ -1511,7 +1512,7 @@ __device__ __forceinline__ void fused_wait(const uint32_t* ctr,
         break;
       }
     }
-    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+    if (!noAcq) __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
   }
   __syncthreads();
 }
@@ -1537,6 +1538,11 @@ struct FusedMeta {
   int upGeom, upMT, upNT, upJobs;
   int dnGeom, dnMT, dnNT, dnJobs;
   int cbTB, cbJobs;          // combine token-block size / jobs
+  int debugBits;             // FM_FUSED_DEBUG bit0: skip per-job drains
+                             // (publish once at walk end), bit1: skip
+                             // acquire fences. MEASUREMENT ONLY - both
+                             // break the visibility protocol; never set
+                             // outside an A/B pricing run.
   int nBlocks;
   int ctlOff;                // LDS offset of the 16-B control slice
   long long spinBound;
@@ -1713,7 +1719,8 @@ __device__ __forceinline__ void gemm_phase_walk(const GemmArgs& a, char* smem,
                                                 int nBlocks, int act,
                                                 int hasBias,
                                                 uint32_t* doneCtr,
-                                                uint32_t* rowDone) {
+                                                uint32_t* rowDone,
+                                                int debugBits) {
   const int qx = nJobs / 8, rx = nJobs % 8;
   for (int jl = blockIdx.x; jl < nJobs; jl += nBlocks) {
     const int xcd = jl % 8, pos = jl / 8;
@@ -1727,7 +1734,7 @@ __device__ __forceinline__ void gemm_phase_walk(const GemmArgs& a, char* smem,
     // only a per-wave drain before the relaxed arrival (Guideline 16
     // R1) - no cache-flushing release fence per block. Row arrivals
     // (rowDone) progressively unlock the down phase's dependent tiles.
-    if (ran) {
+    if (ran && !(debugBits & 1)) {
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       __syncthreads();
     }
@@ -1783,13 +1790,14 @@ __device__ __forceinline__ void gemm_phase_walk_dn(
             }
           }
         }
-        __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+        if (!(f.debugBits & 2))
+          __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
       }
       __syncthreads();
       ran = gemm_job_body<ET, 1, BN, BM, WET, STAGES, CK, CN>(
           a, smem, e, 0, m0, (rem / mT) * BN, 0, hasBias != 0);
     }
-    if (ran) {
+    if (ran && !(f.debugBits & 1)) {
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       __syncthreads();
     }
@@ -1848,7 +1856,7 @@ __global__ __launch_bounds__(512, 1) void k_moe_fused(GemmArgs gl, GemmArgs gu,
     }
   }
   fused_wait(&f.ctl->routeTilesDone, (uint32_t)f.nTiles, f.errWord,
-             f.spinBound);
+             f.spinBound, (f.debugBits & 2) != 0);
   if (tid == 0)
     atomicMax(reinterpret_cast<unsigned long long*>(&f.ctl->clk[1]),
               __builtin_amdgcn_s_memrealtime());
@@ -1858,22 +1866,26 @@ __global__ __launch_bounds__(512, 1) void k_moe_fused(GemmArgs gl, GemmArgs gu,
     case 0:
       gemm_phase_walk<ET, 0, 256, 256, WET, 2, CH, CP>(
           gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, gu.act,
-          gu.bias != nullptr, &f.ctl->upJobsDone, f.upRowDone);
+          gu.bias != nullptr, &f.ctl->upJobsDone, f.upRowDone,
+          f.debugBits);
       break;
     case 1:
       gemm_phase_walk<ET, 0, 128, 256, WET, 3, CH, CP>(
           gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, gu.act,
-          gu.bias != nullptr, &f.ctl->upJobsDone, f.upRowDone);
+          gu.bias != nullptr, &f.ctl->upJobsDone, f.upRowDone,
+          f.debugBits);
       break;
     case 3:
       gemm_phase_walk<ET, 0, 256, 128, WET, 3, CH, CP>(
           gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, gu.act,
-          gu.bias != nullptr, &f.ctl->upJobsDone, f.upRowDone);
+          gu.bias != nullptr, &f.ctl->upJobsDone, f.upRowDone,
+          f.debugBits);
       break;
     default:
       gemm_phase_walk<ET, 0, 128, 128, WET, 2, CH, CP>(
           gu, smem, f.upMT, f.upNT, f.upJobs, f.nBlocks, gu.act,
-          gu.bias != nullptr, &f.ctl->upJobsDone, f.upRowDone);
+          gu.bias != nullptr, &f.ctl->upJobsDone, f.upRowDone,
+          f.debugBits);
       break;
   }
   // no global up->down seam: each down job waits only for the up
@@ -1907,7 +1919,7 @@ __global__ __launch_bounds__(512, 1) void k_moe_fused(GemmArgs gl, GemmArgs gu,
       break;
   }
   fused_wait(&f.ctl->dnJobsDone, (uint32_t)f.dnJobs, f.errWord,
-             f.spinBound);
+             f.spinBound, (f.debugBits & 2) != 0);
   if (tid == 0)
     atomicMax(reinterpret_cast<unsigned long long*>(&f.ctl->clk[3]),
               __builtin_amdgcn_s_memrealtime());
@@ -2312,7 +2324,11 @@ static int moe_forward_fused(hipStream_t st, const void* x, const void* gate_w,
   gu.act = g.cfg.hidden_act;
   gu.H = g.H;
   gu.splitK = 1;
-  gu.sc1Out = 1;  // xM consumed by down-GEMM blocks in this launch
+  static const int dbg = [] {
+    const char* e = getenv("FM_FUSED_DEBUG");
+    return e ? atoi(e) : 0;
+  }();
+  gu.sc1Out = (dbg & 4) ? 0 : 1;  // xM consumed by down-GEMM blocks
 
   gd = gu;
   gd.A = g.xM;
@@ -2366,6 +2382,7 @@ static int moe_forward_fused(hipStream_t st, const void* x, const void* gate_w,
   f.dnJobs = dn.jobs;
   f.cbTB = 8;
   f.cbJobs = DIVUP(g.S, 8);
+  f.debugBits = dbg;
   f.nBlocks = nBlocks;
   f.ctlOff = arena - 16;
   f.spinBound = fusedSpinBound();
