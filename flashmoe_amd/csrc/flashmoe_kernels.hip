@@ -1503,7 +1503,7 @@ __device__ __forceinline__ void fused_wait(const uint32_t* ctr,
     long long spins = 0;
     while (__hip_atomic_load(ctr, __ATOMIC_RELAXED,
                              __HIP_MEMORY_SCOPE_AGENT) < target) {
-      __builtin_amdgcn_s_sleep(8);
+      __builtin_amdgcn_s_sleep(64);
       if (++spins > bound) {
         // give up: poisoned output, host sees the error word
         if (errWord)
@@ -1781,7 +1781,7 @@ __device__ __forceinline__ void gemm_phase_walk_dn(
           while (__hip_atomic_load(ctr, __ATOMIC_RELAXED,
                                    __HIP_MEMORY_SCOPE_AGENT) <
                  (uint32_t)f.upRowTarget) {
-            __builtin_amdgcn_s_sleep(8);
+            __builtin_amdgcn_s_sleep(32);
             if (++spins > f.spinBound) {
               if (f.errWord)
                 __hip_atomic_store(f.errWord, 3u, __ATOMIC_RELAXED,
