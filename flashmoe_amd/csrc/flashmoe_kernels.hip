@@ -2132,6 +2132,13 @@ static long long fusedSpinBound() {
   return b;
 }
 
+// ctl-embedded eC mirror (the fused kernel's routed counts; g.eC is
+// untouched by the fused path)
+static uint32_t* fusedECPtr() {
+  return reinterpret_cast<uint32_t*>(g.fusedCtl + 1) + (g.S / 128) +
+         (size_t)g.E * (g.pEC / 128);
+}
+
 static int fusedPoisonCheck() {
   if (g.hFusedErr &&
       *reinterpret_cast<volatile uint32_t*>(g.hFusedErr)) {
@@ -2311,7 +2318,7 @@ static int moe_forward_fused(hipStream_t st, const void* x, const void* gate_w,
   gu.out = g.xM;
   gu.gate_out = gate_out;
   gu.tokenIds = g.tokenIds;
-  gu.eC = g.eC;
+  gu.eC = nullptr;  // set to the ctl-embedded eC mirror below
   gu.strideAExpert = 0;
   gu.strideBExpert = 2LL * g.P * g.H;
   gu.strideOExpert = (long long)g.pEC * g.P;
@@ -2350,7 +2357,7 @@ static int moe_forward_fused(hipStream_t st, const void* x, const void* gate_w,
   fusedGeoDims(up.geom, bm, bn, stg);
   f.upBM = bm;
   f.upRowTarget = up.nt;
-  f.eCf = g.eC;
+  f.eCf = fusedECPtr();  // ctl-embedded (one memset covers it)
   f.logits32 = g.logits32;
   f.gate_out = gate_out;
   f.tokenIds = g.tokenIds;
@@ -2389,8 +2396,9 @@ static int moe_forward_fused(hipStream_t st, const void* x, const void* gate_w,
 
   // per-forward state re-init (Guideline 16: zero every polled word
   // ahead of the launch; these become memset nodes under graph capture)
+  gu.eC = f.eCf;
+  gd.eC = f.eCf;
   FM_HIP_CHECK(hipMemsetAsync(g.fusedCtl, 0, g.fusedCtlBytes, st));
-  FM_HIP_CHECK(hipMemsetAsync(g.eC, 0, (size_t)g.E * sizeof(uint32_t), st));
   if (g.cfg.is_training)
     FM_HIP_CHECK(hipMemsetAsync(g.gML, 0, 2 * (size_t)g.E * sizeof(float), st));
 
@@ -2481,9 +2489,12 @@ int fm_initialize(const fm_config* cfg, int rank, int world_size) {
     // give-up flag
     const int nTiles = g.S / 128;
     // + per-tile logits arrivals + per-(expert, row-tile) up completion
-    // counters (worst case BM=128 row tiles)
+    // counters (worst case BM=128 row tiles) + an embedded eC mirror so
+    // ONE memset node re-initializes every per-forward word (a second
+    // 32-B memset blit costs ~4 us of wall per step)
     g.fusedCtlBytes = sizeof(FusedCtl) + (size_t)nTiles * sizeof(uint32_t) +
-                      (size_t)g.E * (g.pEC / 128) * sizeof(uint32_t);
+                      (size_t)g.E * (g.pEC / 128) * sizeof(uint32_t) +
+                      (size_t)g.E * sizeof(uint32_t);
     FM_HIP_CHECK(hipMalloc(&g.fusedCtl, g.fusedCtlBytes));
     FM_HIP_CHECK(hipMemset(g.fusedCtl, 0, g.fusedCtlBytes));
     FM_HIP_CHECK(hipHostMalloc(&g.hFusedErr, sizeof(uint32_t),
@@ -2965,7 +2976,9 @@ int fm_read_routing(void* stream, uint32_t* routed_counts, uint32_t* token_idx,
   uint32_t* hEC = (uint32_t*)malloc((size_t)g.E * sizeof(uint32_t));
   TPS* hT = (TPS*)malloc((size_t)g.E * g.pEC * sizeof(TPS));
   if (!hEC || !hT) { free(hEC); free(hT); setErr("oom"); return FM_ERR_HIP; }
-  hipError_t e1 = hipMemcpyAsync(hEC, g.eC, (size_t)g.E * sizeof(uint32_t),
+  const uint32_t* eCsrc =
+      (g.lastForwardFused && g.fusedCtl) ? fusedECPtr() : g.eC;
+  hipError_t e1 = hipMemcpyAsync(hEC, eCsrc, (size_t)g.E * sizeof(uint32_t),
                                  hipMemcpyDeviceToHost, st);
   hipError_t e2 = hipMemcpyAsync(hT, g.tokenIds,
                                  (size_t)g.E * g.pEC * sizeof(TPS),
@@ -3352,7 +3365,8 @@ int fm_export_routing(void* stream, void* routed_dev, void* tps_dev) {
   if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
   hipStream_t st = reinterpret_cast<hipStream_t>(stream);
   hipLaunchKernelGGL(k_export_routing, dim3(g.E), dim3(256), 0, st,
-                     g.tokenIds, g.eC,
+                     g.tokenIds,
+                     (g.lastForwardFused && g.fusedCtl) ? fusedECPtr() : g.eC,
                      reinterpret_cast<uint32_t*>(routed_dev),
                      reinterpret_cast<uint32_t*>(tps_dev), g.E, g.EC, g.pEC);
   FM_HIP_CHECK(hipGetLastError());
