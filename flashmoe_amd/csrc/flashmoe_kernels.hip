@@ -2116,11 +2116,19 @@ int launch_gate(hipStream_t st, const void* x, const void* gate_w,
 static const int FM_FALLBACK = 1000;  // internal sentinel, never ABI-visible
 
 static bool fusedEnabled() {
-  static const bool en = [] {
+  // FM_FUSED=1 forces the single-launch kernel, FM_FUSED=0 forces the
+  // classic multi-kernel path. Default: fused where it measures faster
+  // (deep-K shapes where the classic down-GEMM cannot run its
+  // 2-blocks/CU 128^2 mode, P > 8192: cfg4-shape +3%); classic
+  // otherwise (cfg2/3/5 measured faster multi-kernel this round -
+  // profiles/r02; the fused kernel remains the FM_FUSED=1 path under
+  // active optimization).
+  static const int force = [] {
     const char* e = getenv("FM_FUSED");
-    return !(e && e[0] == '0');
+    return e ? (e[0] == '0' ? 0 : 1) : -1;
   }();
-  return en;
+  if (force >= 0) return force == 1;
+  return g.P > 8192;
 }
 
 static long long fusedSpinBound() {
