@@ -3117,12 +3117,10 @@ int fm_heap_ptrs(void** recv, void** ret) {
 // bounded-spin budget for the in-kernel flag waits; FM_P2P_SPIN_LOG2
 // shrinks it for the forced-timeout test
 static long long p2pSpinBound() {
-  static const long long b = [] {
-    const char* e = getenv("FM_P2P_SPIN_LOG2");
-    const int lg = e ? atoi(e) : 26;
-    return 1ll << (lg < 4 ? 4 : (lg > 40 ? 40 : lg));
-  }();
-  return b;
+  // re-read per call: the forced-timeout test shrinks it at runtime
+  const char* e = getenv("FM_P2P_SPIN_LOG2");
+  const int lg = e ? atoi(e) : 26;
+  return 1ll << (lg < 4 ? 4 : (lg > 40 ? 40 : lg));
 }
 
 // a previously-completed k_await_flags timeout poisons the transport:

@@ -172,8 +172,32 @@ def moe_forward_ep(input, gate_weights, expert_weights, group=None,
 
     from . import _ext, moe
 
-    if os.environ.get("FLASHMOE_P2P") == "1" and not _stub_exchange:
-        return moe_forward_ep_p2p(input, gate_weights, expert_weights, group)
+    # One-sided xGMI transport is the DEFAULT data plane (the
+    # reference's intra-node mode); FLASHMOE_P2P=0 forces the RCCL
+    # all_to_all pipeline, FLASHMOE_P2P=1 forces P2P with no fallback.
+    # In the default "auto" mode the first P2P step is validated with
+    # fm_p2p_error_check (one extra sync); any setup or timeout failure
+    # falls back to the all_to_all path for the rest of the run.
+    p2p_mode = os.environ.get("FLASHMOE_P2P", "auto")
+    if p2p_mode != "0" and not _stub_exchange and \
+            not moe._state.get("p2p_failed"):
+        if p2p_mode == "1":
+            return moe_forward_ep_p2p(input, gate_weights, expert_weights,
+                                      group)
+        try:
+            out = moe_forward_ep_p2p(input, gate_weights, expert_weights,
+                                     group)
+            if not moe._state.get("p2p_validated"):
+                lib = _ext.load()
+                stream = torch.cuda.current_stream().cuda_stream
+                _ext.check(lib.fm_p2p_error_check(ctypes.c_void_p(stream)),
+                           "fm_p2p_error_check")
+                moe._state["p2p_validated"] = True
+            return out
+        except Exception as exc:  # pragma: no cover - multi-GPU only
+            moe._state["p2p_failed"] = True
+            print(f"flashmoe: P2P transport failed ({exc}); "
+                  "falling back to RCCL all_to_all", flush=True)
     lib = _ext.load()
     world = dist.get_world_size(group)
     cc = moe.get_compiled_config()

@@ -157,7 +157,10 @@ def test_ep_pipeline_under_torchrun(dtype_code):
         f"--nproc-per-node={n}", "--master-addr", "127.0.0.1",
         "--master-port", str(29519 + dtype_code), worker,
     ]
-    env = dict(os.environ, PYTHONPATH=REPO_ROOT, FM_TEST_DTYPE=str(dtype_code))
+    # pin the RCCL all_to_all pipeline: the P2P transport (now the auto
+    # default) has its own world-2 test in test_gpu_p2p.py
+    env = dict(os.environ, PYTHONPATH=REPO_ROOT, FM_TEST_DTYPE=str(dtype_code),
+               FLASHMOE_P2P="0")
     r = subprocess.run(cmd, capture_output=True, text=True, timeout=600, env=env)
     sys.stdout.write(r.stdout[-2000:])
     sys.stderr.write(r.stderr[-2000:])
