@@ -1971,6 +1971,7 @@ struct State {
   size_t heapRecvOff = 0, heapRetOff = 0, heapDFlagOff = 0, heapRFlagOff = 0;
   size_t heapBytes = 0;
   void* peerBase[64] = {};   // opened peer heap bases (self = heap)
+  bool peerOpened[64] = {};  // true only for real hipIpcOpenMemHandle maps
   uint64_t* dPeerRecv = nullptr;   // device tables [world]
   uint64_t* dPeerRet = nullptr;
   uint64_t* dPeerDFlag = nullptr;
@@ -2536,7 +2537,10 @@ int fm_finalize(void) {
   if (g.hFusedErr) (void)hipHostFree(g.hFusedErr);
   if (g.heap) {
     for (int p = 0; p < g.world; ++p) {
-      if (p != g.rank && g.peerBase[p]) (void)hipIpcCloseMemHandle(g.peerBase[p]);
+      // close only real IPC mappings: a NULL-handle connect aliases
+      // peers to the local heap, and closing that poisons the sticky
+      // per-thread HIP error state for later launches
+      if (g.peerOpened[p]) (void)hipIpcCloseMemHandle(g.peerBase[p]);
     }
     (void)hipFree(g.heap); (void)hipFree(g.dPeerRecv); (void)hipFree(g.dArrive);
     if (g.hP2pErr) (void)hipHostFree(g.hP2pErr);
@@ -3089,6 +3093,7 @@ int fm_heap_connect(const void* handles /* world x 64B, null = local only */) {
              sizeof(h));
       FM_HIP_CHECK(hipIpcOpenMemHandle(&base, h,
                                        hipIpcMemLazyEnablePeerAccess));
+      g.peerOpened[p] = true;
     }
     g.peerBase[p] = base;
     hRecv[p] = (uint64_t)(uintptr_t)base + g.heapRecvOff;
