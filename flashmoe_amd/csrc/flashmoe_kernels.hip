@@ -1422,6 +1422,37 @@ __global__ void k_export_routing(const TPS* __restrict__ tokenIds,
   }
 }
 
+// MX-scaled MFMA layout probe (test-only): D = diag-scaled A[16x128] x
+// B[128x16] via one mfma_scale_f32_16x16x128_f8f6f4 (fp8 e4m3 A and B,
+// per-32-element-block E8M0 scales). Assumed layout (empirically
+// validated by test_mx_mfma_layout_probe): lane l holds
+// A[row=l&15][k=32*(l>>4)..+32] and B[k=32*(l>>4)..+32][col=l&15] as 8
+// dwords; scale operand = the lane's block scale byte in bits 0..7
+// (E8M0, value 2^(byte-127)); C/D mapping as every 16x16 shape.
+typedef __attribute__((ext_vector_type(8))) int i32x8;
+__global__ void k_mx_mfma_probe(const uint8_t* A, const uint8_t* B,
+                                const uint8_t* sa, const uint8_t* sb,
+                                float* D) {
+  const int lane = threadIdx.x & 63;
+  const int row = lane & 15;
+  const int kb = lane >> 4;  // 32-element K block
+  i32x8 af, bf;
+#pragma unroll
+  for (int d = 0; d < 8; ++d) {
+    af[d] = *reinterpret_cast<const int*>(&A[row * 128 + kb * 32 + d * 4]);
+    bf[d] = *reinterpret_cast<const int*>(&B[row * 128 + kb * 32 + d * 4]);
+    // B given transposed: B_t[col][k] with col=lane&15 - same indexing
+  }
+  f32x4 c{0.f, 0.f, 0.f, 0.f};
+  const int sav = sa[row * 4 + kb];
+  const int sbv = sb[row * 4 + kb];
+  c = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(af, bf, c, 0, 0,
+                                                       0, sav, 0, sbv);
+#pragma unroll
+  for (int r = 0; r < 4; ++r)
+    D[((lane >> 4) * 4 + r) * 16 + (lane & 15)] = c[r];
+}
+
 // MFMA layout probe (test-only): D = A[16x32] x B[32x16] via one
 // mfma_f32_16x16x32_bf16, written with the assumed C/D mapping.
 __global__ void k_fp8cvt_probe(const uint32_t* in, float* out) {
@@ -3502,6 +3533,19 @@ int fm_debug_taskq(void* stream, int ring_sz, long long n_tasks,
   if (out_consumed) *out_consumed = h[1];
   if (out_errors) *out_errors = h[2];
   (void)hipFree(ring); (void)hipFree(seq); (void)hipFree(ctrs); (void)hipFree(acc);
+  return FM_OK;
+}
+
+int fm_debug_mx_mfma(void* stream, const void* A, const void* B,
+                     const void* sa, const void* sb, void* D) {
+  hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+  hipLaunchKernelGGL(k_mx_mfma_probe, dim3(1), dim3(64), 0, st,
+                     reinterpret_cast<const uint8_t*>(A),
+                     reinterpret_cast<const uint8_t*>(B),
+                     reinterpret_cast<const uint8_t*>(sa),
+                     reinterpret_cast<const uint8_t*>(sb),
+                     reinterpret_cast<float*>(D));
+  FM_HIP_CHECK(hipGetLastError());
   return FM_OK;
 }
 
