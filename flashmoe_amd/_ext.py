@@ -89,7 +89,7 @@ def _bind(lib: ctypes.CDLL) -> ctypes.CDLL:
     lib.fm_built_for_gfx950.argtypes = []
     lib.fm_built_for_gfx950.restype = ctypes.c_int
     lib.fm_debug_mfma.argtypes = [p, p, p, p]
-    lib.fm_debug_mx_mfma.argtypes = [p, p, p, p, p, p]
+    lib.fm_debug_mx_mfma.argtypes = [p, p, p, p, p, p, ctypes.c_int]
     lib.fm_debug_mx_mfma.restype = ctypes.c_int
     lib.fm_debug_mfma.restype = ctypes.c_int
     return lib

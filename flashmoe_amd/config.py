@@ -20,27 +20,30 @@ REQUIRED_KEYS = [
 
 # torch_dtype enum (schema): 0 float, 1 tf32 (== fp32 on CDNA4: no xf32),
 # 2 bf16, 3 fp16, 4 fp8e4m3 expert weights with bf16 activations (the
-# config-5 "fp8 weights / bf16 accumulate" regime; extension code - the
-# reference schema stops at 3). torch_dtype_of() is the ACTIVATION dtype;
-# weight_dtype_of() the expert-weight storage dtype.
+# config-5 "fp8 weights / bf16 accumulate" regime), 5 MX fp8: fp8e4m3
+# weights AND runtime-quantized fp8 activations (per-64-element E8M0
+# block scales) on the CDNA4 scaled MFMA - the 2x-rate path (extension
+# codes; the reference schema stops at 3). torch_dtype_of() is the
+# ACTIVATION dtype at the API boundary; weight_dtype_of() the
+# expert-weight storage dtype.
 def torch_dtype_of(code: int):
     import torch
 
     return {0: torch.float32, 1: torch.float32, 2: torch.bfloat16,
-            3: torch.float16, 4: torch.bfloat16}[code]
+            3: torch.float16, 4: torch.bfloat16, 5: torch.bfloat16}[code]
 
 
 def weight_dtype_of(code: int):
     import torch
 
-    if code == 4:
+    if code in (4, 5):
         return torch.float8_e4m3fn
     return torch_dtype_of(code)
 
 
 def element_size_of(code: int) -> int:
     # activation element size (workspace sizing); dtype-4 weights are 1B
-    return {0: 4, 1: 4, 2: 2, 3: 2, 4: 2}[code]
+    return {0: 4, 1: 4, 2: 2, 3: 2, 4: 2, 5: 2}[code]
 
 
 def load_config(path: str | None = None) -> dict:

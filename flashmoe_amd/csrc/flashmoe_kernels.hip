@@ -42,6 +42,7 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(8))) _Float16 halfx8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 typedef __attribute__((ext_vector_type(4))) uint32_t u32x4;
+typedef __attribute__((ext_vector_type(8))) int i32x8;
 
 using fp16 = __half;
 struct fp8e4m3 { uint8_t v; };  // storage-only; dequant at fragment read
@@ -353,6 +354,63 @@ __global__ __launch_bounds__(256) void k_gate_route(
 }
 
 // ---------------------------------------------------------------------------
+// MX-fp8 activation quantization (dtype 5, the cfg5 regime: "CDNA4 fp8
+// MFMA + token-scaling"): bf16 rows -> fp8 e4m3 bytes + per-64-element
+// E8M0 block scales. Block size 64 matches the EMPIRICAL scale
+// granularity of v_mfma_scale_f32_16x16x128_f8f6f4 on gfx950 (probe
+// test_mx_mfma_layout_probe: the scale byte of lane-group g applies to
+// k in [64g, 64g+64); opsel selects the byte). Scale = 2^e with e
+// chosen so blockmax/2^e <= 448 (e4m3 max); rounding RNE via
+// v_cvt_pk_fp8_f32.
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ uint32_t cvt2_fp8(float a, float b) {
+  // pack two fp32 -> two e4m3 bytes (RNE), low half of a dword
+  f32x2n v;
+  v.x = a;
+  v.y = b;
+  return __builtin_amdgcn_cvt_pk_fp8_f32(v.x, v.y, 0u, false);
+}
+
+template <typename T>
+__global__ void k_quant_mx(const T* __restrict__ in, uint8_t* __restrict__ out,
+                           uint8_t* __restrict__ scales, long long nRows,
+                           int K) {
+  // one wave per 64-element block: lane l holds in[row][b*64 + l] ...
+  // simpler: each thread walks one block serially is too slow; use
+  // 64 lanes x 1 element? Layout: block j of row r -> 64 values.
+  // Grid-stride over blocks; blockDim 256 = 4 waves, each wave one
+  // 64-block per iteration (lane = element).
+  const int wave = (int)(threadIdx.x >> 6);
+  const int lane = (int)(threadIdx.x & 63);
+  const long long nBlk = nRows * (K / 64);
+  for (long long b = (long long)blockIdx.x * 4 + wave; b < nBlk;
+       b += (long long)gridDim.x * 4) {
+    const long long row = b / (K / 64);
+    const int j = (int)(b % (K / 64));
+    const float v = toF(in[row * K + j * 64 + lane]);
+    // wave max |v|
+    float m = fabsf(v);
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+      m = fmaxf(m, __shfl_xor(m, off, 64));
+    // E8M0 exponent e: smallest power of two with blockmax/2^e <= 448
+    int e = 0;
+    if (m > 0.0f) {
+      int ex;
+      (void)frexpf(m / 448.0f, &ex);  // m/448 = f * 2^ex, f in [0.5, 1)
+      e = ex;                          // 2^ex >= m/448 > 2^(ex-1)
+      if (e < -126) e = -126;
+      if (e > 127) e = 127;
+    }
+    const float inv = exp2f((float)-e);
+    const uint32_t pk = cvt2_fp8(v * inv, 0.0f);
+    out[row * K + j * 64 + lane] = (uint8_t)(pk & 0xFF);
+    if (lane == 0) scales[row * (K / 64) + j] = (uint8_t)(127 + e);
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Grouped expert GEMM (reference: os/processor/processor.cuh fGET preGEMM
 // :339-468 / postGEMM :711-751, gemm.cuh FAA epilogue, combine :44-205).
 // PHASE 0 (up):   A = x rows gathered via tokenIds[e]; B = Wup[e] [P,H];
@@ -412,6 +470,9 @@ struct GemmArgs {
                              // consumers on other CUs see them after a
                              // vmcnt drain + job-count arrival, with no
                              // per-block release fence
+  const void* aScales;       // MX path (dtype 5): per-64-element E8M0
+                             // block scales of the quantized A operand,
+                             // laid out like A with K/64 bytes per row
 };
 
 // address-space helpers for global_load_lds (direct HBM->LDS DMA)
@@ -1017,6 +1078,266 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
 }
 
 // ---------------------------------------------------------------------------
+// MX-fp8 grouped GEMM (dtype 5): fp8 e4m3 A (runtime-quantized
+// activations, per-64-element E8M0 scales) x fp8 e4m3 B (expert weights,
+// scale 1) on v_mfma_scale_f32_16x16x128_f8f6f4 - the 2x-rate CDNA4
+// path (~4.7 PF/s vs 2.5 bf16). 128x128 tile, BK=128 (one scaled MFMA
+// covers the whole K-tile), 512 threads = 8 waves (2M x 4N), fp32
+// accumulate, bf16 outputs. Staging geometry is byte-identical to the
+// bf16 kernel's (128-B rows, XOR-16B-chunk swizzle, glds double
+// buffer); A/B bytes per K-element halve, so HBM/LDS traffic halves
+// too. Scale->lane mapping empirically pinned by
+// test_mx_mfma_layout_probe: lane-group g's scale byte covers
+// k in [64g, 64g+64) (opsel selects the byte; groups 2..3 unused).
+// PHASE semantics match gemm_job_body (0 up / 1 down+slots / 2 packed).
+// ---------------------------------------------------------------------------
+
+template <int PHASE>
+__device__ __forceinline__ bool mx_gemm_job_body(
+    const GemmArgs& a, char* smemBase, int e, int m0, int n0, int act,
+    bool hasBias) {
+  constexpr int BM = 128, BN = 128, BK = 128;
+  constexpr int SE = 2;
+  uint8_t* Abase = reinterpret_cast<uint8_t*>(smemBase);   // SE x [BM][BK]
+  uint8_t* Bbase = reinterpret_cast<uint8_t*>(smemBase + SE * BM * BK);
+  TPS* sTps = reinterpret_cast<TPS*>(smemBase + SE * BM * BK + SE * BN * BK);
+  uint32_t* sRouted = reinterpret_cast<uint32_t*>(sTps + BM);
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int K = a.K, N = a.N;
+
+  const TPS* tpsE = a.tokenIds ? a.tokenIds + (size_t)e * a.pEC : nullptr;
+  __syncthreads();  // prior job's LDS readers done
+  if (tid == 0)
+    *sRouted = tpsE ? min(a.eC[e], (uint32_t)a.EC) : (uint32_t)a.nRows;
+  __syncthreads();
+  const uint32_t routed = *sRouted;
+  if ((uint32_t)m0 >= routed) return false;
+  const int mCap = a.tokenIds ? a.pEC : a.nRows;
+  const int we = a.segExpert ? a.segExpert[e] : e;
+  if (tid < BM) {
+    TPS t{0u, 1.0f};
+    if ((uint32_t)(m0 + tid) < routed)
+      t = tpsE ? tpsE[m0 + tid] : TPS{(uint32_t)(m0 + tid), 1.0f};
+    sTps[tid] = t;
+  }
+  __syncthreads();
+
+  const uint8_t* __restrict__ Ag = reinterpret_cast<const uint8_t*>(a.A);
+  const uint8_t* __restrict__ Bg =
+      reinterpret_cast<const uint8_t*>(a.B) + (size_t)we * a.strideBExpert;
+  const uint8_t* __restrict__ aS =
+      reinterpret_cast<const uint8_t*>(a.aScales);
+
+  // glds staging: 16 KiB per operand tile = 16 x 1 KiB groups (8 rows of
+  // 128 B each); lane -> (row-in-group = lane/8, 16B chunk = lane%8),
+  // chunk XOR-swizzled by row for conflict-free b128 fragment reads
+  const int grow8 = lane >> 3;
+  const int schunk = (lane & 7) ^ grow8;
+  const size_t aBase = (size_t)e * a.strideAExpert;  // bytes (u8 A)
+  const uint8_t* aSrc[2];
+  const uint8_t* bSrc[2];
+  uint32_t aRowIdx[2];  // scale row of each staged A row (this wave's)
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int row = (wave * 2 + i) * 8 + grow8;
+    const size_t arow = (PHASE == 0) ? (size_t)tpsTok(sTps[row].tokenIdx)
+                                     : (size_t)min(m0 + row, mCap - 1);
+    aSrc[i] = Ag + aBase + arow * (size_t)K + schunk * 16;
+    aRowIdx[i] = (uint32_t)arow;
+  }
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int row = (wave * 2 + i) * 8 + grow8;
+    bSrc[i] = Bg + (size_t)min(n0 + row, N - 1) * K + schunk * 16;
+  }
+  auto stage = [&](int kt, int buf) {
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+      __builtin_amdgcn_global_load_lds(
+          (gas_u32*)(aSrc[i] + kt), (las_u32*)(Abase + buf * BM * BK +
+                                               (wave * 2 + i) * 1024), 16, 0, 0);
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+      __builtin_amdgcn_global_load_lds(
+          (gas_u32*)(bSrc[i] + kt), (las_u32*)(Bbase + buf * BN * BK +
+                                               (wave * 2 + i) * 1024), 16, 0, 0);
+  };
+
+  // wave grid 2(M) x 4(N): 64 rows x 32 cols per wave
+  const int wr = wave >> 2, wc = wave & 3;
+  constexpr int MI = 4, NF = 2;
+  f32x4 accv[MI][NF];
+#pragma unroll
+  for (int i = 0; i < MI; ++i)
+#pragma unroll
+    for (int j = 0; j < NF; ++j) accv[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  // per-lane fragment geometry: row rl, k-group g (32 elements each);
+  // the lane's SCALE byte covers k64 = g (only groups 0..1 are read by
+  // the instruction; 2..3 get a don't-care via g&1)
+  const int rl = lane & 15;
+  const int gk = lane >> 4;
+  const int K64 = K / 64;
+
+  const int nK = K / BK;
+  stage(0, 0);
+  wait_vmcnt<0>();
+  __builtin_amdgcn_s_barrier();
+  for (int t = 0; t < nK; ++t) {
+    const bool stageNow = t + 1 < nK;
+    if (stageNow) stage((t + 1) * BK, (t + 1) % SE);
+    const uint8_t* Al = Abase + (t % SE) * BM * BK;
+    const uint8_t* Bl = Bbase + (t % SE) * BN * BK;
+    __builtin_amdgcn_s_setprio(1);
+    // A-scale bytes: this lane's rows at k64 block (t*2 + (gk&1))
+    const int k64 = t * 2 + (gk & 1);
+    i32x8 af[MI];
+    int sav[MI];
+#pragma unroll
+    for (int mi = 0; mi < MI; ++mi) {
+      const int R = wr * 64 + mi * 16 + rl;
+      const int c0 = (2 * gk) ^ (R & 7);
+      const int c1 = (2 * gk + 1) ^ (R & 7);
+      const u32x4 lo = *reinterpret_cast<const u32x4*>(&Al[R * BK + c0 * 16]);
+      const u32x4 hi = *reinterpret_cast<const u32x4*>(&Al[R * BK + c1 * 16]);
+#pragma unroll
+      for (int d = 0; d < 4; ++d) {
+        af[mi][d] = (int)lo[d];
+        af[mi][4 + d] = (int)hi[d];
+      }
+      // scale row: the STAGED row R's source row (token-gathered for
+      // PHASE 0). Recompute like the staging pass did.
+      const size_t arow = (PHASE == 0)
+          ? (size_t)tpsTok(sTps[R].tokenIdx)
+          : (size_t)min(m0 + R, mCap - 1);
+      sav[mi] = aS[(aBase / 64) + arow * K64 + k64];
+    }
+    i32x8 bfr[NF];
+#pragma unroll
+    for (int ni = 0; ni < NF; ++ni) {
+      const int R = wc * 32 + ni * 16 + rl;
+      const int c0 = (2 * gk) ^ (R & 7);
+      const int c1 = (2 * gk + 1) ^ (R & 7);
+      const u32x4 lo = *reinterpret_cast<const u32x4*>(&Bl[R * BK + c0 * 16]);
+      const u32x4 hi = *reinterpret_cast<const u32x4*>(&Bl[R * BK + c1 * 16]);
+#pragma unroll
+      for (int d = 0; d < 4; ++d) {
+        bfr[ni][d] = (int)lo[d];
+        bfr[ni][4 + d] = (int)hi[d];
+      }
+    }
+#pragma unroll
+    for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < NF; ++ni)
+        accv[mi][ni] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+            af[mi], bfr[ni], accv[mi][ni], 0 /*fp8*/, 0 /*fp8*/, 0, sav[mi],
+            0, 0x7F /*B scale 1.0*/);
+    __builtin_amdgcn_s_setprio(0);
+    wait_vmcnt<0>();
+    __builtin_amdgcn_s_barrier();
+  }
+
+  // epilogue (bf16 outputs; mirrors gemm_job_body's PHASE semantics)
+  const int cl = lane & 15;
+  const int r0 = (lane >> 4) * 4;
+  float* sScale = reinterpret_cast<float*>(Abase);
+  const bool slot = (PHASE == 1) && (a.topk > 1 || a.slotAlways);
+  const bool scaled = (PHASE == 1) && a.topk > 1;
+  if constexpr (PHASE == 1) {
+    if (scaled) {
+      __syncthreads();
+      if (tid < BM) {
+        const TPS tp = sTps[tid];
+        float sc = 0.0f;
+        if ((uint32_t)(m0 + tid) < routed)
+          sc = toF(reinterpret_cast<const bf16*>(
+                   a.gate_out)[(size_t)tpsTok(tp.tokenIdx) * a.PX +
+                               a.expertOffset + e]) / tp.probSum;
+        sScale[tid] = sc;
+      }
+      __syncthreads();
+    }
+  }
+  float bv[NF] = {0.f, 0.f};
+  if (hasBias) {
+    const bf16* bptr = reinterpret_cast<const bf16*>(a.bias) +
+                       (a.strideBExpert ? (size_t)we * N : 0);
+#pragma unroll
+    for (int ni = 0; ni < NF; ++ni) {
+      const int col = n0 + wc * 32 + ni * 16 + cl;
+      if (col < N) bv[ni] = toF(bptr[col]);
+    }
+  }
+#pragma unroll
+  for (int mi = 0; mi < MI; ++mi) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = wr * 64 + mi * 16 + r0 + r;
+      const int m = m0 + row;
+      if ((uint32_t)m >= routed) continue;
+      const TPS tp = sTps[row];
+      const float rowScale = scaled ? sScale[row] : 1.0f;
+#pragma unroll
+      for (int ni = 0; ni < NF; ++ni) {
+        const int col = n0 + wc * 32 + ni * 16 + cl;
+        if (col >= N) continue;
+        float v = accv[mi][ni][r] + bv[ni];
+        if constexpr (PHASE == 0) {
+          v = (act == 0) ? fmaxf(v, 0.0f)
+                         : 0.5f * v * (1.0f + erff(v * 0.70710678118654752f));
+          reinterpret_cast<bf16*>(a.out)[(size_t)e * a.strideOExpert +
+                                         (size_t)m * N + col] =
+              __float2bfloat16(v);
+        } else if constexpr (PHASE == 1) {
+          if (slot) {
+            reinterpret_cast<bf16*>(a.O32)[
+                ((size_t)tpsTok(tp.tokenIdx) * a.topk + tpsJ(tp.tokenIdx)) *
+                    a.H + col] = __float2bfloat16(v * rowScale);
+          } else {
+            reinterpret_cast<bf16*>(
+                a.moe_out)[(size_t)tpsTok(tp.tokenIdx) * a.H + col] =
+                __float2bfloat16(v);
+          }
+        } else {
+          reinterpret_cast<bf16*>(a.out)[(size_t)e * a.strideOExpert +
+                                         (size_t)m * N + col] =
+              __float2bfloat16(v);
+        }
+      }
+    }
+  }
+  return true;
+}
+
+template <int PHASE, int ACT, bool HAS_BIAS>
+__global__ __launch_bounds__(512, 2) void k_group_gemm_mx(GemmArgs a) {
+  constexpr int BM = 128, BN = 128, BK = 128, SE = 2;
+  __shared__ __attribute__((aligned(16))) char smem[
+      SE * BM * BK + SE * BN * BK + BM * 8 + 16];
+  const int mT = a.totalJobs > 0 ? a.jobsMT : gridDim.x;
+  const int nT = a.totalJobs > 0 ? a.jobsNT : gridDim.y;
+  const int gstride = gridDim.x * gridDim.y * gridDim.z;
+  const int nBlocks = a.totalJobs > 0 ? a.totalJobs : gstride;
+  const int lin0 =
+      blockIdx.x + gridDim.x * (blockIdx.y + gridDim.y * blockIdx.z);
+  const int qx = nBlocks / 8, rx = nBlocks % 8;
+  for (int jl = lin0; jl < nBlocks; jl += gstride) {
+    const int xcd = jl % 8, pos = jl / 8;
+    const int swz = a.noRemap
+        ? jl
+        : (xcd < rx ? xcd * (qx + 1) : rx * (qx + 1) + (xcd - rx) * qx) + pos;
+    const int e = swz / (mT * nT);
+    const int rem = swz % (mT * nT);
+    (void)mx_gemm_job_body<PHASE>(a, smem, e, (rem % mT) * BM,
+                                  (rem / mT) * BN, ACT, HAS_BIAS);
+  }
+}
+
+// ---------------------------------------------------------------------------
 // fp32 grouped GEMM (config-1 "CPU correctness plumbing" shapes; VALU
 // tiled). 64x64 tile, 256 threads, each thread a 4x4 sub-block, BK=16.
 // Same phase semantics as the bf16 kernel.
@@ -1429,13 +1750,13 @@ __global__ void k_export_routing(const TPS* __restrict__ tokenIds,
 // A[row=l&15][k=32*(l>>4)..+32] and B[k=32*(l>>4)..+32][col=l&15] as 8
 // dwords; scale operand = the lane's block scale byte in bits 0..7
 // (E8M0, value 2^(byte-127)); C/D mapping as every 16x16 shape.
-typedef __attribute__((ext_vector_type(8))) int i32x8;
+template <int OPS>
 __global__ void k_mx_mfma_probe(const uint8_t* A, const uint8_t* B,
                                 const uint8_t* sa, const uint8_t* sb,
                                 float* D) {
   const int lane = threadIdx.x & 63;
   const int row = lane & 15;
-  const int kb = lane >> 4;  // 32-element K block
+  const int kb = lane >> 4;  // 32-element K block (assumed data layout)
   i32x8 af, bf;
 #pragma unroll
   for (int d = 0; d < 8; ++d) {
@@ -1444,10 +1765,12 @@ __global__ void k_mx_mfma_probe(const uint8_t* A, const uint8_t* B,
     // B given transposed: B_t[col][k] with col=lane&15 - same indexing
   }
   f32x4 c{0.f, 0.f, 0.f, 0.f};
-  const int sav = sa[row * 4 + kb];
-  const int sbv = sb[row * 4 + kb];
+  // sa/sb are PER-LANE raw scale i32s (64 each): the GPU layout test
+  // sweeps single lanes/bytes to pin the hardware's scale->block map
+  const int sav = reinterpret_cast<const int*>(sa)[lane];
+  const int sbv = reinterpret_cast<const int*>(sb)[lane];
   c = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(af, bf, c, 0, 0,
-                                                       0, sav, 0, sbv);
+                                                       OPS, sav, OPS, sbv);
 #pragma unroll
   for (int r = 0; r < 4; ++r)
     D[((lane >> 4) * 4 + r) * 16 + (lane & 15)] = c[r];
@@ -2017,6 +2340,10 @@ struct State {
   void* lastKey[8] = {};
   bool graphValid = false;
   void* xM = nullptr;        // [nLx_alloc, pEC, P] Element
+  uint8_t* x8 = nullptr;     // dtype 5: quantized activations [S, H]
+  uint8_t* xs = nullptr;     // dtype 5: x block scales [S, H/64]
+  uint8_t* xM8 = nullptr;    // dtype 5: quantized intermediates
+  uint8_t* xMs = nullptr;    // dtype 5: xM block scales
   float* O32 = nullptr;      // [S, H] (staged-API combine accumulator)
   void* cbuf = nullptr;      // [S, k, H] Element: non-atomic combine slots
   uint8_t* kept = nullptr;   // [S, k] capacity-kept mask (gate-written)
@@ -2480,10 +2807,16 @@ int fm_initialize(const fm_config* cfg, int rank, int world_size) {
     case 0: case 1: g.esz = 4; break;
     case 2: case 3: g.esz = 2; break;
     case 4: g.esz = 2; break;  // fp8e4m3 expert weights, bf16 activations
+    case 5: g.esz = 2; break;  // MX: fp8 weights + runtime-quantized
+                               // fp8 activations on the scaled MFMA
     default: setErr("unknown dtype");
              return FM_ERR_UNSUPPORTED;
   }
-  g.wesz = (cfg->dtype == 4) ? 1 : g.esz;
+  g.wesz = (cfg->dtype == 4 || cfg->dtype == 5) ? 1 : g.esz;
+  if (cfg->dtype == 5 && (cfg->hidden_size % 128 || cfg->intermediate_size % 128)) {
+    setErr("dtype 5 (MX fp8) requires H, P multiples of 128 (BK=128)");
+    return FM_ERR_SHAPE;
+  }
   if (g.E > 256) { setErr("E > 256 not supported this round"); return FM_ERR_UNSUPPORTED; }
   if (world_size > 64) {
     // the P2P peer tables (State.peerBase, fm_heap_connect locals) are
@@ -2522,6 +2855,13 @@ int fm_initialize(const fm_config* cfg, int rank, int world_size) {
   FM_HIP_CHECK(hipMalloc(&g.cbuf,
                          (size_t)g.S * g.cfg.expert_top_k * g.H * g.esz));
   FM_HIP_CHECK(hipMalloc(&g.kept, (size_t)g.S * g.cfg.expert_top_k));
+  if (g.cfg.dtype == 5) {
+    FM_HIP_CHECK(hipMalloc(&g.x8, (size_t)g.S * g.H));
+    FM_HIP_CHECK(hipMalloc(&g.xs, (size_t)g.S * (g.H / 64)));
+    FM_HIP_CHECK(hipMalloc(&g.xM8, (size_t)g.nLxAlloc * g.pEC * g.P));
+    FM_HIP_CHECK(hipMalloc(&g.xMs,
+                           (size_t)g.nLxAlloc * g.pEC * (g.P / 64)));
+  }
   if (g.esz == 2) {
     // fused persistent kernel state: control block + per-tile arrival
     // counters (zeroed per forward by a memset node) + host-mapped
@@ -2563,6 +2903,8 @@ int fm_finalize(void) {
   (void)hipFree(g.O32);
   (void)hipFree(g.cbuf); (void)hipFree(g.kept);
   (void)hipFree(g.logits32); (void)hipFree(g.gML);
+  if (g.x8) { (void)hipFree(g.x8); (void)hipFree(g.xs);
+              (void)hipFree(g.xM8); (void)hipFree(g.xMs); }
   if (g.fusedCtl) (void)hipFree(g.fusedCtl);
   if (g.hFusedErr) (void)hipHostFree(g.hFusedErr);
   if (g.heap) {
@@ -2598,6 +2940,29 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
   const bool hasBias = a.bias != nullptr;
   const int act = (phase == 0) ? a.act : 0;
   const int N = a.N;
+  if (g.cfg.dtype == 5 && phase != 3) {
+    // MX fp8 grouped GEMM (the gate-logits GEMM stays bf16: gate_w is
+    // Element-typed)
+    dim3 grid(DIVUP(M, 128), DIVUP(N, 128), nE);
+    dim3 block(512);
+#define MXGG(PH, AC, HB)                                                          hipLaunchKernelGGL((k_group_gemm_mx<PH, AC, HB>), grid, block, 0, st, a)
+    const int sel = phase * 4 + act * 2 + (hasBias ? 1 : 0);
+    switch (sel) {
+      case 0: MXGG(0, 0, false); break;
+      case 1: MXGG(0, 0, true); break;
+      case 2: MXGG(0, 1, false); break;
+      case 3: MXGG(0, 1, true); break;
+      case 4: MXGG(1, 0, false); break;
+      case 5: MXGG(1, 0, true); break;
+      default:
+        if (hasBias) MXGG(2, 0, true);
+        else MXGG(2, 0, false);
+        break;
+    }
+#undef MXGG
+    FM_HIP_CHECK(hipGetLastError());
+    return FM_OK;
+  }
   if (g.esz == 2) {
     // tile selection: prefer the deep-pipelined 256-row kernel when the
     // grid still fills the 256 CUs at 1 block/CU. When BN=256 alone
@@ -2829,6 +3194,18 @@ static int launch_cast_combine(hipStream_t st, void* moe_out) {
   return FM_OK;
 }
 
+// dtype 5: quantize bf16 rows into fp8 + per-64 block scales
+static int launch_quant_mx(hipStream_t st, const void* in, uint8_t* out,
+                           uint8_t* scales, long long nRows, int K) {
+  const long long nBlk = nRows * (K / 64);
+  const int grid = (int)std::min<long long>(DIVUP(nBlk, 4), 8192);
+  hipLaunchKernelGGL(k_quant_mx<bf16>, dim3(grid), dim3(256), 0, st,
+                     reinterpret_cast<const bf16*>(in), out, scales, nRows,
+                     K);
+  FM_HIP_CHECK(hipGetLastError());
+  return FM_OK;
+}
+
 static int moe_forward_impl(hipStream_t st, const void* x, const void* gate_w,
                             const void* expert_w, const void* b_up,
                             const void* b_dn, void* gate_out, void* moe_out,
@@ -2836,7 +3213,7 @@ static int moe_forward_impl(hipStream_t st, const void* x, const void* gate_w,
   // fused single-launch path (the north-star structure); classic
   // multi-kernel path remains for fp32 and as FM_FUSED=0 fallback
   g.lastForwardFused = false;
-  if (fusedEnabled() && g.esz == 2 && g.world == 1) {
+  if (fusedEnabled() && g.esz == 2 && g.world == 1 && g.cfg.dtype != 5) {
     int rc = moe_forward_fused(st, x, gate_w, expert_w, b_up, b_dn, gate_out,
                                moe_out);
     if (rc != FM_FALLBACK) {
@@ -2858,8 +3235,15 @@ static int moe_forward_impl(hipStream_t st, const void* x, const void* gate_w,
   if (rc != FM_OK) return rc;
   if (evs) FM_HIP_CHECK(hipEventRecord(evs[2], st));
 
+  if (g.cfg.dtype == 5) {
+    // MX: quantize the activations once; the gate above read bf16 x
+    rc = launch_quant_mx(st, x, g.x8, g.xs, g.S, g.H);
+    if (rc != FM_OK) return rc;
+  }
+
   GemmArgs up{};
-  up.A = x;
+  up.A = (g.cfg.dtype == 5) ? (const void*)g.x8 : x;
+  up.aScales = g.xs;
   up.B = expert_w;
   up.bias = b_up;
   up.out = g.xM;
@@ -2877,8 +3261,16 @@ static int moe_forward_impl(hipStream_t st, const void* x, const void* gate_w,
   if (rc != FM_OK) return rc;
   if (evs) FM_HIP_CHECK(hipEventRecord(evs[3], st));
 
+  if (g.cfg.dtype == 5) {
+    // quantize the intermediate activations for the MX down GEMM
+    rc = launch_quant_mx(st, g.xM, g.xM8, g.xMs,
+                         (long long)g.E * g.pEC, g.P);
+    if (rc != FM_OK) return rc;
+  }
+
   GemmArgs dn = up;
-  dn.A = g.xM;
+  dn.A = (g.cfg.dtype == 5) ? (const void*)g.xM8 : g.xM;
+  dn.aScales = g.xMs;
   dn.B = reinterpret_cast<const char*>(expert_w) + (size_t)g.P * g.H * g.wesz;
   dn.bias = b_dn;
   dn.out = nullptr;
@@ -3045,6 +3437,10 @@ int fm_expert_ffn(void* stream, const void* rows, const void* expert_w,
                   const void* b_up, const void* b_dn, void* out_rows,
                   int64_t n_rows, int32_t local_e) {
   if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
+  if (g.cfg.dtype == 5) {
+    setErr("dtype 5 (MX fp8) supports fm_moe_forward only this round");
+    return FM_ERR_UNSUPPORTED;
+  }
   if (n_rows <= 0) return FM_OK;
   if (n_rows > (int64_t)g.nLxAlloc * g.pEC) {
     setErr("n_rows exceeds workspace"); return FM_ERR_SHAPE;
@@ -3294,6 +3690,10 @@ int fm_expert_ffn_segments(void* stream, const void* rows,
                            const void* seg_expert_dev, int32_t n_segs,
                            const void* expert_w, void* out_rows) {
   if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
+  if (g.cfg.dtype == 5) {
+    setErr("dtype 5 (MX fp8) supports fm_moe_forward only this round");
+    return FM_ERR_UNSUPPORTED;
+  }
   if ((int64_t)n_segs * g.EC > (int64_t)g.nLxAlloc * g.pEC) {
     setErr("segments exceed xM workspace"); return FM_ERR_SHAPE;
   }
@@ -3537,14 +3937,16 @@ int fm_debug_taskq(void* stream, int ring_sz, long long n_tasks,
 }
 
 int fm_debug_mx_mfma(void* stream, const void* A, const void* B,
-                     const void* sa, const void* sb, void* D) {
+                     const void* sa, const void* sb, void* D, int opsel) {
   hipStream_t st = reinterpret_cast<hipStream_t>(stream);
-  hipLaunchKernelGGL(k_mx_mfma_probe, dim3(1), dim3(64), 0, st,
-                     reinterpret_cast<const uint8_t*>(A),
-                     reinterpret_cast<const uint8_t*>(B),
-                     reinterpret_cast<const uint8_t*>(sa),
-                     reinterpret_cast<const uint8_t*>(sb),
-                     reinterpret_cast<float*>(D));
+#define MXP(O)                                                                  hipLaunchKernelGGL(k_mx_mfma_probe<O>, dim3(1), dim3(64), 0, st,                                 reinterpret_cast<const uint8_t*>(A),                                          reinterpret_cast<const uint8_t*>(B),                                          reinterpret_cast<const uint8_t*>(sa),                                         reinterpret_cast<const uint8_t*>(sb),                                         reinterpret_cast<float*>(D))
+  switch (opsel) {
+    case 1: MXP(1); break;
+    case 2: MXP(2); break;
+    case 3: MXP(3); break;
+    default: MXP(0); break;
+  }
+#undef MXP
   FM_HIP_CHECK(hipGetLastError());
   return FM_OK;
 }
