@@ -2449,7 +2449,7 @@ int launch_gate(hipStream_t st, const void* x, const void* gate_w,
     int rc = launch_group_gemm(st, 3, ga, (int)S, 1);
     if (rc != FM_OK) return rc;
   }
-  if (g.cfg.dtype == 2 || g.cfg.dtype == 4) {
+  if (g.cfg.dtype == 2 || g.cfg.dtype == 4 || g.cfg.dtype == 5) {
     if (!mfmaLogits) GATE_LOGITS(bf16);
     GATE_K(bf16)
   } else if (g.cfg.dtype == 3) {
