@@ -334,8 +334,11 @@ def test_odd64_shapes_bf16(fresh_moe):
     """H, P multiples of 64 (not 128): the schema's contract; the tile
     guards (B-row clamp + col < N epilogue mask) must handle the ragged
     N tiles (round-1 rejected these shapes; VERDICT r01 weak #7)."""
+    # capacity_factor 2: EC at CF=1 equals the mean load, so experts
+    # overflow and the kept-set becomes schedule-dependent (flaky vs the
+    # oracle); headroom keeps the comparison deterministic
     cfg, path = make_cfg(hidden_size=192, intermediate_size=320,
-                         sequence_len=256)
+                         sequence_len=256, capacity_factor=2)
     out, gate_out, ref, ocfg = run_pair(fresh_moe, cfg, path)
     assert_values(gate_out, ref["gate_out"], "bf16", "gate_out")
     assert_values(out, ref["moe_out"], "bf16", "moe_out")
@@ -525,10 +528,13 @@ def test_error_behaviour(fresh_moe):
 def test_mx_mfma_layout_probe():
     """Pin the empirically determined mfma_scale_f32_16x16x128_f8f6f4
     contract the MX GEMM builds on: lane l holds
-    A[row=l&15][k=32*(l>>4)..+32] / B[k-block][col=l&15] as 8 dwords;
-    the SCALE byte of lane-group g (opsel-selected) covers
-    k in [64g, 64g+64) - effective 64-element scale granularity; groups
-    2..3 are ignored. E8M0: value = 2^(byte-127)."""
+    A[row=l&15][k=32*(l>>4)..+32] / B[k-block][col=l&15] as 8 dwords.
+    The SCALE byte of lane-group g (opsel-selected, E8M0 = 2^(b-127))
+    covers the interleaved 16-element chunk pair
+    {k: k>>6 == g&1 and (k>>4)&1 == g>>1} - so supplying the scale of
+    CONTIGUOUS 64-block (g&1) on every lane-group scales each hardware
+    block uniformly and correctly (what k_quant_mx + the MX GEMM do).
+    Wrong lane wiring fails loudly here."""
     import flashmoe_amd._ext as _ext
 
     lib = _ext.load()
@@ -545,13 +551,14 @@ def test_mx_mfma_layout_probe():
         scale = (2.0 ** (torch.tensor(sa64[:, b]).float() - 127)).unsqueeze(1) * \
                 (2.0 ** (torch.tensor(sb64[:, b]).float() - 127)).unsqueeze(0)
         want += scale * blk
-    # per-lane i32 scale words: lane-group g supplies block g's byte 0
+    # per-lane i32 scale words: lane-group g supplies the contiguous
+    # 64-block (g & 1)'s byte (groups 0,2 -> block 0; 1,3 -> block 1)
     saL = np.full(64, 0x7F7F7F7F, dtype=np.uint32)
     sbL = np.full(64, 0x7F7F7F7F, dtype=np.uint32)
     for row in range(16):
-        for gblk in range(2):
-            saL[gblk * 16 + row] = (0x7F7F7F00 | int(sa64[row, gblk]))
-            sbL[gblk * 16 + row] = (0x7F7F7F00 | int(sb64[row, gblk]))
+        for g in range(4):
+            saL[g * 16 + row] = (0x7F7F7F00 | int(sa64[row, g & 1]))
+            sbL[g * 16 + row] = (0x7F7F7F00 | int(sb64[row, g & 1]))
     dA = A.cuda().view(torch.uint8).contiguous()
     dB = Bt.cuda().view(torch.uint8).contiguous()
     dsa = torch.from_numpy(saL.view(np.int32).copy()).cuda()
@@ -766,7 +773,8 @@ def test_mx_fp8_gelu_bias(fresh_moe):
     import flashmoe_amd._ext as _ext
 
     cfg, path = make_cfg(torch_dtype=5, hidden_act=1, sequence_len=256,
-                         hidden_size=128, intermediate_size=384)
+                         hidden_size=128, intermediate_size=384,
+                         capacity_factor=2)  # no-overflow headroom
     fresh_moe.initialize(path, rank=0, world_size=1)
     lib = _ext.load()
     S, H, P, E = 256, 128, 384, 8
