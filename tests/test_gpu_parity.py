@@ -794,7 +794,7 @@ def test_mx_fp8_gelu_bias(fresh_moe):
         ctypes.c_void_p(gate_out.data_ptr()), ctypes.c_void_p(out.data_ptr()),
         S), "fm_moe_forward")
     torch.cuda.synchronize()
-    ocfg = OracleConfig(num_experts=E, expert_top_k=2, capacity_factor=1,
+    ocfg = OracleConfig(num_experts=E, expert_top_k=2, capacity_factor=2,
                         hidden_act=1, element="bf16", mx_fp8=True)
     ref = oracle_forward(x.float().cpu().numpy(),
                          gw.float().cpu().numpy().reshape(-1),
