@@ -1185,20 +1185,36 @@ __device__ __forceinline__ bool mx_gemm_job_body(
   const int gk = lane >> 4;
   const int K64 = K / 64;
 
+  // per-lane A-scale source rows (this lane's 4 fragment rows), hoisted
+  const uint8_t* sRow[MI];
+#pragma unroll
+  for (int mi = 0; mi < MI; ++mi) {
+    const int R = wr * 64 + mi * 16 + rl;
+    const size_t arow = (PHASE == 0) ? (size_t)tpsTok(sTps[R].tokenIdx)
+                                     : (size_t)min(m0 + R, mCap - 1);
+    sRow[mi] = aS + (aBase / 64) + arow * K64 + (gk & 1);
+  }
   const int nK = K / BK;
   stage(0, 0);
+  // software-pipelined scale bytes: tile t's 4 scattered global byte
+  // loads are issued one tile AHEAD so they never sit on the MFMA
+  // dependency chain
+  int sav[MI], savNext[MI];
+#pragma unroll
+  for (int mi = 0; mi < MI; ++mi) sav[mi] = sRow[mi][0];
   wait_vmcnt<0>();
   __builtin_amdgcn_s_barrier();
   for (int t = 0; t < nK; ++t) {
     const bool stageNow = t + 1 < nK;
-    if (stageNow) stage((t + 1) * BK, (t + 1) % SE);
+    if (stageNow) {
+      stage((t + 1) * BK, (t + 1) % SE);
+#pragma unroll
+      for (int mi = 0; mi < MI; ++mi) savNext[mi] = sRow[mi][(t + 1) * 2];
+    }
     const uint8_t* Al = Abase + (t % SE) * BM * BK;
     const uint8_t* Bl = Bbase + (t % SE) * BN * BK;
     __builtin_amdgcn_s_setprio(1);
-    // A-scale bytes: this lane's rows at k64 block (t*2 + (gk&1))
-    const int k64 = t * 2 + (gk & 1);
     i32x8 af[MI];
-    int sav[MI];
 #pragma unroll
     for (int mi = 0; mi < MI; ++mi) {
       const int R = wr * 64 + mi * 16 + rl;
@@ -1211,12 +1227,6 @@ __device__ __forceinline__ bool mx_gemm_job_body(
         af[mi][d] = (int)lo[d];
         af[mi][4 + d] = (int)hi[d];
       }
-      // scale row: the STAGED row R's source row (token-gathered for
-      // PHASE 0). Recompute like the staging pass did.
-      const size_t arow = (PHASE == 0)
-          ? (size_t)tpsTok(sTps[R].tokenIdx)
-          : (size_t)min(m0 + R, mCap - 1);
-      sav[mi] = aS[(aBase / 64) + arow * K64 + k64];
     }
     i32x8 bfr[NF];
 #pragma unroll
@@ -1240,6 +1250,8 @@ __device__ __forceinline__ bool mx_gemm_job_body(
             af[mi], bfr[ni], accv[mi][ni], 0 /*fp8*/, 0 /*fp8*/, 0, sav[mi],
             0, 0x7F /*B scale 1.0*/);
     __builtin_amdgcn_s_setprio(0);
+#pragma unroll
+    for (int mi = 0; mi < MI; ++mi) sav[mi] = savNext[mi];
     wait_vmcnt<0>();
     __builtin_amdgcn_s_barrier();
   }
@@ -2945,9 +2957,23 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
   const int N = a.N;
   if (g.cfg.dtype == 5 && phase != 3) {
     // MX fp8 grouped GEMM (the gate-logits GEMM stays bf16: gate_w is
-    // Element-typed)
+    // Element-typed). Persistent grid at 2 blocks/CU: per-job setup
+    // (sTps/scale-row pointers) amortizes over many 128^2 tiles.
+    GemmArgs aa = a;
+    aa.splitK = 1;
+    aa.totalJobs = 0;
     dim3 grid(DIVUP(M, 128), DIVUP(N, 128), nE);
     dim3 block(512);
+    {
+      const int J = grid.x * grid.y * nE;
+      const int resident = 2 * (g.nCU > 0 ? g.nCU : 256);
+      if (J > resident) {
+        aa.totalJobs = J;
+        aa.jobsMT = grid.x;
+        aa.jobsNT = grid.y;
+        grid = dim3(resident, 1, 1);
+      }
+    }
 #define MXGG(PH, AC, HB)                                                          hipLaunchKernelGGL((k_group_gemm_mx<PH, AC, HB>), grid, block, 0, st, a)
     const int sel = phase * 4 + act * 2 + (hasBias ? 1 : 0);
     switch (sel) {
