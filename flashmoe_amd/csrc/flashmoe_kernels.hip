@@ -2974,7 +2974,7 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
         grid = dim3(resident, 1, 1);
       }
     }
-#define MXGG(PH, AC, HB)                                                          hipLaunchKernelGGL((k_group_gemm_mx<PH, AC, HB>), grid, block, 0, st, a)
+#define MXGG(PH, AC, HB)                                                          hipLaunchKernelGGL((k_group_gemm_mx<PH, AC, HB>), grid, block, 0, st, aa)
     const int sel = phase * 4 + act * 2 + (hasBias ? 1 : 0);
     switch (sel) {
       case 0: MXGG(0, 0, false); break;
