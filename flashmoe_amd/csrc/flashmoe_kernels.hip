@@ -1094,11 +1094,11 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
 // PHASE semantics match gemm_job_body (0 up / 1 down+slots / 2 packed).
 // ---------------------------------------------------------------------------
 
-template <int PHASE>
+template <int PHASE, int BM = 128, int BN = 128>
 __device__ __forceinline__ bool mx_gemm_job_body(
     const GemmArgs& a, char* smemBase, int e, int m0, int n0, int act,
     bool hasBias) {
-  constexpr int BM = 128, BN = 128, BK = 128;
+  constexpr int BK = 128;
   constexpr int SE = 2;
   uint8_t* Abase = reinterpret_cast<uint8_t*>(smemBase);   // SE x [BM][BK]
   uint8_t* Bbase = reinterpret_cast<uint8_t*>(smemBase + SE * BM * BK);
@@ -1138,39 +1138,41 @@ __device__ __forceinline__ bool mx_gemm_job_body(
   // chunk XOR-swizzled by row for conflict-free b128 fragment reads
   const int grow8 = lane >> 3;
   const int schunk = (lane & 7) ^ grow8;
+  constexpr int GPW_A = BM / 8 / 8;  // 1 KiB glds groups per wave (A)
+  constexpr int GPW_B = BN / 8 / 8;
   const size_t aBase = (size_t)e * a.strideAExpert;  // bytes (u8 A)
-  const uint8_t* aSrc[2];
-  const uint8_t* bSrc[2];
-  uint32_t aRowIdx[2];  // scale row of each staged A row (this wave's)
+  const uint8_t* aSrc[GPW_A];
+  const uint8_t* bSrc[GPW_B];
 #pragma unroll
-  for (int i = 0; i < 2; ++i) {
-    const int row = (wave * 2 + i) * 8 + grow8;
+  for (int i = 0; i < GPW_A; ++i) {
+    const int row = (wave * GPW_A + i) * 8 + grow8;
     const size_t arow = (PHASE == 0) ? (size_t)tpsTok(sTps[row].tokenIdx)
                                      : (size_t)min(m0 + row, mCap - 1);
     aSrc[i] = Ag + aBase + arow * (size_t)K + schunk * 16;
-    aRowIdx[i] = (uint32_t)arow;
   }
 #pragma unroll
-  for (int i = 0; i < 2; ++i) {
-    const int row = (wave * 2 + i) * 8 + grow8;
+  for (int i = 0; i < GPW_B; ++i) {
+    const int row = (wave * GPW_B + i) * 8 + grow8;
     bSrc[i] = Bg + (size_t)min(n0 + row, N - 1) * K + schunk * 16;
   }
   auto stage = [&](int kt, int buf) {
 #pragma unroll
-    for (int i = 0; i < 2; ++i)
+    for (int i = 0; i < GPW_A; ++i)
       __builtin_amdgcn_global_load_lds(
-          (gas_u32*)(aSrc[i] + kt), (las_u32*)(Abase + buf * BM * BK +
-                                               (wave * 2 + i) * 1024), 16, 0, 0);
+          (gas_u32*)(aSrc[i] + kt),
+          (las_u32*)(Abase + buf * BM * BK + (wave * GPW_A + i) * 1024), 16,
+          0, 0);
 #pragma unroll
-    for (int i = 0; i < 2; ++i)
+    for (int i = 0; i < GPW_B; ++i)
       __builtin_amdgcn_global_load_lds(
-          (gas_u32*)(bSrc[i] + kt), (las_u32*)(Bbase + buf * BN * BK +
-                                               (wave * 2 + i) * 1024), 16, 0, 0);
+          (gas_u32*)(bSrc[i] + kt),
+          (las_u32*)(Bbase + buf * BN * BK + (wave * GPW_B + i) * 1024), 16,
+          0, 0);
   };
 
-  // wave grid 2(M) x 4(N): 64 rows x 32 cols per wave
+  // wave grid 2(M) x 4(N): BM/2 rows x BN/4 cols per wave
   const int wr = wave >> 2, wc = wave & 3;
-  constexpr int MI = 4, NF = 2;
+  constexpr int MI = BM / 32, NF = BN / 64;
   f32x4 accv[MI][NF];
 #pragma unroll
   for (int i = 0; i < MI; ++i)
@@ -1189,7 +1191,7 @@ __device__ __forceinline__ bool mx_gemm_job_body(
   const uint8_t* sRow[MI];
 #pragma unroll
   for (int mi = 0; mi < MI; ++mi) {
-    const int R = wr * 64 + mi * 16 + rl;
+    const int R = wr * (BM / 2) + mi * 16 + rl;
     const size_t arow = (PHASE == 0) ? (size_t)tpsTok(sTps[R].tokenIdx)
                                      : (size_t)min(m0 + R, mCap - 1);
     sRow[mi] = aS + (aBase / 64) + arow * K64 + (gk & 1);
@@ -1217,7 +1219,7 @@ __device__ __forceinline__ bool mx_gemm_job_body(
     i32x8 af[MI];
 #pragma unroll
     for (int mi = 0; mi < MI; ++mi) {
-      const int R = wr * 64 + mi * 16 + rl;
+      const int R = wr * (BM / 2) + mi * 16 + rl;
       const int c0 = (2 * gk) ^ (R & 7);
       const int c1 = (2 * gk + 1) ^ (R & 7);
       const u32x4 lo = *reinterpret_cast<const u32x4*>(&Al[R * BK + c0 * 16]);
@@ -1231,7 +1233,7 @@ __device__ __forceinline__ bool mx_gemm_job_body(
     i32x8 bfr[NF];
 #pragma unroll
     for (int ni = 0; ni < NF; ++ni) {
-      const int R = wc * 32 + ni * 16 + rl;
+      const int R = wc * (BN / 4) + ni * 16 + rl;
       const int c0 = (2 * gk) ^ (R & 7);
       const int c1 = (2 * gk + 1) ^ (R & 7);
       const u32x4 lo = *reinterpret_cast<const u32x4*>(&Bl[R * BK + c0 * 16]);
@@ -1277,13 +1279,15 @@ __device__ __forceinline__ bool mx_gemm_job_body(
       __syncthreads();
     }
   }
-  float bv[NF] = {0.f, 0.f};
+  float bv[NF];
+#pragma unroll
+  for (int ni = 0; ni < NF; ++ni) bv[ni] = 0.f;
   if (hasBias) {
     const bf16* bptr = reinterpret_cast<const bf16*>(a.bias) +
                        (a.strideBExpert ? (size_t)we * N : 0);
 #pragma unroll
     for (int ni = 0; ni < NF; ++ni) {
-      const int col = n0 + wc * 32 + ni * 16 + cl;
+      const int col = n0 + wc * (BN / 4) + ni * 16 + cl;
       if (col < N) bv[ni] = toF(bptr[col]);
     }
   }
@@ -1291,14 +1295,14 @@ __device__ __forceinline__ bool mx_gemm_job_body(
   for (int mi = 0; mi < MI; ++mi) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int row = wr * 64 + mi * 16 + r0 + r;
+      const int row = wr * (BM / 2) + mi * 16 + r0 + r;
       const int m = m0 + row;
       if ((uint32_t)m >= routed) continue;
       const TPS tp = sTps[row];
       const float rowScale = scaled ? sScale[row] : 1.0f;
 #pragma unroll
       for (int ni = 0; ni < NF; ++ni) {
-        const int col = n0 + wc * 32 + ni * 16 + cl;
+        const int col = n0 + wc * (BN / 4) + ni * 16 + cl;
         if (col >= N) continue;
         float v = accv[mi][ni][r] + bv[ni];
         if constexpr (PHASE == 0) {
@@ -1328,9 +1332,9 @@ __device__ __forceinline__ bool mx_gemm_job_body(
   return true;
 }
 
-template <int PHASE, int ACT, bool HAS_BIAS>
+template <int PHASE, int ACT, bool HAS_BIAS, int BM = 128, int BN = 128>
 __global__ __launch_bounds__(512, 2) void k_group_gemm_mx(GemmArgs a) {
-  constexpr int BM = 128, BN = 128, BK = 128, SE = 2;
+  constexpr int BK = 128, SE = 2;
   __shared__ __attribute__((aligned(16))) char smem[
       SE * BM * BK + SE * BN * BK + BM * 8 + 16];
   const int mT = a.totalJobs > 0 ? a.jobsMT : gridDim.x;
@@ -1347,8 +1351,8 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_mx(GemmArgs a) {
         : (xcd < rx ? xcd * (qx + 1) : rx * (qx + 1) + (xcd - rx) * qx) + pos;
     const int e = swz / (mT * nT);
     const int rem = swz % (mT * nT);
-    (void)mx_gemm_job_body<PHASE>(a, smem, e, (rem % mT) * BM,
-                                  (rem / mT) * BN, ACT, HAS_BIAS);
+    (void)mx_gemm_job_body<PHASE, BM, BN>(a, smem, e, (rem % mT) * BM,
+                                          (rem / mT) * BN, ACT, HAS_BIAS);
   }
 }
 
@@ -2957,16 +2961,21 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
   const int N = a.N;
   if (g.cfg.dtype == 5 && phase != 3) {
     // MX fp8 grouped GEMM (the gate-logits GEMM stays bf16: gate_w is
-    // Element-typed). Persistent grid at 2 blocks/CU: per-job setup
-    // (sTps/scale-row pointers) amortizes over many 128^2 tiles.
+    // Element-typed). 256x256 tile when the shape affords it (staging
+    // bytes/flop halve: the 128^2 MX tile is glds-rate-bound at ~140
+    // GB/s/CU demand vs the ~108 measured ceiling); 128^2 otherwise.
+    // Persistent grid amortizes per-job setup.
     GemmArgs aa = a;
     aa.splitK = 1;
     aa.totalJobs = 0;
-    dim3 grid(DIVUP(M, 128), DIVUP(N, 128), nE);
+    const bool big = (M >= 256) && (N >= 256) &&
+                     DIVUP(M, 256) * DIVUP(N, 256) * nE >= (g.nCU > 0 ? g.nCU : 256);
+    const int bm = big ? 256 : 128, bn = big ? 256 : 128;
+    dim3 grid(DIVUP(M, bm), DIVUP(N, bn), nE);
     dim3 block(512);
     {
       const int J = grid.x * grid.y * nE;
-      const int resident = 2 * (g.nCU > 0 ? g.nCU : 256);
+      const int resident = (big ? 1 : 2) * (g.nCU > 0 ? g.nCU : 256);
       if (J > resident) {
         aa.totalJobs = J;
         aa.jobsMT = grid.x;
@@ -2974,7 +2983,15 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
         grid = dim3(resident, 1, 1);
       }
     }
-#define MXGG(PH, AC, HB)                                                          hipLaunchKernelGGL((k_group_gemm_mx<PH, AC, HB>), grid, block, 0, st, aa)
+#define MXGG(PH, AC, HB)                                                      \
+    do {                                                                      \
+      if (big)                                                                \
+        hipLaunchKernelGGL((k_group_gemm_mx<PH, AC, HB, 256, 256>), grid,     \
+                           block, 0, st, aa);                                 \
+      else                                                                    \
+        hipLaunchKernelGGL((k_group_gemm_mx<PH, AC, HB, 128, 128>), grid,     \
+                           block, 0, st, aa);                                 \
+    } while (0)
     const int sel = phase * 4 + act * 2 + (hasBias ? 1 : 0);
     switch (sel) {
       case 0: MXGG(0, 0, false); break;
