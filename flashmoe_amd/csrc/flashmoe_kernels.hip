@@ -229,75 +229,95 @@ __global__ __launch_bounds__(256) void k_gate_logits(
               part[tok * (Ec + 1) + e]);
 }
 
+// ---------------------------------------------------------------------------
+// Softmax / top-k / ordered placement for one 128-token tile (reference
+// gate.cuh:557-718 semantics): the shared core of k_gate_route (classic
+// path) and the fused kernel's inline route. 512 threads; the per-token
+// top-k runs on FOUR threads per token, each owning a <=64-expert chunk
+// with a u64 taken-mask and combining candidates with width-4 shuffles
+// (larger value wins, equal value -> smaller index == the reference's
+// ascending strict-> first-index scan). The serial one-thread-per-token
+// scan was the dominant gate cost at E=256 (~144 us at the cfg5 shape).
+// Reads the tile's logits from logits32 and RE-ZEROES them (keeps the
+// split-K logits accumulate memset-free).
+// ---------------------------------------------------------------------------
 template <typename T, int K>
-__global__ __launch_bounds__(256) void k_gate_route(
-    const float* __restrict__ logits32, T* __restrict__ gate_out,
-    TPS* __restrict__ tokenIds, uint32_t* __restrict__ eC, int S, int E,
-    int PX, int EC, int pEC, float* __restrict__ gML,
-    float* __restrict__ gMeC, uint8_t* __restrict__ kept) {
-  // gML/gMeC: training-mode aux-loss accumulators (gate.cuh:273-299,
-  // 763-773; types.cuh:936-958): gML[e] += column-mean of softmax probs,
-  // gMeC[e] += routed fraction. Null in inference mode.
+__device__ __forceinline__ void route_tile_core(
+    char* smem, float* __restrict__ logits32, T* __restrict__ gate_out,
+    TPS* __restrict__ tokenIds, uint32_t* __restrict__ eC,
+    uint8_t* __restrict__ kept, float* __restrict__ gML,
+    float* __restrict__ gMeC, int S, int E, int PX, int EC, int pEC,
+    int m0) {
   constexpr int BM = 128;
-  extern __shared__ __attribute__((aligned(16))) char smem[];
   float* logits = reinterpret_cast<float*>(smem);                  // [BM][E+1]
   uint16_t* sel = reinterpret_cast<uint16_t*>(logits + BM * (E + 1));
   uint16_t* localIdx = sel + BM * K;
   // +2: one dump slot so the counting scan's store is UNCONDITIONAL
   // (an exec-masked conditional store compiled into ~520 serial
-  // s_and_saveexec blocks = ~80% of this kernel's runtime, see ISA)
+  // s_and_saveexec blocks = ~80% of the round-1 kernel's runtime)
   uint32_t* base = reinterpret_cast<uint32_t*>(localIdx + BM * K + 2);  // [E]
   float* sInv = reinterpret_cast<float*>(base + E);                // [BM] 1/d
   float* sMax = sInv + BM;                                         // [BM]
+  float* sCw = sMax + BM;                                          // [BM] mCw
   const int tid = threadIdx.x;
-  const int m0 = blockIdx.x * BM;
-  // cooperative coalesced copy of the tile's logits into LDS, then
-  // re-zero the global tile (keeps logits32 zero between forwards so
-  // the split-K logits GEMM can atomically accumulate without a memset)
-  for (int i = tid; i < BM * E; i += 256) {
+  for (int i = tid; i < BM * E; i += 512) {
     const int row = i / E, col = i % E;
-    float* src = const_cast<float*>(&logits32[(size_t)(m0 + row) * E + col]);
+    float* src = &logits32[(size_t)(m0 + row) * E + col];
     logits[row * (E + 1) + col] = *src;
     *src = 0.0f;
   }
   __syncthreads();
 
-  float mCw = 0.0f;
-  uint16_t mySel[K];
-  if (tid < BM) {
-    const float* lrow = logits + tid * (E + 1);
-    float m = -INFINITY;
-    for (int e = 0; e < E; ++e) m = fmaxf(m, lrow[e]);
-    float d = 0.0f;
-    for (int e = 0; e < E; ++e) d += __expf(lrow[e] - m);
-    const float inv_d = 1.0f / d;
-    sInv[tid] = inv_d;  // always: the coalesced gate_out pass below
-    sMax[tid] = m;      // recomputes probs from these
-    // iterative argmax, strict >, first index wins; "taken" by direct
-    // comparison with earlier selections (E up to 256)
-#pragma unroll
-    for (int i = 0; i < K; ++i) {
-      float sV = -INFINITY;
-      int sIdx = 0;
-      for (int e = 0; e < E; ++e) {
-        bool taken = false;
-#pragma unroll
-        for (int j = 0; j < K; ++j) taken |= (j < i) && (mySel[j] == e);
-        if (!taken && lrow[e] > sV) { sV = lrow[e]; sIdx = e; }
+  {
+    const int token = tid >> 2;  // 4 threads per token
+    const int sub = tid & 3;
+    if (token < BM) {
+      const float* lrow = logits + token * (E + 1);
+      const int chunk = (E + 3) >> 2;  // <= 64: the taken-mask is a u64
+      const int e0 = sub * chunk;
+      const int e1 = min(E, e0 + chunk);
+      float m = -INFINITY;
+      for (int e = e0; e < e1; ++e) m = fmaxf(m, lrow[e]);
+      m = fmaxf(m, __shfl_xor(m, 1, 4));
+      m = fmaxf(m, __shfl_xor(m, 2, 4));
+      float d = 0.0f;
+      for (int e = e0; e < e1; ++e) d += __expf(lrow[e] - m);
+      d += __shfl_xor(d, 1, 4);
+      d += __shfl_xor(d, 2, 4);
+      const float inv_d = 1.0f / d;
+      if (sub == 0) {
+        sInv[token] = inv_d;  // the coalesced gate_out pass below
+        sMax[token] = m;      // recomputes probs from these
       }
-      mySel[i] = (uint16_t)sIdx;
-      mCw += __expf(sV - m) * inv_d;
-    }
+      unsigned long long takenM = 0ull;
+      float mCw = 0.0f;
 #pragma unroll
-    for (int i = 0; i < K; ++i) sel[tid * K + i] = mySel[i];
+      for (int i = 0; i < K; ++i) {
+        float lv = -INFINITY;
+        int li = E;  // sentinel sorts after every real index
+        for (int e = e0; e < e1; ++e) {
+          const bool tk = (takenM >> (e - e0)) & 1ull;
+          if (!tk && lrow[e] > lv) { lv = lrow[e]; li = e; }
+        }
+#pragma unroll
+        for (int off = 1; off < 4; off <<= 1) {
+          const float ov = __shfl_xor(lv, off, 4);
+          const int oi = __shfl_xor(li, off, 4);
+          if (ov > lv || (ov == lv && oi < li)) { lv = ov; li = oi; }
+        }
+        if (li >= e0 && li < e1) takenM |= 1ull << (li - e0);
+        if (sub == 0) {
+          sel[token * K + i] = (uint16_t)li;
+          mCw += __expf(lv - m) * inv_d;
+        }
+      }
+      if (sub == 0) sCw[token] = mCw;
+    }
   }
   __syncthreads();
-  // cooperative COALESCED gate_out store: the old per-token loop wrote
-  // each prob as its own 2-byte store at 2*PX-byte thread stride (one
-  // cache line per element - measured the dominant cost of this
-  // kernel); here 256 threads sweep [BM][PX] row-major in 8-element
-  // vectors recomputing exp(l-max)*inv from LDS
-  for (int i = tid * 8; i < BM * PX; i += 256 * 8) {
+  // cooperative COALESCED gate_out store (recompute probs from LDS; a
+  // per-token scalar store pattern was one cache line per element)
+  for (int i = tid * 8; i < BM * PX; i += 512 * 8) {
     const int row = i / PX, col0 = i % PX;
     const float mR = sMax[row], ivR = sInv[row];
     const float* lrow = logits + row * (E + 1);
@@ -321,13 +341,13 @@ __global__ __launch_bounds__(256) void k_gate_route(
       for (int q = 0; q < 8; ++q) {
         const uint32_t w = v[q >> 1];
         const uint16_t sv = (q & 1) ? (uint16_t)(w >> 16) : (uint16_t)(w & 0xffff);
-        // branchless: non-matching entries write the dump slot
-        const bool m = (sv == (uint16_t)tid);
-        localIdx[m ? (mj + q) : BM * K] = (uint16_t)cnt;
-        cnt += m;
+        const bool mt = (sv == (uint16_t)tid);
+        localIdx[mt ? (mj + q) : BM * K] = (uint16_t)cnt;
+        cnt += mt;
       }
     }
-    base[tid] = atomicAdd(eC + tid, cnt);
+    base[tid] = __hip_atomic_fetch_add(eC + tid, cnt, __ATOMIC_RELAXED,
+                                       __HIP_MEMORY_SCOPE_AGENT);
     if (gML) {
       // column sum of probs for this expert over the tile's tokens
       float colSum = 0.0f;
@@ -341,16 +361,29 @@ __global__ __launch_bounds__(256) void k_gate_route(
   if (tid < BM) {
 #pragma unroll
     for (int i = 0; i < K; ++i) {
-      const int e = mySel[i];
+      const int e = sel[tid * K + i];
       const uint32_t slot = base[e] + localIdx[tid * K + i];
       const bool keep = slot < (uint32_t)EC;
       if (keep) {
         tokenIds[(size_t)e * pEC + slot] =
-            TPS{(uint32_t)(m0 + tid) | ((uint32_t)i << 28), mCw};
+            TPS{(uint32_t)(m0 + tid) | ((uint32_t)i << 28), sCw[tid]};
       }
       if (kept) kept[(size_t)(m0 + tid) * K + i] = keep ? 1 : 0;
     }
   }
+}
+
+template <typename T, int K>
+__global__ __launch_bounds__(512) void k_gate_route(
+    float* __restrict__ logits32, T* __restrict__ gate_out,
+    TPS* __restrict__ tokenIds, uint32_t* __restrict__ eC, int S, int E,
+    int PX, int EC, int pEC, float* __restrict__ gML,
+    float* __restrict__ gMeC, uint8_t* __restrict__ kept) {
+  // gML/gMeC: training-mode aux-loss accumulators (gate.cuh:273-299,
+  // 763-773; types.cuh:936-958). Null in inference mode.
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  route_tile_core<T, K>(smem, logits32, gate_out, tokenIds, eC, kept, gML,
+                        gMeC, S, E, PX, EC, pEC, blockIdx.x * 128);
 }
 
 // ---------------------------------------------------------------------------
@@ -1921,115 +1954,15 @@ struct FusedMeta {
   long long spinBound;
 };
 
-// softmax / top-k / ordered placement for one 128-token tile
-// (k_gate_route semantics at 512 threads; reference gate.cuh:557-718).
-// Reads the tile's logits from logits32 (atomic partials complete - the
-// caller acquired after the last arrival), re-zeroes them (keeps the
-// split-K logits accumulate memset-free), writes gate_out / tokenIds /
-// kept and bumps eC.
+// softmax / top-k / ordered placement for one 128-token tile inside the
+// fused kernel: thin wrapper over route_tile_core (shared with the
+// classic k_gate_route)
 template <typename T, int K>
 __device__ __forceinline__ void route_tile(char* smem, const FusedMeta& f,
                                            int tile) {
-  const int E = f.E, PX = f.PX, S = f.S;
-  float* logits = reinterpret_cast<float*>(smem);                  // [128][E+1]
-  uint16_t* sel = reinterpret_cast<uint16_t*>(logits + 128 * (E + 1));
-  uint16_t* localIdx = sel + 128 * K;
-  // +2: dump slot so the counting scan's store is UNCONDITIONAL
-  uint32_t* base = reinterpret_cast<uint32_t*>(localIdx + 128 * K + 2);  // [E]
-  float* sInv = reinterpret_cast<float*>(base + E);                // [128]
-  float* sMax = sInv + 128;                                        // [128]
-  const int tid = threadIdx.x;
-  const int m0 = tile * 128;
-  T* gate_out = reinterpret_cast<T*>(f.gate_out);
-  for (int i = tid; i < 128 * E; i += 512) {
-    const int row = i / E, col = i % E;
-    float* src = &f.logits32[(size_t)(m0 + row) * E + col];
-    logits[row * (E + 1) + col] = *src;
-    *src = 0.0f;
-  }
-  __syncthreads();
-
-  float mCw = 0.0f;
-  uint16_t mySel[K];
-  if (tid < 128) {
-    const float* lrow = logits + tid * (E + 1);
-    float m = -INFINITY;
-    for (int e = 0; e < E; ++e) m = fmaxf(m, lrow[e]);
-    float d = 0.0f;
-    for (int e = 0; e < E; ++e) d += __expf(lrow[e] - m);
-    const float inv_d = 1.0f / d;
-    sInv[tid] = inv_d;
-    sMax[tid] = m;
-    // iterative argmax, strict >, first index wins (gate.cuh:662-668)
-#pragma unroll
-    for (int i = 0; i < K; ++i) {
-      float sV = -INFINITY;
-      int sIdx = 0;
-      for (int e = 0; e < E; ++e) {
-        bool taken = false;
-#pragma unroll
-        for (int j = 0; j < K; ++j) taken |= (j < i) && (mySel[j] == e);
-        if (!taken && lrow[e] > sV) { sV = lrow[e]; sIdx = e; }
-      }
-      mySel[i] = (uint16_t)sIdx;
-      mCw += __expf(sV - m) * inv_d;
-    }
-#pragma unroll
-    for (int i = 0; i < K; ++i) sel[tid * K + i] = mySel[i];
-  }
-  __syncthreads();
-  // coalesced gate_out store (recompute probs from LDS)
-  for (int i = tid * 8; i < 128 * PX; i += 512 * 8) {
-    const int row = i / PX, col0 = i % PX;
-    const float mR = sMax[row], ivR = sInv[row];
-    const float* lrow = logits + row * (E + 1);
-    struct __attribute__((aligned(16))) V8 { T x[8]; } v;
-#pragma unroll
-    for (int q = 0; q < 8; ++q) {
-      const int col = col0 + q;
-      const float pv = (col < E) ? __expf(lrow[col] - mR) * ivR : 0.0f;
-      fromF(pv, v.x[q]);
-    }
-    *reinterpret_cast<V8*>(gate_out + (size_t)(m0 + row) * PX + col0) = v;
-  }
-  __syncthreads();
-  if (tid < E) {
-    uint32_t cnt = 0;
-    for (int mj = 0; mj < 128 * K; mj += 8) {
-      const u32x4 v = *reinterpret_cast<const u32x4*>(sel + mj);
-#pragma unroll
-      for (int q = 0; q < 8; ++q) {
-        const uint32_t w = v[q >> 1];
-        const uint16_t sv = (q & 1) ? (uint16_t)(w >> 16) : (uint16_t)(w & 0xffff);
-        const bool m = (sv == (uint16_t)tid);
-        localIdx[m ? (mj + q) : 128 * K] = (uint16_t)cnt;
-        cnt += m;
-      }
-    }
-    base[tid] = __hip_atomic_fetch_add(f.eCf + tid, cnt, __ATOMIC_RELAXED,
-                                       __HIP_MEMORY_SCOPE_AGENT);
-    if (f.gML) {
-      float colSum = 0.0f;
-      for (int mj = 0; mj < 128; ++mj)
-        colSum += __expf(logits[mj * (E + 1) + tid] - sMax[mj]) * sInv[mj];
-      atomicAdd(f.gML + tid, colSum / (float)S);
-      atomicAdd(f.gMeC + tid, (float)cnt / (float)S);
-    }
-  }
-  __syncthreads();
-  if (tid < 128) {
-#pragma unroll
-    for (int i = 0; i < K; ++i) {
-      const int e = mySel[i];
-      const uint32_t slot = base[e] + localIdx[tid * K + i];
-      const bool keep = slot < (uint32_t)f.EC;
-      if (keep) {
-        f.tokenIds[(size_t)e * f.pEC + slot] =
-            TPS{(uint32_t)(m0 + tid) | ((uint32_t)i << 28), mCw};
-      }
-      f.kept[(size_t)(m0 + tid) * K + i] = keep ? 1 : 0;
-    }
-  }
+  route_tile_core<T, K>(smem, f.logits32, reinterpret_cast<T*>(f.gate_out),
+                        f.tokenIds, f.eCf, f.kept, f.gML, f.gMeC, f.S, f.E,
+                        f.PX, f.EC, f.pEC, tile * 128);
 }
 
 // masked slot reduction (k_cast_combine semantics) over a token block
@@ -2413,7 +2346,7 @@ int launch_gate(hipStream_t st, const void* x, const void* gate_w,
   const size_t ldsL = gate_lds_bytes(Ec, g.esz);
   const size_t ldsR = 128 * (g.E + 1) * sizeof(float) +
                       128 * 8 * 2 * sizeof(uint16_t) + g.E * sizeof(uint32_t) +
-                      2 * 128 * sizeof(float) + 64;
+                      3 * 128 * sizeof(float) + 64;
   float* gMLp = g.cfg.is_training ? g.gML : nullptr;
   float* gMeCp = g.cfg.is_training ? g.gMeC : nullptr;
   if (g.cfg.is_training)
@@ -2435,7 +2368,7 @@ int launch_gate(hipStream_t st, const void* x, const void* gate_w,
       (void)hipFuncSetAttribute(                                                    \
           reinterpret_cast<const void*>(&k_gate_route<T, KK>),                \
           hipFuncAttributeMaxDynamicSharedMemorySize, (int)ldsR);             \
-    hipLaunchKernelGGL((k_gate_route<T, KK>), dim3(tiles), dim3(256), ldsR,   \
+    hipLaunchKernelGGL((k_gate_route<T, KK>), dim3(tiles), dim3(512), ldsR,   \
                        st, g.logits32, reinterpret_cast<T*>(gate_out),        \
                        g.tokenIds, g.eC, (int)S, g.E, g.PX, g.EC, g.pEC,      \
                        gMLp, gMeCp, g.kept);                                  \
@@ -2676,7 +2609,7 @@ static int moe_forward_fused(hipStream_t st, const void* x, const void* gate_w,
   fusedGeoDims(dn.geom, bm, bn, stg);
   arena = std::max(arena, gemmLdsBytesRT(bm, bn, BEZ, stg));
   const int routeBytes = 128 * (g.E + 1) * 4 + 128 * K * 2 +
-                         (128 * K + 2) * 2 + g.E * 4 + 2 * 128 * 4;
+                         (128 * K + 2) * 2 + g.E * 4 + 3 * 128 * 4;
   arena = std::max(arena, routeBytes);
   arena = (arena + 15) / 16 * 16 + 16;
   if (arena > 160 * 1024) return FM_FALLBACK;
