@@ -2808,8 +2808,11 @@ int fm_initialize(const fm_config* cfg, int rank, int world_size) {
                          (size_t)g.S * g.cfg.expert_top_k * g.H * g.esz));
   FM_HIP_CHECK(hipMalloc(&g.kept, (size_t)g.S * g.cfg.expert_top_k));
   if (g.cfg.dtype == 5) {
-    FM_HIP_CHECK(hipMalloc(&g.x8, (size_t)g.S * g.H));
-    FM_HIP_CHECK(hipMalloc(&g.xs, (size_t)g.S * (g.H / 64)));
+    // sized for BOTH the single-rank forward (S rows) and the EP
+    // segments path (up to nLxAlloc*pEC recv rows)
+    const size_t qRows = std::max((size_t)g.S, (size_t)g.nLxAlloc * g.pEC);
+    FM_HIP_CHECK(hipMalloc(&g.x8, qRows * g.H));
+    FM_HIP_CHECK(hipMalloc(&g.xs, qRows * (g.H / 64)));
     FM_HIP_CHECK(hipMalloc(&g.xM8, (size_t)g.nLxAlloc * g.pEC * g.P));
     FM_HIP_CHECK(hipMalloc(&g.xMs,
                            (size_t)g.nLxAlloc * g.pEC * (g.P / 64)));
@@ -3669,16 +3672,19 @@ int fm_expert_ffn_segments(void* stream, const void* rows,
                            const void* seg_expert_dev, int32_t n_segs,
                            const void* expert_w, void* out_rows) {
   if (!g.initialized) { setErr("not initialized"); return FM_ERR_STATE; }
-  if (g.cfg.dtype == 5) {
-    setErr("dtype 5 (MX fp8) supports fm_moe_forward only this round");
-    return FM_ERR_UNSUPPORTED;
-  }
   if ((int64_t)n_segs * g.EC > (int64_t)g.nLxAlloc * g.pEC) {
     setErr("segments exceed xM workspace"); return FM_ERR_SHAPE;
   }
   hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+  if (g.cfg.dtype == 5) {
+    // MX: quantize the received rows once; the MX GEMM reads x8/scales
+    int qrc = launch_quant_mx(st, rows, g.x8, g.xs,
+                              (long long)n_segs * g.EC, g.H);
+    if (qrc != FM_OK) return qrc;
+  }
   GemmArgs up{};
-  up.A = rows;
+  up.A = (g.cfg.dtype == 5) ? (const void*)g.x8 : rows;
+  up.aScales = g.xs;
   up.B = expert_w;
   up.bias = nullptr;
   up.out = g.xM;
@@ -3701,8 +3707,14 @@ int fm_expert_ffn_segments(void* stream, const void* rows,
   up.segExpert = reinterpret_cast<const int32_t*>(seg_expert_dev);
   int rc = launch_group_gemm(st, 0, up, g.EC, n_segs);
   if (rc != FM_OK) return rc;
+  if (g.cfg.dtype == 5) {
+    int qrc = launch_quant_mx(st, g.xM, g.xM8, g.xMs,
+                              (long long)n_segs * g.EC, g.P);
+    if (qrc != FM_OK) return qrc;
+  }
   GemmArgs dn = up;
-  dn.A = g.xM;
+  dn.A = (g.cfg.dtype == 5) ? (const void*)g.xM8 : g.xM;
+  dn.aScales = g.xMs;
   dn.B = reinterpret_cast<const char*>(expert_w) + (size_t)g.P * g.H * g.wesz;
   dn.out = out_rows;
   dn.strideAExpert = (long long)g.EC * g.P;

@@ -146,10 +146,11 @@ def test_p2p_transport_world1_matches_single_rank():
         moe.finalize()
 
 
-@pytest.mark.parametrize("dtype_code", [2, 4])
+@pytest.mark.parametrize("dtype_code", [2, 4, 5])
 def test_ep_pipeline_under_torchrun(dtype_code):
     """Full EP pipeline over torch.distributed (RCCL) at world = #GPUs.
-    dtype 4 = fp8e4m3 expert weights through fm_expert_ffn_segments."""
+    dtype 4 = fp8e4m3 expert weights, dtype 5 = MX-scaled fp8 (quantized
+    activations), both through fm_expert_ffn_segments."""
     n = torch.cuda.device_count()
     worker = os.path.join(REPO_ROOT, "tests", "ep_gpu_worker.py")
     cmd = [

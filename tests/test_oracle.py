@@ -216,3 +216,23 @@ def test_golden_regression(case):
     assert np.array_equal(r["eC"], data["eC"])
     np.testing.assert_allclose(r["moe_out"], data["moe_out"], rtol=1e-6, atol=1e-7)
     np.testing.assert_allclose(r["gate_out"], data["gate_out"], rtol=1e-6, atol=1e-7)
+
+
+def test_e4m3_model_matches_torch():
+    """The oracle's numpy e4m3 RNE model (the MX parity pin) must match
+    torch.float8_e4m3fn casting bit-for-bit, including ties-to-even and
+    saturation; and mx_quant_rows must be idempotent."""
+    torch = pytest.importorskip("torch")
+    from oracle.moe_oracle import e4m3_round, mx_quant_rows
+
+    rng = np.random.default_rng(7)
+    v = (rng.standard_normal(8192) * np.exp(rng.uniform(-6, 6, 8192))
+         ).astype(np.float32)
+    ours = e4m3_round(np.clip(v, -448, 448))
+    ref = torch.tensor(np.clip(v, -448, 448)).to(torch.float8_e4m3fn)
+    assert np.array_equal(ours, ref.float().numpy())
+    a = rng.standard_normal((16, 256)).astype(np.float32) * 10
+    q = mx_quant_rows(a)
+    assert np.array_equal(q, mx_quant_rows(q))  # fixed point
+    # block scales are powers of two: q / e4m3-grid ratio check via
+    # re-quantization at the same scales (covered by idempotence above)
