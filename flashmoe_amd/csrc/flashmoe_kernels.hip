@@ -508,6 +508,12 @@ struct GemmArgs {
   const void* aScales;       // MX path (dtype 5): per-64-element E8M0
                              // block scales of the quantized A operand,
                              // laid out like A with K/64 bytes per row
+  void* out8;                // MX PHASE 0 (BN=256 geometry): quantize the
+  void* outScales;           // activation tile in-register in the
+                             // epilogue (bf16-round -> 16-lane blockmax
+                             // -> e4m3) and store fp8 + E8M0 scales
+                             // directly - skips the k_quant_mx pass and
+                             // the bf16 xM round-trip entirely
 };
 
 // address-space helpers for global_load_lds (direct HBM->LDS DMA)
@@ -1341,9 +1347,10 @@ __device__ __forceinline__ bool mx_gemm_job_body(
         if constexpr (PHASE == 0) {
           v = (act == 0) ? fmaxf(v, 0.0f)
                          : 0.5f * v * (1.0f + erff(v * 0.70710678118654752f));
-          reinterpret_cast<bf16*>(a.out)[(size_t)e * a.strideOExpert +
-                                         (size_t)m * N + col] =
-              __float2bfloat16(v);
+          if (!(a.out8 && BN == 256))
+            reinterpret_cast<bf16*>(a.out)[(size_t)e * a.strideOExpert +
+                                           (size_t)m * N + col] =
+                __float2bfloat16(v);
         } else if constexpr (PHASE == 1) {
           if (slot) {
             reinterpret_cast<bf16*>(a.O32)[
@@ -1358,6 +1365,61 @@ __device__ __forceinline__ bool mx_gemm_job_body(
           reinterpret_cast<bf16*>(a.out)[(size_t)e * a.strideOExpert +
                                          (size_t)m * N + col] =
               __float2bfloat16(v);
+        }
+      }
+    }
+  }
+  if constexpr (PHASE == 0 && BN == 256) {
+    if (a.out8) {
+      // epilogue MX quantization: each quarter-wave (16 lanes x NF=4
+      // cols each) holds exactly ONE contiguous 64-col block of a row.
+      // bf16-round first (the oracle's intermediate is Element-rounded
+      // before quantization), 16-lane shuffle blockmax, E8M0 scale,
+      // e4m3 RNE pack via v_cvt_pk_fp8_f32.
+      uint8_t* o8 = reinterpret_cast<uint8_t*>(a.out8) +
+                    (size_t)e * a.strideOExpert;
+      uint8_t* oS = reinterpret_cast<uint8_t*>(a.outScales) +
+                    (size_t)e * (a.strideOExpert / 64);
+      const int blk = n0 / 64 + wc;  // this quarter-wave's 64-block
+#pragma unroll
+      for (int mi = 0; mi < MI; ++mi) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = wr * (BM / 2) + mi * 16 + r0 + r;
+          const int m = m0 + row;
+          if ((uint32_t)m >= routed) continue;
+          float vq[NF];
+          float mx = 0.0f;
+#pragma unroll
+          for (int ni = 0; ni < NF; ++ni) {
+            float v = accv[mi][ni][r] + bv[ni];
+            v = (act == 0)
+                    ? fmaxf(v, 0.0f)
+                    : 0.5f * v * (1.0f + erff(v * 0.70710678118654752f));
+            vq[ni] = __bfloat162float(__float2bfloat16(v));
+            mx = fmaxf(mx, fabsf(vq[ni]));
+          }
+#pragma unroll
+          for (int off = 1; off < 16; off <<= 1)
+            mx = fmaxf(mx, __shfl_xor(mx, off, 16));
+          int eE = 0;
+          if (mx > 0.0f) {
+            int ex;
+            (void)frexpf(mx / 448.0f, &ex);
+            eE = ex < -126 ? -126 : (ex > 127 ? 127 : ex);
+          }
+          const float inv = exp2f((float)-eE);
+          const uint32_t p01 = cvt2_fp8(vq[0] * inv, vq[1] * inv);
+          const uint32_t p23 = cvt2_fp8(vq[2] * inv, vq[3] * inv);
+          uint8_t* orow = o8 + (size_t)m * N;
+#pragma unroll
+          for (int ni = 0; ni < NF; ++ni) {
+            const int col = n0 + wc * (BN / 4) + ni * 16 + cl;
+            const uint32_t src = (ni < 2) ? p01 : p23;
+            orow[col] = (uint8_t)((src >> ((ni & 1) * 8)) & 0xFF);
+          }
+          if (cl == 0)
+            oS[(size_t)m * (N / 64) + blk] = (uint8_t)(127 + eE);
         }
       }
     }
@@ -2327,6 +2389,14 @@ void setErr(const char* msg) {
     }                                                                        \
   } while (0)
 
+// dtype 5: does the MX grouped GEMM pick the 256^2 tile for (M, N)?
+// (mirrored by launch_group_gemm's MX branch; the 256-tile up epilogue
+// quantizes in-register, making the xM quant pass unnecessary)
+static bool mxBigGeom(int M, int N, int nE) {
+  return (M >= 256) && (N >= 256) &&
+         DIVUP(M, 256) * DIVUP(N, 256) * nE >= (g.nCU > 0 ? g.nCU : 256);
+}
+
 size_t gate_lds_bytes(int E, size_t esz) {
   // mirrors the k_gate arena: logits + A chunk + G chunk
   return 128 * (E + 1) * sizeof(float) + (128 + E) * (64 + 8) * esz;
@@ -2904,8 +2974,7 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
     GemmArgs aa = a;
     aa.splitK = 1;
     aa.totalJobs = 0;
-    const bool big = (M >= 256) && (N >= 256) &&
-                     DIVUP(M, 256) * DIVUP(N, 256) * nE >= (g.nCU > 0 ? g.nCU : 256);
+    const bool big = mxBigGeom(M, N, nE);
     const int bm = big ? 256 : 128, bn = big ? 256 : 128;
     dim3 grid(DIVUP(M, bm), DIVUP(N, bn), nE);
     dim3 block(512);
@@ -3239,11 +3308,17 @@ static int moe_forward_impl(hipStream_t st, const void* x, const void* gate_w,
   up.EC = g.EC; up.pEC = g.pEC; up.PX = g.PX;
   up.topk = g.cfg.expert_top_k; up.act = g.cfg.hidden_act;
   up.expertOffset = 0; up.nRows = 0; up.H = g.H;
+  const bool mxEpiQuant =
+      (g.cfg.dtype == 5) && mxBigGeom(g.pEC, g.P, g.E);
+  if (mxEpiQuant) {
+    up.out8 = g.xM8;       // 256-tile up epilogue quantizes in-register
+    up.outScales = g.xMs;  // (skips the separate k_quant_mx xM pass)
+  }
   rc = launch_group_gemm(st, 0, up, g.pEC, g.E);
   if (rc != FM_OK) return rc;
   if (evs) FM_HIP_CHECK(hipEventRecord(evs[3], st));
 
-  if (g.cfg.dtype == 5) {
+  if (g.cfg.dtype == 5 && !mxEpiQuant) {
     // quantize the intermediate activations for the MX down GEMM
     rc = launch_quant_mx(st, g.xM, g.xM8, g.xMs,
                          (long long)g.E * g.pEC, g.P);
@@ -3251,6 +3326,8 @@ static int moe_forward_impl(hipStream_t st, const void* x, const void* gate_w,
   }
 
   GemmArgs dn = up;
+  dn.out8 = nullptr;
+  dn.outScales = nullptr;
   dn.A = (g.cfg.dtype == 5) ? (const void*)g.xM8 : g.xM;
   dn.aScales = g.xMs;
   dn.B = reinterpret_cast<const char*>(expert_w) + (size_t)g.P * g.H * g.wesz;
@@ -3705,14 +3782,22 @@ int fm_expert_ffn_segments(void* stream, const void* rows,
   up.H = g.H;
   up.splitK = 1;
   up.segExpert = reinterpret_cast<const int32_t*>(seg_expert_dev);
+  const bool mxEpiQuant =
+      (g.cfg.dtype == 5) && mxBigGeom(g.EC, g.P, n_segs);
+  if (mxEpiQuant) {
+    up.out8 = g.xM8;
+    up.outScales = g.xMs;
+  }
   int rc = launch_group_gemm(st, 0, up, g.EC, n_segs);
   if (rc != FM_OK) return rc;
-  if (g.cfg.dtype == 5) {
+  if (g.cfg.dtype == 5 && !mxEpiQuant) {
     int qrc = launch_quant_mx(st, g.xM, g.xM8, g.xMs,
                               (long long)n_segs * g.EC, g.P);
     if (qrc != FM_OK) return qrc;
   }
   GemmArgs dn = up;
+  dn.out8 = nullptr;
+  dn.outScales = nullptr;
   dn.A = (g.cfg.dtype == 5) ? (const void*)g.xM8 : g.xM;
   dn.aScales = g.xMs;
   dn.B = reinterpret_cast<const char*>(expert_w) + (size_t)g.P * g.H * g.wesz;
