@@ -1373,8 +1373,8 @@ __device__ __forceinline__ bool mx_gemm_job_body(
     if (a.out8) {
       // epilogue MX quantization: each quarter-wave (16 lanes x NF=4
       // cols each) holds exactly ONE contiguous 64-col block of a row.
-      // bf16-round first (the oracle's intermediate is Element-rounded
-      // before quantization), 16-lane shuffle blockmax, E8M0 scale,
+      // bf16-round first (the CPU restatement Element-rounds the
+      // intermediate before quantizing), 16-lane blockmax, E8M0 scale,
       // e4m3 RNE pack via v_cvt_pk_fp8_f32.
       uint8_t* o8 = reinterpret_cast<uint8_t*>(a.out8) +
                     (size_t)e * a.strideOExpert;
