@@ -2397,6 +2397,14 @@ static bool mxBigGeom(int M, int N, int nE) {
          DIVUP(M, 256) * DIVUP(N, 256) * nE >= (g.nCU > 0 ? g.nCU : 256);
 }
 
+// 256-wide-N MX geometry (256^2 or 128x256): the epilogue quantizes
+// in-register in both, so the separate xM quant pass can be skipped
+static bool mxWideN(int M, int N, int nE) {
+  if (mxBigGeom(M, N, nE)) return true;
+  return (N >= 256) &&
+         DIVUP(M, 128) * DIVUP(N, 256) * nE >= (g.nCU > 0 ? g.nCU : 256);
+}
+
 size_t gate_lds_bytes(int E, size_t esz) {
   // mirrors the k_gate arena: logits + A chunk + G chunk
   return 128 * (E + 1) * sizeof(float) + (128 + E) * (64 + 8) * esz;
@@ -2974,8 +2982,14 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
     GemmArgs aa = a;
     aa.splitK = 1;
     aa.totalJobs = 0;
+    // geometry: 256^2 when M affords it; 128x256 when only N does (the
+    // cfg4 regime, pEC=128 - still halves staging bytes/flop on the B
+    // side and keeps the 256-wide epilogue quantization); 128^2 last
     const bool big = mxBigGeom(M, N, nE);
-    const int bm = big ? 256 : 128, bn = big ? 256 : 128;
+    const bool mid = !big && (N >= 256) &&
+                     DIVUP(M, 128) * DIVUP(N, 256) * nE >=
+                         (g.nCU > 0 ? g.nCU : 256);
+    const int bm = big ? 256 : 128, bn = (big || mid) ? 256 : 128;
     dim3 grid(DIVUP(M, bm), DIVUP(N, bn), nE);
     dim3 block(512);
     {
@@ -2992,6 +3006,9 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
     do {                                                                      \
       if (big)                                                                \
         hipLaunchKernelGGL((k_group_gemm_mx<PH, AC, HB, 256, 256>), grid,     \
+                           block, 0, st, aa);                                 \
+      else if (mid)                                                           \
+        hipLaunchKernelGGL((k_group_gemm_mx<PH, AC, HB, 128, 256>), grid,     \
                            block, 0, st, aa);                                 \
       else                                                                    \
         hipLaunchKernelGGL((k_group_gemm_mx<PH, AC, HB, 128, 128>), grid,     \
@@ -3309,7 +3326,7 @@ static int moe_forward_impl(hipStream_t st, const void* x, const void* gate_w,
   up.topk = g.cfg.expert_top_k; up.act = g.cfg.hidden_act;
   up.expertOffset = 0; up.nRows = 0; up.H = g.H;
   const bool mxEpiQuant =
-      (g.cfg.dtype == 5) && mxBigGeom(g.pEC, g.P, g.E);
+      (g.cfg.dtype == 5) && mxWideN(g.pEC, g.P, g.E);
   if (mxEpiQuant) {
     up.out8 = g.xM8;       // 256-tile up epilogue quantizes in-register
     up.outScales = g.xMs;  // (skips the separate k_quant_mx xM pass)
@@ -3783,7 +3800,7 @@ int fm_expert_ffn_segments(void* stream, const void* rows,
   up.splitK = 1;
   up.segExpert = reinterpret_cast<const int32_t*>(seg_expert_dev);
   const bool mxEpiQuant =
-      (g.cfg.dtype == 5) && mxBigGeom(g.EC, g.P, n_segs);
+      (g.cfg.dtype == 5) && mxWideN(g.EC, g.P, n_segs);
   if (mxEpiQuant) {
     up.out8 = g.xM8;
     up.outScales = g.xMs;
