@@ -804,3 +804,26 @@ def test_mx_fp8_gelu_bias(fresh_moe):
     # same double-quantization tolerance note as the top-4 test above
     assert_values(out, ref["moe_out"], "bf16", "moe_out",
                   rtol=5e-2, atol_scale=5e-3)
+
+
+def test_mx_fp8_mid_geometry_epilogue_quant(fresh_moe):
+    """Shapes that engage the MX 128x256 tile WITH in-epilogue
+    quantization (jobs >= one wave of CUs): the path the big-shape
+    benches run, which the small parity shapes above never reach."""
+    cfg, path = make_cfg(torch_dtype=5, sequence_len=2048, num_experts=256,
+                         expert_top_k=2, hidden_size=128,
+                         intermediate_size=256, capacity_factor=4)
+    out, gate_out, ref, ocfg = run_pair(fresh_moe, cfg, path)
+    assert_values(out, ref["moe_out"], "bf16", "moe_out",
+                  rtol=5e-2, atol_scale=5e-3)
+
+
+def test_mx_fp8_big_geometry_epilogue_quant(fresh_moe):
+    """Shapes that engage the MX 256x256 tile + epilogue quantization
+    (M = pEC >= 256 via a high capacity factor)."""
+    cfg, path = make_cfg(torch_dtype=5, sequence_len=4096, num_experts=128,
+                         expert_top_k=2, hidden_size=128,
+                         intermediate_size=256, capacity_factor=8)
+    out, gate_out, ref, ocfg = run_pair(fresh_moe, cfg, path)
+    assert_values(out, ref["moe_out"], "bf16", "moe_out",
+                  rtol=5e-2, atol_scale=5e-3)
