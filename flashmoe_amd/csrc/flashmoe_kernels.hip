@@ -2512,11 +2512,9 @@ static bool fusedEnabled() {
   // otherwise (cfg2/3/5 measured faster multi-kernel this round -
   // profiles/r02; the fused kernel remains the FM_FUSED=1 path under
   // active optimization).
-  static const int force = [] {
-    const char* e = getenv("FM_FUSED");
-    return e ? (e[0] == '0' ? 0 : 1) : -1;
-  }();
-  if (force >= 0) return force == 1;
+  // re-read per call: tests toggle it to force fused-path coverage
+  const char* e = getenv("FM_FUSED");
+  if (e) return e[0] != '0';
   return g.P > 8192;
 }
 
@@ -3386,7 +3384,8 @@ int fm_moe_forward(void* stream, const void* x, const void* gate_w,
   void* key[8] = {const_cast<void*>(x), const_cast<void*>(gate_w),
                   const_cast<void*>(expert_w), const_cast<void*>(b_up),
                   const_cast<void*>(b_dn), gate_out, moe_out,
-                  reinterpret_cast<void*>(static_cast<uintptr_t>(S))};
+                  reinterpret_cast<void*>(static_cast<uintptr_t>(
+                      (uint64_t)S << 1 | (fusedEnabled() ? 1 : 0)))};
   if (g.graphValid && memcmp(key, g.graphKey, sizeof(key)) == 0) {
     if (hipGraphLaunch(g.graphExec, st) == hipSuccess) return FM_OK;
     g.graphValid = false;  // replay failed: rebuild next time
