@@ -248,42 +248,53 @@ __device__ __forceinline__ void route_tile_core(
     uint8_t* __restrict__ kept, float* __restrict__ gML,
     float* __restrict__ gMeC, int S, int E, int PX, int EC, int pEC,
     int m0) {
-  constexpr int BM = 128;
-  float* logits = reinterpret_cast<float*>(smem);                  // [BM][E+1]
-  uint16_t* sel = reinterpret_cast<uint16_t*>(logits + BM * (E + 1));
-  uint16_t* localIdx = sel + BM * K;
+  // TT tokens per pass: 128 normally; 64 half-passes at E > 256 so the
+  // fp32 logits tile fits LDS (64 x 513 x 4 B = 131 KB at E = 512).
+  // SUBS = 512/TT threads cooperate per token (chunk <= 64 experts each,
+  // so the taken-mask stays one u64). Pass order preserves token order,
+  // so intra-tile placement semantics are unchanged.
+  const int TT = (E > 256) ? 64 : 128;
+  const int SUBS = 512 / TT;
+  float* logits = reinterpret_cast<float*>(smem);                  // [TT][E+1]
+  uint16_t* sel = reinterpret_cast<uint16_t*>(logits + TT * (E + 1));
+  uint16_t* localIdx = sel + TT * K;
   // +2: one dump slot so the counting scan's store is UNCONDITIONAL
   // (an exec-masked conditional store compiled into ~520 serial
   // s_and_saveexec blocks = ~80% of the round-1 kernel's runtime)
-  uint32_t* base = reinterpret_cast<uint32_t*>(localIdx + BM * K + 2);  // [E]
-  float* sInv = reinterpret_cast<float*>(base + E);                // [BM] 1/d
-  float* sMax = sInv + BM;                                         // [BM]
-  float* sCw = sMax + BM;                                          // [BM] mCw
+  uint32_t* base = reinterpret_cast<uint32_t*>(localIdx + TT * K + 2);  // [E]
+  float* sInv = reinterpret_cast<float*>(base + E);                // [TT] 1/d
+  float* sMax = sInv + TT;                                         // [TT]
+  float* sCw = sMax + TT;                                          // [TT] mCw
   const int tid = threadIdx.x;
-  for (int i = tid; i < BM * E; i += 512) {
+  for (int pass = 0; pass < 128 / TT; ++pass) {
+  const int pm0 = m0 + pass * TT;
+  __syncthreads();  // prior pass's LDS readers done
+  for (int i = tid; i < TT * E; i += 512) {
     const int row = i / E, col = i % E;
-    float* src = &logits32[(size_t)(m0 + row) * E + col];
+    float* src = &logits32[(size_t)(pm0 + row) * E + col];
     logits[row * (E + 1) + col] = *src;
     *src = 0.0f;
   }
   __syncthreads();
 
   {
-    const int token = tid >> 2;  // 4 threads per token
-    const int sub = tid & 3;
-    if (token < BM) {
+    const int token = tid / SUBS;
+    const int sub = tid % SUBS;
+    if (token < TT) {
       const float* lrow = logits + token * (E + 1);
-      const int chunk = (E + 3) >> 2;  // <= 64: the taken-mask is a u64
+      const int chunk = (E + SUBS - 1) / SUBS;  // <= 64: u64 taken-mask
       const int e0 = sub * chunk;
       const int e1 = min(E, e0 + chunk);
       float m = -INFINITY;
       for (int e = e0; e < e1; ++e) m = fmaxf(m, lrow[e]);
-      m = fmaxf(m, __shfl_xor(m, 1, 4));
-      m = fmaxf(m, __shfl_xor(m, 2, 4));
+#pragma unroll
+      for (int off = 1; off < 8; off <<= 1)
+        if (off < SUBS) m = fmaxf(m, __shfl_xor(m, off, SUBS));
       float d = 0.0f;
       for (int e = e0; e < e1; ++e) d += __expf(lrow[e] - m);
-      d += __shfl_xor(d, 1, 4);
-      d += __shfl_xor(d, 2, 4);
+#pragma unroll
+      for (int off = 1; off < 8; off <<= 1)
+        if (off < SUBS) d += __shfl_xor(d, off, SUBS);
       const float inv_d = 1.0f / d;
       if (sub == 0) {
         sInv[token] = inv_d;  // the coalesced gate_out pass below
@@ -300,10 +311,12 @@ __device__ __forceinline__ void route_tile_core(
           if (!tk && lrow[e] > lv) { lv = lrow[e]; li = e; }
         }
 #pragma unroll
-        for (int off = 1; off < 4; off <<= 1) {
-          const float ov = __shfl_xor(lv, off, 4);
-          const int oi = __shfl_xor(li, off, 4);
-          if (ov > lv || (ov == lv && oi < li)) { lv = ov; li = oi; }
+        for (int off = 1; off < 8; off <<= 1) {
+          if (off < SUBS) {
+            const float ov = __shfl_xor(lv, off, SUBS);
+            const int oi = __shfl_xor(li, off, SUBS);
+            if (ov > lv || (ov == lv && oi < li)) { lv = ov; li = oi; }
+          }
         }
         if (li >= e0 && li < e1) takenM |= 1ull << (li - e0);
         if (sub == 0) {
@@ -317,7 +330,7 @@ __device__ __forceinline__ void route_tile_core(
   __syncthreads();
   // cooperative COALESCED gate_out store (recompute probs from LDS; a
   // per-token scalar store pattern was one cache line per element)
-  for (int i = tid * 8; i < BM * PX; i += 512 * 8) {
+  for (int i = tid * 8; i < TT * PX; i += 512 * 8) {
     const int row = i / PX, col0 = i % PX;
     const float mR = sMax[row], ivR = sInv[row];
     const float* lrow = logits + row * (E + 1);
@@ -328,37 +341,37 @@ __device__ __forceinline__ void route_tile_core(
       const float pv = (col < E) ? __expf(lrow[col] - mR) * ivR : 0.0f;
       fromF(pv, v.x[q]);
     }
-    *reinterpret_cast<V8*>(gate_out + (size_t)(m0 + row) * PX + col0) = v;
+    *reinterpret_cast<V8*>(gate_out + (size_t)(pm0 + row) * PX + col0) = v;
   }
   __syncthreads();
   if (tid < E) {
     // 16-B LDS reads, 8 selections per iteration (a scalar dependent
     // scan exposed ~50 cycles of LDS latency per entry: ~7 us/fwd)
     uint32_t cnt = 0;
-    for (int mj = 0; mj < BM * K; mj += 8) {
+    for (int mj = 0; mj < TT * K; mj += 8) {
       const u32x4 v = *reinterpret_cast<const u32x4*>(sel + mj);
 #pragma unroll
       for (int q = 0; q < 8; ++q) {
         const uint32_t w = v[q >> 1];
         const uint16_t sv = (q & 1) ? (uint16_t)(w >> 16) : (uint16_t)(w & 0xffff);
         const bool mt = (sv == (uint16_t)tid);
-        localIdx[mt ? (mj + q) : BM * K] = (uint16_t)cnt;
+        localIdx[mt ? (mj + q) : TT * K] = (uint16_t)cnt;
         cnt += mt;
       }
     }
     base[tid] = __hip_atomic_fetch_add(eC + tid, cnt, __ATOMIC_RELAXED,
                                        __HIP_MEMORY_SCOPE_AGENT);
     if (gML) {
-      // column sum of probs for this expert over the tile's tokens
+      // column sum of probs for this expert over the pass's tokens
       float colSum = 0.0f;
-      for (int mj = 0; mj < BM; ++mj)
+      for (int mj = 0; mj < TT; ++mj)
         colSum += __expf(logits[mj * (E + 1) + tid] - sMax[mj]) * sInv[mj];
       atomicAdd(gML + tid, colSum / (float)S);
       atomicAdd(gMeC + tid, (float)cnt / (float)S);
     }
   }
   __syncthreads();
-  if (tid < BM) {
+  if (tid < TT) {
 #pragma unroll
     for (int i = 0; i < K; ++i) {
       const int e = sel[tid * K + i];
@@ -366,11 +379,12 @@ __device__ __forceinline__ void route_tile_core(
       const bool keep = slot < (uint32_t)EC;
       if (keep) {
         tokenIds[(size_t)e * pEC + slot] =
-            TPS{(uint32_t)(m0 + tid) | ((uint32_t)i << 28), sCw[tid]};
+            TPS{(uint32_t)(pm0 + tid) | ((uint32_t)i << 28), sCw[tid]};
       }
-      if (kept) kept[(size_t)(m0 + tid) * K + i] = keep ? 1 : 0;
+      if (kept) kept[(size_t)(pm0 + tid) * K + i] = keep ? 1 : 0;
     }
   }
+  }  // pass loop
 }
 
 template <typename T, int K>
@@ -2422,9 +2436,11 @@ int launch_gate(hipStream_t st, const void* x, const void* gate_w,
   const int Ec = g.E < 128 ? g.E : 128;
   const bool mfmaLogits = (g.esz == 2);  // bf16/fp16: MFMA logits GEMM
   const size_t ldsL = gate_lds_bytes(Ec, g.esz);
-  const size_t ldsR = 128 * (g.E + 1) * sizeof(float) +
-                      128 * 8 * 2 * sizeof(uint16_t) + g.E * sizeof(uint32_t) +
-                      3 * 128 * sizeof(float) + 64;
+  // route arena: TT = 64-token half-passes at E > 256 (route_tile_core)
+  const int rTT = (g.E > 256) ? 64 : 128;
+  const size_t ldsR = (size_t)rTT * (g.E + 1) * sizeof(float) +
+                      rTT * 8 * 2 * sizeof(uint16_t) + g.E * sizeof(uint32_t) +
+                      3 * rTT * sizeof(float) + 64;
   float* gMLp = g.cfg.is_training ? g.gML : nullptr;
   float* gMeCp = g.cfg.is_training ? g.gMeC : nullptr;
   if (g.cfg.is_training)
@@ -2684,8 +2700,9 @@ static int moe_forward_fused(hipStream_t st, const void* x, const void* gate_w,
   arena = std::max(arena, gemmLdsBytesRT(bm, bn, BEZ, stg));
   fusedGeoDims(dn.geom, bm, bn, stg);
   arena = std::max(arena, gemmLdsBytesRT(bm, bn, BEZ, stg));
-  const int routeBytes = 128 * (g.E + 1) * 4 + 128 * K * 2 +
-                         (128 * K + 2) * 2 + g.E * 4 + 3 * 128 * 4;
+  const int rTT = (g.E > 256) ? 64 : 128;  // route half-passes at big E
+  const int routeBytes = rTT * (g.E + 1) * 4 + rTT * K * 2 +
+                         (rTT * K + 2) * 2 + g.E * 4 + 3 * rTT * 4;
   arena = std::max(arena, routeBytes);
   arena = (arena + 15) / 16 * 16 + 16;
   if (arena > 160 * 1024) return FM_FALLBACK;
@@ -2845,7 +2862,7 @@ int fm_initialize(const fm_config* cfg, int rank, int world_size) {
     setErr("dtype 5 (MX fp8) requires H, P multiples of 128 (BK=128)");
     return FM_ERR_SHAPE;
   }
-  if (g.E > 256) { setErr("E > 256 not supported this round"); return FM_ERR_UNSUPPORTED; }
+  if (g.E > 512) { setErr("E > 512 not supported this round"); return FM_ERR_UNSUPPORTED; }
   if (world_size > 64) {
     // the P2P peer tables (State.peerBase, fm_heap_connect locals) are
     // sized for 64 ranks - a full 8-node envelope; reject beyond it

@@ -897,3 +897,27 @@ def test_fused_training_aux(force_fused, fresh_moe):
     assert np.allclose(gML, want_gML, rtol=2e-2, atol=1e-3)
     assert np.allclose(gMeC, want_gMeC, rtol=1e-6, atol=1e-6)
     assert_values(out, ref["moe_out"], "bf16", "moe_out")
+
+
+def test_many_experts_512(fresh_moe):
+    """E = 512 (the round-2 cap lift; >256 runs the route's 64-token
+    half-pass mode with 8 threads/token): single tile, routing bit-exact
+    vs the oracle, values at the bf16 bar."""
+    cfg, path = make_cfg(num_experts=512, expert_top_k=4, sequence_len=128,
+                         hidden_size=128, intermediate_size=128,
+                         capacity_factor=2)
+    out, gate_out, ref, ocfg = run_pair(fresh_moe, cfg, path)
+    E, EC = 512, 2 * 4 * 1  # ceil(128/512)=1, CF=2, k=4
+    counts, tok, ps = routing_from_lib(E, EC)
+    clipped = np.minimum(ref["eC"], EC)
+    assert np.array_equal(counts.astype(np.int64), clipped)
+    assert_values(gate_out, ref["gate_out"], "bf16", "gate_out")
+    assert_values(out, ref["moe_out"], "bf16", "moe_out")
+
+
+def test_many_experts_384_multi_tile(fresh_moe):
+    cfg, path = make_cfg(num_experts=384, expert_top_k=2, sequence_len=512,
+                         hidden_size=128, intermediate_size=128,
+                         capacity_factor=8)
+    out, gate_out, ref, ocfg = run_pair(fresh_moe, cfg, path)
+    assert_values(out, ref["moe_out"], "bf16", "moe_out")

@@ -30,7 +30,7 @@ def EC_of(cfg):
 
 fails = 0
 for case in range(n_cases):
-    E = rng.choice([1, 2, 4, 8, 16, 32, 64, 96, 128, 192, 256])
+    E = rng.choice([1, 2, 4, 8, 16, 32, 64, 96, 128, 192, 256, 384, 512])
     k = rng.choice([kk for kk in (1, 2, 3, 4, 5, 6, 7, 8) if kk <= E])
     H = rng.choice([128, 256, 512, 1024])
     P = rng.choice([128, 256, 512, 1024, 2048])
