@@ -58,6 +58,7 @@ template <int N> __device__ __forceinline__ void wait_vmcnt() {
   else if constexpr (N == 5) asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
   else if constexpr (N == 6) asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
   else if constexpr (N == 8) asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+  else if constexpr (N == 10) asm volatile("s_waitcnt vmcnt(10)" ::: "memory");
   else if constexpr (N == 12) asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
   else static_assert(N == 0, "unsupported vmcnt literal");
 }
@@ -1147,12 +1148,19 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_bf16_big(GemmArgs a) {
 // PHASE semantics match gemm_job_body (0 up / 1 down+slots / 2 packed).
 // ---------------------------------------------------------------------------
 
-template <int PHASE, int BM = 128, int BN = 128>
+#define MI_OF(BMV) ((BMV) / 32)
+template <int PHASE, int BM = 128, int BN = 128, int STAGES = 2>
 __device__ __forceinline__ bool mx_gemm_job_body(
     const GemmArgs& a, char* smemBase, int e, int m0, int n0, int act,
     bool hasBias) {
   constexpr int BK = 128;
-  constexpr int SE = 2;
+  // triple buffering (SE=3) gives each stage TWO compute phases of
+  // latency budget (the 2-stage pipeline's per-tile vmcnt(0) drain is
+  // the measured stall: MFMA pipe ~12%, profiles/r02_mx_pmc.txt);
+  // degrade when it would not fit LDS
+  constexpr int SE =
+      (STAGES == 3 && 3 * (BM + BN) * BK + BM * 8 + 16 > 160 * 1024)
+          ? 2 : STAGES;
   uint8_t* Abase = reinterpret_cast<uint8_t*>(smemBase);   // SE x [BM][BK]
   uint8_t* Bbase = reinterpret_cast<uint8_t*>(smemBase + SE * BM * BK);
   TPS* sTps = reinterpret_cast<TPS*>(smemBase + SE * BM * BK + SE * BN * BK);
@@ -1250,19 +1258,34 @@ __device__ __forceinline__ bool mx_gemm_job_body(
     sRow[mi] = aS + (aBase / 64) + arow * K64 + (gk & 1);
   }
   const int nK = K / BK;
+  // counted end-of-tile wait: with SE=3, tile t+1 must have landed but
+  // tile t+2's GPT glds AND the MI prefetched scale-byte loads (issued
+  // after them, completing in order) may stay in flight
+  constexpr int GPW_AB = GPW_A + GPW_B;
+  constexpr int WN3 = (GPW_AB + MI_OF(BM) <= 10) ? 10 : 12;
   stage(0, 0);
-  // software-pipelined scale bytes: tile t's 4 scattered global byte
+  // software-pipelined scale bytes: tile t's scattered global byte
   // loads are issued one tile AHEAD so they never sit on the MFMA
   // dependency chain
   int sav[MI], savNext[MI];
 #pragma unroll
   for (int mi = 0; mi < MI; ++mi) sav[mi] = sRow[mi][0];
-  wait_vmcnt<0>();
+  if constexpr (SE == 3) {
+    if (nK > 1) {
+      stage(BK, 1);
+      wait_vmcnt<GPW_AB>();  // tile 0 landed; tile 1 in flight
+    } else {
+      wait_vmcnt<0>();
+    }
+  } else {
+    wait_vmcnt<0>();
+  }
   __builtin_amdgcn_s_barrier();
   for (int t = 0; t < nK; ++t) {
-    const bool stageNow = t + 1 < nK;
-    if (stageNow) {
-      stage((t + 1) * BK, (t + 1) % SE);
+    const int pre = t + SE - 1;      // tile staged this iteration
+    const bool stageNow = pre < nK;
+    if (stageNow) stage(pre * BK, pre % SE);
+    if (t + 1 < nK) {
 #pragma unroll
       for (int mi = 0; mi < MI; ++mi) savNext[mi] = sRow[mi][(t + 1) * 2];
     }
@@ -1307,7 +1330,11 @@ __device__ __forceinline__ bool mx_gemm_job_body(
     __builtin_amdgcn_s_setprio(0);
 #pragma unroll
     for (int mi = 0; mi < MI; ++mi) sav[mi] = savNext[mi];
-    wait_vmcnt<0>();
+    if constexpr (SE == 3) {
+      if (stageNow) wait_vmcnt<WN3>(); else wait_vmcnt<0>();
+    } else {
+      wait_vmcnt<0>();
+    }
     __builtin_amdgcn_s_barrier();
   }
 
@@ -1441,9 +1468,13 @@ __device__ __forceinline__ bool mx_gemm_job_body(
   return true;
 }
 
-template <int PHASE, int ACT, bool HAS_BIAS, int BM = 128, int BN = 128>
+template <int PHASE, int ACT, bool HAS_BIAS, int BM = 128, int BN = 128,
+          int STAGES = 2>
 __global__ __launch_bounds__(512, 2) void k_group_gemm_mx(GemmArgs a) {
-  constexpr int BK = 128, SE = 2;
+  constexpr int BK = 128;
+  constexpr int SE =
+      (STAGES == 3 && 3 * (BM + BN) * BK + BM * 8 + 16 > 160 * 1024)
+          ? 2 : STAGES;
   __shared__ __attribute__((aligned(16))) char smem[
       SE * BM * BK + SE * BN * BK + BM * 8 + 16];
   const int mT = a.totalJobs > 0 ? a.jobsMT : gridDim.x;
@@ -1460,8 +1491,8 @@ __global__ __launch_bounds__(512, 2) void k_group_gemm_mx(GemmArgs a) {
         : (xcd < rx ? xcd * (qx + 1) : rx * (qx + 1) + (xcd - rx) * qx) + pos;
     const int e = swz / (mT * nT);
     const int rem = swz % (mT * nT);
-    (void)mx_gemm_job_body<PHASE, BM, BN>(a, smem, e, (rem % mT) * BM,
-                                          (rem / mT) * BN, ACT, HAS_BIAS);
+    (void)mx_gemm_job_body<PHASE, BM, BN, STAGES>(
+        a, smem, e, (rem % mT) * BM, (rem / mT) * BN, ACT, HAS_BIAS);
   }
 }
 
@@ -3005,6 +3036,10 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
                      DIVUP(M, 128) * DIVUP(N, 256) * nE >=
                          (g.nCU > 0 ? g.nCU : 256);
     const int bm = big ? 256 : 128, bn = (big || mid) ? 256 : 128;
+    static const bool mxS3 = [] {
+      const char* e2 = getenv("FM_MX_STAGES");
+      return !(e2 && e2[0] == '2');
+    }();
     dim3 grid(DIVUP(M, bm), DIVUP(N, bn), nE);
     dim3 block(512);
     {
@@ -3021,6 +3056,9 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
     do {                                                                      \
       if (big)                                                                \
         hipLaunchKernelGGL((k_group_gemm_mx<PH, AC, HB, 256, 256>), grid,     \
+                           block, 0, st, aa);                                 \
+      else if (mid && mxS3)                                                   \
+        hipLaunchKernelGGL((k_group_gemm_mx<PH, AC, HB, 128, 256, 3>), grid,  \
                            block, 0, st, aa);                                 \
       else if (mid)                                                           \
         hipLaunchKernelGGL((k_group_gemm_mx<PH, AC, HB, 128, 256>), grid,     \
