@@ -2442,12 +2442,13 @@ static bool mxBigGeom(int M, int N, int nE) {
          DIVUP(M, 256) * DIVUP(N, 256) * nE >= (g.nCU > 0 ? g.nCU : 256);
 }
 
-// 256-wide-N MX geometry (256^2 or 128x256): the epilogue quantizes
-// in-register in both, so the separate xM quant pass can be skipped
+// 256-wide-N MX geometry (128x256 default or FM_MX_GEOM=big): the
+// epilogue quantizes in-register in both, so the xM quant pass is
+// skipped whenever the up GEMM runs a 256-wide tile
 static bool mxWideN(int M, int N, int nE) {
-  if (mxBigGeom(M, N, nE)) return true;
   return (N >= 256) &&
-         DIVUP(M, 128) * DIVUP(N, 256) * nE >= (g.nCU > 0 ? g.nCU : 256);
+         (mxBigGeom(M, N, nE) ||
+          DIVUP(M, 128) * DIVUP(N, 256) * nE >= (g.nCU > 0 ? g.nCU : 256));
 }
 
 size_t gate_lds_bytes(int E, size_t esz) {
@@ -3031,7 +3032,11 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
     // geometry: 256^2 when M affords it; 128x256 when only N does (the
     // cfg4 regime, pEC=128 - still halves staging bytes/flop on the B
     // side and keeps the 256-wide epilogue quantization); 128^2 last
-    const bool big = mxBigGeom(M, N, nE);
+    static const int forceGeom = [] {  // 0 auto, 1 force mid (A/B knob)
+      const char* e2 = getenv("FM_MX_FORCE_MID");
+      return (e2 && e2[0] == '1') ? 1 : 0;
+    }();
+    const bool big = !forceGeom && mxBigGeom(M, N, nE);
     const bool mid = !big && (N >= 256) &&
                      DIVUP(M, 128) * DIVUP(N, 256) * nE >=
                          (g.nCU > 0 ? g.nCU : 256);
