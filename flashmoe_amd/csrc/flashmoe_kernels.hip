@@ -3029,14 +3029,17 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
     GemmArgs aa = a;
     aa.splitK = 1;
     aa.totalJobs = 0;
-    // geometry: 256^2 when M affords it; 128x256 when only N does (the
-    // cfg4 regime, pEC=128 - still halves staging bytes/flop on the B
-    // side and keeps the 256-wide epilogue quantization); 128^2 last
-    static const int forceGeom = [] {  // 0 auto, 1 force mid (A/B knob)
-      const char* e2 = getenv("FM_MX_FORCE_MID");
-      return (e2 && e2[0] == '1') ? 1 : 0;
+    // geometry: 128x256 triple-buffered is the measured default
+    // wherever N affords 256-wide tiles (cfg3/4/5 shapes: -7..-12% -
+    // the 2-stage pipeline's per-tile drain was the stall,
+    // profiles/r02_mx_pmc.txt; the 256^2 SE2 tile halves B re-reads but
+    // loses more to latency, kept behind FM_MX_GEOM=big for A/B);
+    // 128^2 for narrow N
+    static const int forceBig = [] {
+      const char* e2 = getenv("FM_MX_GEOM");
+      return (e2 && e2[0] == 'b') ? 1 : 0;
     }();
-    const bool big = !forceGeom && mxBigGeom(M, N, nE);
+    const bool big = forceBig && mxBigGeom(M, N, nE);
     const bool mid = !big && (N >= 256) &&
                      DIVUP(M, 128) * DIVUP(N, 256) * nE >=
                          (g.nCU > 0 ? g.nCU : 256);
