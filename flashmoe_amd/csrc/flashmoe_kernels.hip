@@ -3175,7 +3175,18 @@ static int launch_group_gemm(hipStream_t st, int phase, const GemmArgs& a,
         if (a3.eC)  // PHASE 3 big kernel does not zero eC; do it here
           FM_HIP_CHECK(hipMemsetAsync(const_cast<uint32_t*>(a3.eC), 0,
                                       (size_t)a3.EC * sizeof(uint32_t), st));
-        dim3 g3(DIVUP(M, bmSel), DIVUP(N, bnSel), 1);
+        // K-split to fill the chip when the logits tile grid is narrow
+        // (cfg5: 128 x 2 tiles left half the CUs idle): partials
+        // atomicAdd into logits32, which is kept zero between forwards
+        // by the route's re-zeroing pass
+        {
+          const int b3 = DIVUP(M, bmSel) * DIVUP(N, bnSel);
+          while (a3.splitK < 8 && b3 * a3.splitK * 2 <= 512 &&
+                 (a.K / 64) % (a3.splitK * 2) == 0)
+            a3.splitK *= 2;
+          a3.atomicLogits = (a3.splitK > 1) ? 1 : 0;
+        }
+        dim3 g3(DIVUP(M, bmSel), DIVUP(N, bnSel), a3.splitK);
         if (g.cfg.dtype == 3) {
           if (mode == 0)
             hipLaunchKernelGGL((k_group_gemm_bf16_big<fp16, 3, 0, false, 256, 256>), g3, dim3(512), 0, st, a3);
