@@ -921,3 +921,20 @@ def test_many_experts_384_multi_tile(fresh_moe):
                          capacity_factor=8)
     out, gate_out, ref, ocfg = run_pair(fresh_moe, cfg, path)
     assert_values(out, ref["moe_out"], "bf16", "moe_out")
+
+
+def test_fused_many_experts_384(force_fused, fresh_moe):
+    """Fused kernel + E>256 (the route's 64-token half-pass mode runs
+    INSIDE k_moe_fused's inline route)."""
+    cfg, path = make_cfg(num_experts=384, expert_top_k=2, sequence_len=512,
+                         hidden_size=128, intermediate_size=128,
+                         capacity_factor=8)
+    out, gate_out, ref, _ = run_pair(fresh_moe, cfg, path)
+    assert_values(out, ref["moe_out"], "bf16", "moe_out")
+
+
+def test_fused_odd64_shapes(force_fused, fresh_moe):
+    cfg, path = make_cfg(hidden_size=192, intermediate_size=320,
+                         sequence_len=256, capacity_factor=2)
+    out, gate_out, ref, _ = run_pair(fresh_moe, cfg, path)
+    assert_values(out, ref["moe_out"], "bf16", "moe_out")
