@@ -81,6 +81,9 @@ def finalize():
     _state.update(initialized=False, cfg=None, gate_out=None)
     _state.pop("p2p", None)        # heap freed by fm_finalize
     _state.pop("ep_buffers", None)
+    # a P2P setup failure is scoped to one initialize/finalize cycle
+    _state.pop("p2p_failed", None)
+    _state.pop("p2p_validated", None)
 
 
 def get_compiled_config() -> dict:
