@@ -27,6 +27,12 @@ CASES = {
     "small_bf16_top2": dict(
         S=128, H=64, P=96, cfg=dict(num_experts=8, expert_top_k=2, element="bf16")
     ),
+    # dtype 5 (MX fp8): pins the e4m3 RNE + per-64-block E8M0 activation
+    # quantization model (H, P multiples of 64 so blocks are whole)
+    "small_mx_top2": dict(
+        S=128, H=64, P=128, cfg=dict(num_experts=8, expert_top_k=2,
+                                     element="bf16", mx_fp8=True)
+    ),
 }
 
 

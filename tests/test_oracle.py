@@ -201,7 +201,8 @@ def test_empty_input_and_single_token():
     assert r1["eC"].sum() == 1
 
 
-@pytest.mark.parametrize("case", ["small_fp32_top1", "small_fp32_top2", "small_bf16_top2"])
+@pytest.mark.parametrize("case", ["small_fp32_top1", "small_fp32_top2",
+                                  "small_bf16_top2", "small_mx_top2"])
 def test_golden_regression(case):
     path = os.path.join(GOLDEN_DIR, f"{case}.npz")
     meta_path = os.path.join(GOLDEN_DIR, f"{case}.json")
