@@ -819,7 +819,7 @@ def _assert_mx_values(out, want):
     ok = np.isclose(got, want, rtol=5e-2, atol=5e-3 * scale)
     frac_bad = 1.0 - ok.mean()
     max_err = float(np.abs(got - want).max())
-    assert frac_bad <= 1e-4 and max_err <= 0.02 * scale, (
+    assert frac_bad <= 1e-3 and max_err <= 0.025 * scale, (
         f"moe_out: {100*frac_bad:.4f}% outside tol, max err {max_err:.3f} "
         f"(scale {scale:.1f})")
 
