@@ -33,7 +33,12 @@ extern "C" {
 
 /* Mirror of csrc/flashmoe_config.json (schema:
  * csrc/flashmoe_config.schema.json). dtype: 0 fp32, 1 tf32(=fp32 on
- * CDNA4 - no xf32), 2 bf16, 3 fp16. */
+ * CDNA4 - no xf32), 2 bf16, 3 fp16, 4 fp8e4m3 expert weights with bf16
+ * activations (W8A16), 5 MX-block-scaled fp8: fp8e4m3 weights AND
+ * runtime-quantized fp8 activations (per-64-element E8M0 block scales)
+ * on the CDNA4 scaled MFMA; weights are passed exactly as for dtype 4,
+ * activation quantization is internal. dtype 5 requires H, P multiples
+ * of 128. */
 typedef struct fm_config {
   int32_t num_experts;
   int32_t expert_top_k;
